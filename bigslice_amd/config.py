@@ -1,0 +1,48 @@
+"""Global configuration knobs for the engine.
+
+The reference makes its global chunk size a flag
+(internal/defaultsize/size.go:14-19, default 128 rows).  On MI355X the unit of
+work is a device batch: a column-parallel chunk sized to keep all 256 CUs busy
+and to amortize kernel-launch overhead against 288 GB of HBM3E.  We therefore
+default to multi-million-row device batches and small (64k) host batches.
+"""
+
+import os
+
+
+def _env_int(name: str, default: int) -> int:
+    try:
+        return int(os.environ.get(name, default))
+    except ValueError:
+        return default
+
+
+# Rows per device batch flowing through a task pipeline on GPU.
+DEVICE_CHUNK_ROWS = _env_int("BIGSLICE_DEVICE_CHUNK_ROWS", 8 << 20)
+
+# Rows per host (CPU) batch; CPU paths are for plumbing/tests, keep it modest.
+HOST_CHUNK_ROWS = _env_int("BIGSLICE_HOST_CHUNK_ROWS", 1 << 16)
+
+# Seed used by the combiner hash table so a previous partitioning step does
+# not strip hash entropy (reference: exec/combiner.go:43).
+COMBINER_HASH_SEED = 0x9ACB0442
+
+# In-memory target number of keys for a spilling combiner before it spills
+# (reference: exec/bigmachine.go:1105 uses chunk*100; ours is device-scaled).
+COMBINER_TARGET_KEYS = _env_int("BIGSLICE_COMBINER_TARGET_KEYS", 1 << 26)
+
+# Spill target bytes for external sort runs (reference cogroup.go:126-127
+# uses 32 MiB; device-scaled default 4 GiB keeps runs HBM-resident).
+SORT_SPILL_TARGET_BYTES = _env_int("BIGSLICE_SORT_SPILL_BYTES", 4 << 30)
+
+# Maximum consecutive losses of a single task before giving up
+# (reference: exec/eval.go:30).
+MAX_CONSECUTIVE_LOST = 5
+
+# Default parallelism for the local executor (concurrent shard tasks).
+DEFAULT_PARALLELISM = _env_int("BIGSLICE_PARALLELISM", 8)
+
+
+def chunk_rows(device: str) -> int:
+    """Default batch size for a given device kind ('cuda' or 'cpu')."""
+    return DEVICE_CHUNK_ROWS if device.startswith("cuda") else HOST_CHUNK_ROWS

@@ -1,0 +1,240 @@
+"""Device-resident columnar batches.
+
+Role-parity with the reference's frame package (frame/frame.go:82-92): a Frame
+is a set of equal-length typed columns with a key prefix, supporting zero-copy
+slicing, copy/append, per-row hashing of the prefix, and lexicographic
+ordering by prefix.  Here each numeric column is a torch tensor living in
+MI355X HBM3E (or host memory on the CPU path); OBJECT columns are Python lists
+on the host and only flow through CPU pipelines.
+
+The heavy per-row "ops" of the reference (frame/ops.go: hash/less/swap/encode)
+become whole-column device kernels: hashing is the murmur3 HIP kernel
+(bigslice_amd/csrc), ordering is torch/rocPRIM sort, and copies are
+hipMemcpyAsync under torch.
+"""
+
+from __future__ import annotations
+
+from typing import Iterable, List, Optional, Sequence, Union
+
+import numpy as np
+import torch
+
+from . import hashing
+from .schema import OBJECT, Schema, infer_dtype, is_object
+
+
+Column = Union[torch.Tensor, list]
+
+
+def _col_len(col: Column) -> int:
+    return col.shape[0] if isinstance(col, torch.Tensor) else len(col)
+
+
+def _col_dtype(col: Column):
+    return col.dtype if isinstance(col, torch.Tensor) else OBJECT
+
+
+class Frame:
+    """Equal-length typed columns with a key prefix."""
+
+    __slots__ = ("columns", "prefix")
+
+    def __init__(self, columns: Sequence[Column], prefix: int = None):
+        self.columns: List[Column] = list(columns)
+        n = None
+        for c in self.columns:
+            cl = _col_len(c)
+            if n is None:
+                n = cl
+            elif cl != n:
+                raise ValueError(f"ragged frame: column lengths {n} vs {cl}")
+        if prefix is None:
+            prefix = 1 if self.columns else 0
+        self.prefix = prefix
+
+    # -- construction ----------------------------------------------------
+
+    @staticmethod
+    def from_lists(cols: Sequence[list], prefix: int = None,
+                   schema: Schema = None, device: str = "cpu") -> "Frame":
+        """Build a frame from Python lists, inferring dtypes per column."""
+        out: List[Column] = []
+        for i, c in enumerate(cols):
+            dt = schema.dtypes[i] if schema else (
+                infer_dtype(c[0]) if c else OBJECT)
+            if is_object(dt):
+                out.append(list(c))
+            else:
+                out.append(torch.tensor(c, dtype=dt, device=device))
+        if schema is not None and prefix is None:
+            prefix = schema.prefix
+        return Frame(out, prefix)
+
+    @staticmethod
+    def empty(schema: Schema, device: str = "cpu") -> "Frame":
+        cols: List[Column] = []
+        for dt in schema.dtypes:
+            if is_object(dt):
+                cols.append([])
+            else:
+                cols.append(torch.empty(0, dtype=dt, device=device))
+        return Frame(cols, schema.prefix)
+
+    # -- basic properties -------------------------------------------------
+
+    def __len__(self) -> int:
+        return _col_len(self.columns[0]) if self.columns else 0
+
+    @property
+    def num_columns(self) -> int:
+        return len(self.columns)
+
+    @property
+    def schema(self) -> Schema:
+        return Schema([_col_dtype(c) for c in self.columns], self.prefix)
+
+    @property
+    def device(self) -> str:
+        for c in self.columns:
+            if isinstance(c, torch.Tensor):
+                return str(c.device)
+        return "cpu"
+
+    @property
+    def has_objects(self) -> bool:
+        return any(not isinstance(c, torch.Tensor) for c in self.columns)
+
+    def nbytes(self) -> int:
+        total = 0
+        for c in self.columns:
+            if isinstance(c, torch.Tensor):
+                total += c.numel() * c.element_size()
+            else:
+                total += sum(len(str(x)) for x in c)  # rough
+        return total
+
+    # -- slicing / copying -------------------------------------------------
+
+    def slice(self, start: int, stop: int) -> "Frame":
+        """Zero-copy row range (reference frame.Slice, frame.go:244)."""
+        return Frame(
+            [c[start:stop] for c in self.columns], self.prefix)
+
+    def select(self, indices: torch.Tensor) -> "Frame":
+        """Gather rows by index tensor (device gather kernel)."""
+        cols: List[Column] = []
+        for c in self.columns:
+            if isinstance(c, torch.Tensor):
+                cols.append(c[indices.to(c.device)])
+            else:
+                idx = indices.cpu().tolist()
+                cols.append([c[i] for i in idx])
+        return Frame(cols, self.prefix)
+
+    def mask(self, keep: torch.Tensor) -> "Frame":
+        """Filter rows by boolean mask (compaction; reference Filter's
+        frame.Copy loop, slice.go:688-722, as one device kernel)."""
+        cols: List[Column] = []
+        for c in self.columns:
+            if isinstance(c, torch.Tensor):
+                cols.append(c[keep.to(c.device)])
+            else:
+                km = keep.cpu().numpy()
+                cols.append([x for x, k in zip(c, km) if k])
+        return Frame(cols, self.prefix)
+
+    @staticmethod
+    def concat(frames: Sequence["Frame"]) -> "Frame":
+        """Concatenate frames row-wise (reference AppendFrame)."""
+        frames = [f for f in frames if len(f) > 0]
+        if not frames:
+            raise ValueError("concat of no rows; use Frame.empty")
+        first = frames[0]
+        if len(frames) == 1:
+            return first
+        cols: List[Column] = []
+        for i in range(first.num_columns):
+            parts = [f.columns[i] for f in frames]
+            if isinstance(parts[0], torch.Tensor):
+                cols.append(torch.cat(parts))
+            else:
+                merged: list = []
+                for p in parts:
+                    merged.extend(p)
+                cols.append(merged)
+        return Frame(cols, first.prefix)
+
+    def to(self, device: str, non_blocking: bool = False) -> "Frame":
+        cols: List[Column] = []
+        for c in self.columns:
+            if isinstance(c, torch.Tensor):
+                cols.append(c.to(device, non_blocking=non_blocking))
+            else:
+                if device != "cpu":
+                    raise ValueError(
+                        "object columns cannot move to device memory")
+                cols.append(c)
+        return Frame(cols, self.prefix)
+
+    def clone(self) -> "Frame":
+        return Frame(
+            [c.clone() if isinstance(c, torch.Tensor) else list(c)
+             for c in self.columns], self.prefix)
+
+    def with_prefix(self, prefix: int) -> "Frame":
+        if not (0 < prefix <= self.num_columns):
+            raise ValueError(f"invalid prefix {prefix}")
+        return Frame(self.columns, prefix)
+
+    # -- key operations ----------------------------------------------------
+
+    def hash(self, seed: int = 0) -> torch.Tensor:
+        """32-bit murmur3 hash of the prefix columns, XOR-combined
+        (bit-identical to reference frame.HashWithSeed, frame/frame.go:395-401
+        + ops_builtin.go:140-164, so partition assignment is reproducible).
+
+        Returns a uint32-valued tensor (as int64 on CPU path for numpy ease,
+        uint32 on device).
+        """
+        return hashing.hash_columns(self.columns[: self.prefix], seed)
+
+    def argsort_by_prefix(self) -> torch.Tensor:
+        """Row permutation sorting lexicographically by prefix columns
+        (reference frame.Less, frame/frame.go:375-385)."""
+        n = len(self)
+        cols = self.columns[: self.prefix]
+        if any(not isinstance(c, torch.Tensor) for c in cols):
+            keys = list(zip(*[list(c) if not isinstance(c, torch.Tensor)
+                              else c.cpu().tolist() for c in cols]))
+            order = sorted(range(n), key=lambda i: keys[i])
+            return torch.tensor(order, dtype=torch.int64)
+        # Stable sorts applied from least- to most-significant column give
+        # a lexicographic order.
+        perm = torch.arange(n, device=cols[0].device)
+        for c in reversed(cols):
+            _, o = torch.sort(c[perm], stable=True)
+            perm = perm[o]
+        return perm
+
+    def sort_by_prefix(self) -> "Frame":
+        return self.select(self.argsort_by_prefix())
+
+    # -- conversion ---------------------------------------------------------
+
+    def column_lists(self) -> List[list]:
+        out = []
+        for c in self.columns:
+            if isinstance(c, torch.Tensor):
+                out.append(c.cpu().tolist())
+            else:
+                out.append(list(c))
+        return out
+
+    def rows(self) -> list:
+        """Rows as tuples (single-column frames yield scalars in scan())."""
+        return list(zip(*self.column_lists())) if self.columns else []
+
+    def __repr__(self):
+        return (f"Frame({len(self)} rows, {self.schema}, "
+                f"device={self.device})")

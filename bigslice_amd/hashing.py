@@ -1,0 +1,153 @@
+"""Murmur3-32 hashing, bit-identical to the reference.
+
+The reference hashes key columns with seeded murmur3-32 over little-endian
+bytes (frame/ops_builtin.go:140-164: scalars of width<=4 via 4 LE bytes,
+8-byte scalars via 8 LE bytes; strings over their bytes; bool -> seed+{0,1};
+frame/frame.go:395-401 XOR-combines across prefix columns).  Keeping the hash
+bit-identical keeps partition assignment reproducible against the reference.
+
+Host path: numpy-vectorized murmur3 (uint32 wraparound arithmetic).
+Device path: the same algorithm as a HIP kernel (csrc/hash_partition.hip),
+dispatched through bigslice_amd.kernels.
+"""
+
+from __future__ import annotations
+
+from typing import Sequence
+
+import numpy as np
+import torch
+
+from .schema import OBJECT
+
+_C1 = np.uint32(0xCC9E2D51)
+_C2 = np.uint32(0x1B873593)
+
+
+def _rotl32(x: np.ndarray, r: int) -> np.ndarray:
+    return (x << np.uint32(r)) | (x >> np.uint32(32 - r))
+
+
+def _mix_block(h: np.ndarray, k: np.ndarray) -> np.ndarray:
+    k = k * _C1
+    k = _rotl32(k, 15)
+    k = k * _C2
+    h = h ^ k
+    h = _rotl32(h, 13)
+    return h * np.uint32(5) + np.uint32(0xE6546B64)
+
+
+def _fmix(h: np.ndarray, total_len: int) -> np.ndarray:
+    h = h ^ np.uint32(total_len)
+    h = h ^ (h >> np.uint32(16))
+    h = h * np.uint32(0x85EBCA6B)
+    h = h ^ (h >> np.uint32(13))
+    h = h * np.uint32(0xC2B2AE35)
+    h = h ^ (h >> np.uint32(16))
+    return h
+
+
+def murmur3_u32(vals: np.ndarray, seed: int) -> np.ndarray:
+    """murmur3_x86_32 of each 4-byte LE value (reference hash32)."""
+    h = np.full(vals.shape, np.uint32(seed), dtype=np.uint32)
+    h = _mix_block(h, vals.astype(np.uint32, copy=False))
+    return _fmix(h, 4)
+
+
+def murmur3_u64(vals: np.ndarray, seed: int) -> np.ndarray:
+    """murmur3_x86_32 of each 8-byte LE value (reference hash64)."""
+    v = vals.astype(np.uint64, copy=False)
+    lo = (v & np.uint64(0xFFFFFFFF)).astype(np.uint32)
+    hi = (v >> np.uint64(32)).astype(np.uint32)
+    h = np.full(v.shape, np.uint32(seed), dtype=np.uint32)
+    h = _mix_block(h, lo)
+    h = _mix_block(h, hi)
+    return _fmix(h, 8)
+
+
+def murmur3_bytes(data: bytes, seed: int) -> int:
+    """Scalar murmur3_x86_32 over arbitrary bytes (strings)."""
+    with np.errstate(over="ignore"):
+        h = np.uint32(seed)
+        n = len(data)
+        nblocks = n // 4
+        if nblocks:
+            blocks = np.frombuffer(data[: nblocks * 4], dtype="<u4")
+            for k in blocks:
+                h = _mix_block(h, np.uint32(k))
+        tail = data[nblocks * 4:]
+        k1 = np.uint32(0)
+        if len(tail) >= 3:
+            k1 ^= np.uint32(tail[2]) << np.uint32(16)
+        if len(tail) >= 2:
+            k1 ^= np.uint32(tail[1]) << np.uint32(8)
+        if len(tail) >= 1:
+            k1 ^= np.uint32(tail[0])
+            k1 = k1 * _C1
+            k1 = _rotl32(k1, 15)
+            k1 = k1 * _C2
+            h = h ^ k1
+        return int(_fmix(h, n))
+
+
+def _hash_host_column(col, seed: int) -> np.ndarray:
+    """Hash one host column -> uint32 numpy array."""
+    with np.errstate(over="ignore"):
+        if isinstance(col, torch.Tensor):
+            dt = col.dtype
+            if dt == torch.bool:
+                return np.uint32(seed) + col.numpy().astype(np.uint32)
+            a = col.numpy()
+            if dt in (torch.int64, torch.uint64, torch.float64):
+                if dt == torch.float64:
+                    a = a.view(np.uint64)
+                return murmur3_u64(a.astype(np.int64).view(np.uint64)
+                                   if a.dtype != np.uint64 else a, seed)
+            if dt == torch.float32:
+                return murmur3_u32(a.view(np.uint32), seed)
+            # narrower ints: Go converts via uint32(v), sign-extending
+            # signed types (two's complement reinterpretation).
+            return murmur3_u32(a.astype(np.int64).astype(np.uint32)
+                               if a.dtype.kind == "i"
+                               else a.astype(np.uint32), seed)
+        # object column (strings / arbitrary python values)
+        out = np.empty(len(col), dtype=np.uint32)
+        for i, v in enumerate(col):
+            if isinstance(v, str):
+                out[i] = murmur3_bytes(v.encode("utf-8"), seed)
+            elif isinstance(v, bytes):
+                out[i] = murmur3_bytes(v, seed)
+            elif isinstance(v, bool):
+                out[i] = np.uint32(seed) + np.uint32(v)
+            elif isinstance(v, int):
+                out[i] = murmur3_u64(np.array([v], dtype=np.int64)
+                                     .view(np.uint64), seed)[0]
+            elif isinstance(v, float):
+                out[i] = murmur3_u64(np.array([v], dtype=np.float64)
+                                     .view(np.uint64), seed)[0]
+            elif isinstance(v, tuple):
+                # hash tuples by hashing the concatenated member hashes
+                h = np.uint32(seed)
+                for m in v:
+                    hv = _hash_host_column([m], seed)[0]
+                    h = _mix_block(h, np.uint32(hv))
+                out[i] = _fmix(h, 4 * len(v))
+            else:
+                raise TypeError(f"unhashable object column value {type(v)}")
+        return out
+
+
+def hash_columns(cols: Sequence, seed: int = 0) -> torch.Tensor:
+    """XOR-combined 32-bit hash of key columns (reference
+    frame.HashWithSeed).  Device tensors dispatch to the HIP kernel."""
+    if not cols:
+        raise ValueError("hash of zero key columns")
+    first = cols[0]
+    if isinstance(first, torch.Tensor) and first.is_cuda:
+        from . import kernels
+        return kernels.hash_columns_device(list(cols), seed)
+    h = None
+    for c in cols:
+        hc = _hash_host_column(c, seed)
+        h = hc if h is None else (h ^ hc)
+    return torch.from_numpy(h.astype(np.int64))  # non-negative values

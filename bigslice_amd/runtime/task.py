@@ -1,0 +1,133 @@
+"""Tasks: the schedulable unit (one shard of one fused pipeline phase).
+
+Role-parity: exec/task.go — Task{Name, Do, Deps, Partitioner, NumPartition,
+Combiner, Group, state machine INIT->WAITING->RUNNING->OK/ERR/LOST with
+broadcast waiters (task.go:41-86, :325-418)}.
+"""
+
+from __future__ import annotations
+
+import enum
+import threading
+from typing import Callable, List, Optional, Sequence
+
+
+class TaskState(enum.IntEnum):
+    INIT = 0
+    WAITING = 1
+    RUNNING = 2
+    OK = 3
+    ERR = 4
+    LOST = 5
+
+
+class TaskDep:
+    """Dependency on a phase of producer tasks (task.go:91-128).
+
+    head_tasks: the tasks of the producing phase (all shards).
+    partition: which partition of the producers' output this task reads.
+    expand: hand each producer stream to the consumer unmerged.
+    combiner: the consumer's combiner (producers pre-combine partitions).
+    """
+
+    __slots__ = ("head_tasks", "partition", "expand", "combiner")
+
+    def __init__(self, head_tasks: Sequence["Task"], partition: int,
+                 expand: bool = False, combiner=None):
+        self.head_tasks = list(head_tasks)
+        self.partition = partition
+        self.expand = expand
+        self.combiner = combiner
+
+
+class Task:
+    """One shard of a compiled phase."""
+
+    def __init__(self, name: str, invocation_index: int,
+                 do: Callable, deps: List[TaskDep],
+                 num_partitions: int = 1, partitioner=None,
+                 combiner=None, group: Optional[List["Task"]] = None,
+                 num_out_columns: int = 1, pragma=None,
+                 schema=None, shard: int = 0, num_shards: int = 1):
+        self.name = name
+        self.invocation_index = invocation_index
+        self.do = do  # do(dep_readers, ctx) -> sliceio.Reader
+        self.deps = deps
+        self.num_partitions = num_partitions
+        self.partitioner = partitioner
+        self.combiner = combiner
+        self.group: List[Task] = group if group is not None else [self]
+        self.num_out_columns = num_out_columns
+        self.pragma = pragma
+        self.schema = schema
+        self.shard = shard
+        self.num_shards = num_shards
+
+        self._state = TaskState.INIT
+        self._err: Optional[BaseException] = None
+        self._cond = threading.Condition()
+        self.consecutive_lost = 0
+
+    # -- state machine ----------------------------------------------------
+
+    @property
+    def state(self) -> TaskState:
+        with self._cond:
+            return self._state
+
+    @property
+    def error(self) -> Optional[BaseException]:
+        with self._cond:
+            return self._err
+
+    def set_state(self, s: TaskState, err: BaseException = None) -> None:
+        with self._cond:
+            self._state = s
+            if s == TaskState.ERR:
+                self._err = err
+            if s == TaskState.OK:
+                self.consecutive_lost = 0
+            elif s == TaskState.LOST:
+                self.consecutive_lost += 1
+            self._cond.notify_all()
+
+    def wait_state(self, min_state: TaskState,
+                   timeout: Optional[float] = None) -> TaskState:
+        """Block until state >= min_state (task.go WaitState)."""
+        with self._cond:
+            self._cond.wait_for(lambda: self._state >= min_state,
+                                timeout=timeout)
+            return self._state
+
+    def __repr__(self):
+        return f"Task({self.name}, {self.state.name})"
+
+
+def phase_of(task: Task) -> List[Task]:
+    """All sibling tasks of the task's phase (task.go Group semantics)."""
+    return task.group
+
+
+def all_deps_ok(task: Task) -> bool:
+    return all(t.state == TaskState.OK
+               for dep in task.deps for t in dep.head_tasks)
+
+
+def graph_string(roots: Sequence[Task]) -> str:
+    """Debug rendering of the task graph (task.go GraphString), used by
+    compiler golden tests."""
+    lines: List[str] = []
+    seen = set()
+
+    def visit(t: Task, depth: int):
+        lines.append("  " * depth + t.name)
+        if t.name in seen:
+            return
+        seen.add(t.name)
+        for dep in t.deps:
+            for h in dep.head_tasks:
+                visit(h, depth + 1)
+
+    for r in roots:
+        visit(r, 0)
+    return "\n".join(lines)
