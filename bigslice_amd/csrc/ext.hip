@@ -1,0 +1,285 @@
+// bigslice_amd._C — torch extension binding the CDNA4 kernels:
+//   hash_columns      (K3 murmur3 row hash)
+//   hash_partition /  (K4 fused hash+histogram+scatter partitioner)
+//   scatter_by_partition
+//   groupby           (K9 hash-aggregate)
+//   radix_argsort     (K6 device radix sort)
+//
+// Single translation unit: the kernel files are included directly.
+
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+#include "hash_partition.hip"
+#include "groupby.hip"
+#include "sort.hip"
+
+namespace {
+
+#define HIP_CHECK(expr)                                              \
+  do {                                                               \
+    hipError_t _e = (expr);                                          \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ",                     \
+                hipGetErrorString(_e));                              \
+  } while (0)
+
+int32_t dtype_code(const torch::Tensor& t) {
+  switch (t.scalar_type()) {
+    case torch::kInt8: return DT_I8;
+    case torch::kUInt8: return DT_U8;
+    case torch::kInt16: return DT_I16;
+    case torch::kInt32: return DT_I32;
+    case torch::kInt64: return DT_I64;
+    case torch::kFloat32: return DT_F32;
+    case torch::kFloat64: return DT_F64;
+    case torch::kBool: return DT_BOOL;
+    case torch::kUInt32: return DT_U32;
+    case torch::kUInt64: return DT_U64;
+    default:
+      TORCH_CHECK(false, "unsupported column dtype ", t.scalar_type());
+  }
+}
+
+KeyCols make_key_cols(const std::vector<torch::Tensor>& cols) {
+  TORCH_CHECK(!cols.empty() && (int)cols.size() <= MAX_KEY_COLS,
+              "1..", MAX_KEY_COLS, " key columns supported");
+  KeyCols k;
+  k.n = (int)cols.size();
+  for (size_t i = 0; i < cols.size(); ++i) {
+    TORCH_CHECK(cols[i].is_cuda() && cols[i].is_contiguous(),
+                "key columns must be contiguous device tensors");
+    k.cols[i] = {cols[i].data_ptr(), dtype_code(cols[i])};
+  }
+  return k;
+}
+
+hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+constexpr int64_t ROWS_PER_BLOCK = 16384;
+constexpr int32_t MAX_LDS_PARTS = 4096;
+
+// ---------------------------------------------------------------- hash
+
+torch::Tensor hash_columns(std::vector<torch::Tensor> cols, int64_t seed) {
+  KeyCols k = make_key_cols(cols);
+  int64_t n = cols[0].size(0);
+  auto out = torch::empty({n}, cols[0].options().dtype(torch::kUInt32));
+  if (n == 0) return out;
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 65535);
+  hipLaunchKernelGGL(k_hash_rows, dim3(blocks), dim3(THREADS), 0,
+                     current_stream(), k, n, (uint32_t)seed,
+                     out.data_ptr<uint32_t>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
+// ----------------------------------------------------------- partition
+
+std::tuple<std::vector<torch::Tensor>, torch::Tensor> partition_common(
+    const std::vector<torch::Tensor>& cols, torch::Tensor pids,
+    torch::Tensor block_hist, int64_t n, int64_t nparts, int64_t nblocks) {
+  // Destination offsets: per-partition base + this block's running sum
+  // (tiny tensors; stays on device).
+  auto hist = block_hist.view({nblocks, nparts}).to(torch::kInt64);
+  auto part_counts = hist.sum(0);
+  auto part_start = part_counts.cumsum(0) - part_counts;
+  auto within = hist.cumsum(0) - hist;
+  auto block_off = (part_start.unsqueeze(0) + within).contiguous();
+
+  ScatterCols sc;
+  sc.n = (int)cols.size();
+  TORCH_CHECK(sc.n <= MAX_COLS, "at most ", MAX_COLS, " columns");
+  std::vector<torch::Tensor> dst;
+  for (int c = 0; c < sc.n; ++c) {
+    TORCH_CHECK(cols[c].is_cuda() && cols[c].is_contiguous(),
+                "columns must be contiguous device tensors");
+    dst.push_back(torch::empty_like(cols[c]));
+    sc.src[c] = {cols[c].data_ptr(), dtype_code(cols[c])};
+    sc.dst[c] = {dst[c].data_ptr(), dtype_code(cols[c])};
+  }
+  size_t lds = (size_t)nparts * sizeof(uint32_t);
+  hipLaunchKernelGGL(k_part_scatter, dim3((uint32_t)nblocks), dim3(THREADS),
+                     lds, current_stream(), sc,
+                     pids.data_ptr<int32_t>(), n, (int32_t)nparts,
+                     ROWS_PER_BLOCK, block_off.data_ptr<int64_t>());
+  HIP_CHECK(hipGetLastError());
+  return {dst, part_counts.cpu()};
+}
+
+std::tuple<std::vector<torch::Tensor>, torch::Tensor> hash_partition(
+    std::vector<torch::Tensor> cols, std::vector<torch::Tensor> key_cols,
+    int64_t nparts, int64_t seed) {
+  TORCH_CHECK(nparts >= 1 && nparts <= MAX_LDS_PARTS,
+              "nparts must be in [1, ", MAX_LDS_PARTS, "]");
+  KeyCols k = make_key_cols(key_cols);
+  int64_t n = cols.at(0).size(0);
+  int64_t nblocks = (n + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+  auto pids = torch::empty({n}, cols[0].options().dtype(torch::kInt32));
+  auto block_hist = torch::empty({nblocks * nparts},
+                                 cols[0].options().dtype(torch::kUInt32));
+  size_t lds = (size_t)nparts * sizeof(uint32_t);
+  hipLaunchKernelGGL(k_part_hist, dim3((uint32_t)nblocks), dim3(THREADS),
+                     lds, current_stream(), k, n, (int32_t)nparts,
+                     (uint32_t)seed, ROWS_PER_BLOCK,
+                     pids.data_ptr<int32_t>(),
+                     block_hist.data_ptr<uint32_t>());
+  HIP_CHECK(hipGetLastError());
+  return partition_common(cols, pids, block_hist, n, nparts, nblocks);
+}
+
+std::tuple<std::vector<torch::Tensor>, torch::Tensor> scatter_by_partition(
+    std::vector<torch::Tensor> cols, torch::Tensor pids, int64_t nparts) {
+  TORCH_CHECK(nparts >= 1 && nparts <= MAX_LDS_PARTS);
+  TORCH_CHECK(pids.scalar_type() == torch::kInt32 && pids.is_cuda());
+  pids = pids.contiguous();
+  int64_t n = cols.at(0).size(0);
+  int64_t nblocks = (n + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+  auto block_hist = torch::empty({nblocks * nparts},
+                                 cols[0].options().dtype(torch::kUInt32));
+  size_t lds = (size_t)nparts * sizeof(uint32_t);
+  hipLaunchKernelGGL(k_pids_hist, dim3((uint32_t)nblocks), dim3(THREADS),
+                     lds, current_stream(), pids.data_ptr<int32_t>(), n,
+                     (int32_t)nparts, ROWS_PER_BLOCK,
+                     block_hist.data_ptr<uint32_t>());
+  HIP_CHECK(hipGetLastError());
+  return partition_common(cols, pids, block_hist, n, nparts, nblocks);
+}
+
+// ------------------------------------------------------------- groupby
+
+torch::Tensor agg_identity(torch::Tensor like, int32_t agg, int64_t size) {
+  auto opt = like.options();
+  switch (agg) {
+    case AGG_SUM: return torch::zeros({size}, opt);
+    case AGG_PROD: return torch::ones({size}, opt);
+    case AGG_MIN:
+      if (like.is_floating_point())
+        return torch::full({size}, std::numeric_limits<double>::infinity(),
+                           opt);
+      return torch::full({size},
+                         like.scalar_type() == torch::kInt64
+                             ? std::numeric_limits<int64_t>::max()
+                             : (int64_t)std::numeric_limits<int32_t>::max(),
+                         opt);
+    default:  // AGG_MAX
+      if (like.is_floating_point())
+        return torch::full({size}, -std::numeric_limits<double>::infinity(),
+                           opt);
+      return torch::full({size},
+                         like.scalar_type() == torch::kInt64
+                             ? std::numeric_limits<int64_t>::min()
+                             : (int64_t)std::numeric_limits<int32_t>::min(),
+                         opt);
+  }
+}
+
+std::vector<torch::Tensor> groupby(torch::Tensor keys,
+                                   std::vector<torch::Tensor> vals,
+                                   std::vector<int64_t> aggs) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64,
+              "groupby keys must be int64 device tensors");
+  TORCH_CHECK(vals.size() == aggs.size());
+  TORCH_CHECK((int)vals.size() <= MAX_COLS);
+  keys = keys.contiguous();
+  int64_t n = keys.size(0);
+
+  int64_t cap = 1024;
+  while (cap < 2 * n && cap < (1ll << 26)) cap <<= 1;
+
+  const uint32_t seed = 0x9acb0442u;  // reference combiner hashSeed
+  for (;;) {
+    auto tkeys = torch::full({cap + 1}, (int64_t)GB_SENTINEL,
+                             keys.options());
+    ValCols vc;
+    vc.n = (int)vals.size();
+    std::vector<torch::Tensor> tabs;
+    for (size_t c = 0; c < vals.size(); ++c) {
+      TORCH_CHECK(vals[c].is_cuda() && vals[c].is_contiguous());
+      tabs.push_back(agg_identity(vals[c], (int32_t)aggs[c], cap + 1));
+      vc.src[c] = {vals[c].data_ptr(), dtype_code(vals[c])};
+      vc.tab[c] = {tabs[c].data_ptr(), dtype_code(vals[c])};
+      vc.agg[c] = (int32_t)aggs[c];
+    }
+    auto flags = torch::zeros({2}, keys.options().dtype(torch::kInt32));
+    int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
+    if (n > 0) {
+      hipLaunchKernelGGL(k_groupby_insert, dim3(blocks), dim3(THREADS), 0,
+                         current_stream(), keys.data_ptr<int64_t>(), n, vc,
+                         tkeys.data_ptr<int64_t>(), cap, seed,
+                         flags.data_ptr<int32_t>(),
+                         flags.data_ptr<int32_t>() + 1);
+      HIP_CHECK(hipGetLastError());
+    }
+    auto flags_h = flags.cpu();
+    if (flags_h[1].item<int>() != 0) {  // overflow: grow x2 and retry
+      TORCH_CHECK(cap < (1ll << 30), "groupby table overflow at cap ", cap);
+      cap <<= 1;
+      continue;
+    }
+    bool sentinel_seen = flags_h[0].item<int>() != 0;
+    auto body_keys = tkeys.narrow(0, 0, cap);
+    auto mask = body_keys.ne((int64_t)GB_SENTINEL);
+    std::vector<torch::Tensor> out;
+    auto out_keys = body_keys.masked_select(mask);
+    if (sentinel_seen) {
+      out_keys = torch::cat(
+          {out_keys, torch::full({1}, (int64_t)GB_SENTINEL, keys.options())});
+    }
+    out.push_back(out_keys);
+    for (size_t c = 0; c < vals.size(); ++c) {
+      auto v = tabs[c].narrow(0, 0, cap).masked_select(mask);
+      if (sentinel_seen)
+        v = torch::cat({v, tabs[c].narrow(0, cap, 1)});
+      out.push_back(v);
+    }
+    return out;
+  }
+}
+
+// ---------------------------------------------------------------- sort
+
+torch::Tensor radix_argsort(torch::Tensor keys) {
+  TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
+  int64_t n = keys.size(0);
+  auto perm_in = torch::arange(n, keys.options().dtype(torch::kInt64));
+  auto perm_out = torch::empty_like(perm_in);
+  auto keys_out = torch::empty_like(keys);
+  auto run = [&](auto fn) {
+    size_t temp_bytes = 0;
+    fn(keys.data_ptr(), keys_out.data_ptr(),
+       perm_in.data_ptr<int64_t>(), perm_out.data_ptr<int64_t>(), n,
+       nullptr, temp_bytes, current_stream());
+    auto temp = torch::empty({(int64_t)temp_bytes},
+                             keys.options().dtype(torch::kUInt8));
+    fn(keys.data_ptr(), keys_out.data_ptr(),
+       perm_in.data_ptr<int64_t>(), perm_out.data_ptr<int64_t>(), n,
+       temp.data_ptr(), temp_bytes, current_stream());
+  };
+  switch (keys.scalar_type()) {
+    case torch::kInt64: run(radix_sort_pairs_int64_t); break;
+    case torch::kInt32: run(radix_sort_pairs_int32_t); break;
+    case torch::kFloat32: run(radix_sort_pairs_float); break;
+    case torch::kFloat64: run(radix_sort_pairs_double); break;
+    default:
+      TORCH_CHECK(false, "radix_argsort: unsupported dtype");
+  }
+  return perm_out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("hash_columns", &hash_columns, "murmur3 row hash (K3)");
+  m.def("hash_partition", &hash_partition,
+        "fused hash+histogram+scatter partitioner (K4)");
+  m.def("scatter_by_partition", &scatter_by_partition,
+        "scatter rows by precomputed partition ids");
+  m.def("groupby", &groupby, "hash-aggregate group-by (K9)");
+  m.def("radix_argsort", &radix_argsort, "device radix argsort (K6)");
+}

@@ -1,0 +1,122 @@
+// K9: hash-aggregate group-by over device-resident columns.
+//
+// Role-parity with the reference's combiningFrame (exec/combiner.go:
+// open-addressing, power-of-two capacity, grow x2 on pressure, seeded with
+// hashSeed 0x9acb0442 so a previous partitioning step does not strip hash
+// entropy).  The CDNA4 redesign: one pass of global-memory atomics into an
+// open-addressing table in HBM/LLC; on table overflow the host doubles
+// capacity and retries (the reference's rehash-x2, done batch-wise).
+// Slot `cap` (one extra) is reserved for the sentinel key itself.
+
+#include <hip/hip_runtime.h>
+
+#include "columns.h"
+
+#define THREADS 256
+
+// Sentinel marking an empty slot; rows whose key equals the sentinel are
+// accumulated in the reserved extra slot (index == capacity).
+#define GB_SENTINEL 0x8000000000000000LL
+
+enum AggCode : int32_t { AGG_SUM = 0, AGG_MIN = 1, AGG_MAX = 2, AGG_PROD = 3 };
+
+struct ValCols {
+  ColDesc src[MAX_COLS];
+  MutColDesc tab[MAX_COLS];
+  int32_t agg[MAX_COLS];
+  int n;
+};
+
+template <typename T>
+__device__ __forceinline__ void atomic_cas_combine(T* addr, T v, int agg);
+
+template <typename T, typename U>
+__device__ __forceinline__ void cas_loop(T* addr, T v, int agg) {
+  U* a = (U*)addr;
+  U old = *a, assumed;
+  do {
+    assumed = old;
+    T cur = __builtin_bit_cast(T, assumed);
+    T nv;
+    switch (agg) {
+      case AGG_MIN: nv = v < cur ? v : cur; break;
+      case AGG_MAX: nv = v > cur ? v : cur; break;
+      default: nv = cur * v; break;
+    }
+    if (nv == cur) return;
+    old = atomicCAS(a, assumed, __builtin_bit_cast(U, nv));
+  } while (old != assumed);
+}
+
+__device__ __forceinline__ void accum(const MutColDesc& t, int64_t slot,
+                                      const ColDesc& s, int64_t i,
+                                      int agg) {
+  switch (t.code) {
+    case DT_I32: {
+      int32_t v = ((const int32_t*)s.ptr)[i];
+      int32_t* a = (int32_t*)t.ptr + slot;
+      if (agg == AGG_SUM) atomicAdd((int*)a, (int)v);
+      else if (agg == AGG_MIN) atomicMin((int*)a, (int)v);
+      else if (agg == AGG_MAX) atomicMax((int*)a, (int)v);
+      else cas_loop<int32_t, unsigned int>(a, v, agg);
+      break;
+    }
+    case DT_I64: {
+      int64_t v = ((const int64_t*)s.ptr)[i];
+      int64_t* a = (int64_t*)t.ptr + slot;
+      if (agg == AGG_SUM)
+        atomicAdd((unsigned long long*)a, (unsigned long long)v);
+      else
+        cas_loop<int64_t, unsigned long long>(a, v, agg);
+      break;
+    }
+    case DT_F32: {
+      float v = ((const float*)s.ptr)[i];
+      float* a = (float*)t.ptr + slot;
+      if (agg == AGG_SUM) atomicAdd(a, v);
+      else cas_loop<float, unsigned int>(a, v, agg);
+      break;
+    }
+    case DT_F64: {
+      double v = ((const double*)s.ptr)[i];
+      double* a = (double*)t.ptr + slot;
+      if (agg == AGG_SUM) atomicAdd(a, v);
+      else cas_loop<double, unsigned long long>(a, v, agg);
+      break;
+    }
+  }
+}
+
+extern "C" __global__ void k_groupby_insert(
+    const int64_t* keys, int64_t n, ValCols vals, int64_t* tkeys,
+    int64_t cap, uint32_t seed, int32_t* sentinel_seen, int32_t* overflow) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t mask = (uint64_t)cap - 1;
+  for (; i < n; i += stride) {
+    int64_t k = keys[i];
+    int64_t slot;
+    if (k == GB_SENTINEL) {
+      // reserved extra slot
+      if (atomicOr(sentinel_seen, 1) == 0) {}
+      slot = cap;
+    } else {
+      uint64_t h = mm3_u64((uint64_t)k, seed) & mask;
+      int64_t probes = 0;
+      for (;;) {
+        long long prev = atomicCAS((unsigned long long*)&tkeys[h],
+                                   (unsigned long long)GB_SENTINEL,
+                                   (unsigned long long)k);
+        if (prev == GB_SENTINEL || prev == k) break;
+        h = (h + 1) & mask;
+        if (++probes >= cap) {
+          atomicOr(overflow, 1);
+          return;
+        }
+      }
+      slot = (int64_t)h;
+    }
+    for (int c = 0; c < vals.n; ++c)
+      accum(vals.tab[c], slot, vals.src[c], i, vals.agg[c]);
+  }
+}

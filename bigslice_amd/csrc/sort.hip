@@ -1,0 +1,37 @@
+// K6: device radix sort (argsort) via rocPRIM headers compiled into this
+// extension (no external library dependency).  Replaces the reference's
+// comparison sort of spill chunks (sortio/sort.go:52) with an LSD radix
+// sort over whole device batches; the k-way merge of sorted runs (K7/K8)
+// consumes its output.
+
+#include <hip/hip_runtime.h>
+#include <rocprim/device/device_radix_sort.hpp>
+
+#include <stdexcept>
+
+template <typename K>
+static void radix_sort_pairs_impl(const K* keys_in, K* keys_out,
+                                  const int64_t* vals_in, int64_t* vals_out,
+                                  int64_t n, void* temp, size_t& temp_bytes,
+                                  hipStream_t stream) {
+  hipError_t err = rocprim::radix_sort_pairs(
+      temp, temp_bytes, keys_in, keys_out, vals_in, vals_out, (size_t)n, 0,
+      sizeof(K) * 8, stream);
+  if (err != hipSuccess)
+    throw std::runtime_error(std::string("rocprim::radix_sort_pairs: ") +
+                             hipGetErrorString(err));
+}
+
+// Explicit type dispatch used by ext.hip.
+#define INSTANTIATE_SORT(K)                                               \
+  void radix_sort_pairs_##K(const void* ki, void* ko, const int64_t* vi,  \
+                            int64_t* vo, int64_t n, void* temp,           \
+                            size_t& temp_bytes, hipStream_t s) {          \
+    radix_sort_pairs_impl<K>((const K*)ki, (K*)ko, vi, vo, n, temp,       \
+                             temp_bytes, s);                              \
+  }
+
+INSTANTIATE_SORT(int64_t)
+INSTANTIATE_SORT(int32_t)
+INSTANTIATE_SORT(float)
+INSTANTIATE_SORT(double)

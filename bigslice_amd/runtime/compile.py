@@ -90,23 +90,24 @@ class Compiler:
     def compile(self, slice_: Slice) -> List[Task]:
         """Compile the root slice; returns its tasks (one per shard)."""
         return self._compile(slice_, num_partitions=1, partitioner=None,
-                             combiner=None)
+                             combiner=None, shuffle_out=False)
 
     def _compile(self, slice_: Slice, num_partitions: int,
-                 partitioner, combiner) -> List[Task]:
+                 partitioner, combiner, shuffle_out: bool = False
+                 ) -> List[Task]:
         from .session import Result
         target = unwrap(slice_)
         if isinstance(target, Result):
             return self._reuse_result(target, num_partitions, partitioner,
-                                      combiner)
-        key = (id(slice_), num_partitions)
+                                      combiner, shuffle_out)
+        key = (id(slice_), num_partitions, shuffle_out)
         if key in self.memo:
             return self.memo[key]
 
         chain = pipeline_slices(slice_)
         if not chain:  # slice unwraps straight to a Result
             return self._reuse_result(target, num_partitions, partitioner,
-                                      combiner)
+                                      combiner, shuffle_out)
         bottom = chain[-1]
         nshard = chain[0].num_shards
 
@@ -119,9 +120,10 @@ class Compiler:
                 dtasks = self._compile(
                     dep.slice, num_partitions=bottom.num_shards,
                     partitioner=dep.partitioner,
-                    combiner=bottom.combiner)
+                    combiner=bottom.combiner, shuffle_out=True)
             else:
-                dtasks = self._compile(dep.slice, 1, None, None)
+                dtasks = self._compile(dep.slice, 1, None, None,
+                                       shuffle_out=False)
                 if dep.slice.num_shards != bottom.num_shards:
                     raise ValueError(
                         f"non-shuffle dep shard mismatch: "
@@ -167,6 +169,7 @@ class Compiler:
                 num_partitions=num_partitions,
                 partitioner=partitioner,
                 combiner=combiner,
+                shuffle_out=shuffle_out,
                 group=group,
                 num_out_columns=chain[0].schema.num_columns,
                 pragma=chain[0].pragma,
@@ -178,14 +181,15 @@ class Compiler:
         return tasks
 
     def _reuse_result(self, result, num_partitions: int, partitioner,
-                      combiner) -> List[Task]:
+                      combiner, shuffle_out: bool = False) -> List[Task]:
         """Reuse a prior invocation's tasks (exec/compile.go:226-261),
         inserting pass-through _shuffle tasks when the consumer needs a
         different partitioning."""
         prev = result.tasks
-        needs_shuffle = (num_partitions > 1 and any(
+        needs_shuffle = (shuffle_out and any(
             t.num_partitions != num_partitions or
-            t.combiner is not combiner for t in prev))
+            t.combiner is not combiner or not t.shuffle_out
+            for t in prev))
         if not needs_shuffle:
             return prev
         key = (id(result), num_partitions)
@@ -207,6 +211,7 @@ class Compiler:
                 num_partitions=num_partitions,
                 partitioner=partitioner,
                 combiner=combiner,
+                shuffle_out=True,
                 group=group,
                 num_out_columns=prev[shard].num_out_columns,
                 schema=prev[shard].schema,

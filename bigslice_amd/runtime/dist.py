@@ -1,0 +1,187 @@
+"""Distributed executor: SPMD, one process per GPU, RCCL over xGMI.
+
+Role-parity with the reference's bigmachine executor (exec/bigmachine.go),
+redesigned for the MI355X execution model: instead of a driver pushing
+tasks to workers over RPC and workers pulling shuffle data through a
+storage layer, every rank compiles the identical task graph (the CompileEnv
+/ Func-registry discipline, verified by digest at start) and executes it
+phase-synchronously: each rank runs its own shards of a phase, partitions
+output with the fused K4 kernel (pre-combining when the consumer declares a
+combiner), and the phase boundary is ONE RCCL all-to-allv over xGMI.
+
+Shard placement: shard s of every phase runs on rank s % world, and
+partition p of any shuffle output is owned by rank p % world — placement
+is static and known to all ranks, so there is no location metadata plane.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+from ..ops.slice_base import TaskContext
+from ..parallel.comm import Comm
+from ..sliceio import MultiReader, Reader
+from .eval import Executor
+from .partition import PartitionWriter
+from .store import MemoryStore
+from .task import Task, TaskState
+
+
+def owner_rank(shard: int, world: int) -> int:
+    return shard % world
+
+
+class DistExecutor(Executor):
+    """Phase-synchronous SPMD executor.  All Session.run / Result.scan
+    calls are collective: every rank must make the same calls in the
+    same order."""
+
+    def __init__(self, comm: Comm):
+        self.comm = comm
+        self.store = MemoryStore()
+        self.device = comm.device
+
+    # The session calls executor.evaluate when present (instead of the
+    # generic task-pull evaluator).
+    def evaluate(self, roots: Sequence[Task]) -> None:
+        phases = self._phase_order(roots)
+        for phase in phases:
+            self._run_phase(phase)
+        self.comm.barrier()
+
+    def _phase_order(self, roots: Sequence[Task]) -> List[List[Task]]:
+        """Topological order of phase groups (deterministic across
+        ranks: graph construction is deterministic)."""
+        order: List[List[Task]] = []
+        seen = set()
+
+        def visit(task: Task):
+            gid = id(task.group[0])
+            if gid in seen:
+                return
+            seen.add(gid)
+            for t in task.group:
+                for dep in t.deps:
+                    for h in dep.head_tasks:
+                        visit(h)
+            order.append(list(task.group))
+
+        for r in roots:
+            visit(r)
+        return order
+
+    def _run_phase(self, phase: List[Task]) -> None:
+        comm = self.comm
+        world = comm.world
+        # Skip phases that already ran (Result reuse across invocations).
+        if all(t.state == TaskState.OK for t in phase):
+            return
+        my_tasks = [t for t in phase
+                    if owner_rank(t.shard, world) == comm.rank]
+        exemplar = phase[0]
+        shuffled = exemplar.num_partitions > 1
+        send: List[List[Tuple[str, int, object]]] = \
+            [[] for _ in range(world)]
+        err: Optional[BaseException] = None
+        try:
+            for t in my_tasks:
+                buckets = self._run_task(t)
+                if buckets is None:
+                    continue
+                for p, frames in enumerate(buckets):
+                    if shuffled:
+                        d = owner_rank(p, world)
+                        for f in frames:
+                            send[d].append((t.name, p, f))
+                        if not frames:
+                            pass
+                    else:
+                        rows = sum(len(f) for f in frames)
+                        self.store.put(t.name, p, frames, rows)
+        except BaseException as e:
+            err = e
+        # Surface errors collectively so every rank raises.
+        errs = comm.all_gather_obj(repr(err) if err else None)
+        if err is not None:
+            for t in phase:
+                t.set_state(TaskState.ERR, err)
+            raise err
+        first = next((e for e in errs if e), None)
+        if first:
+            e = RuntimeError(f"remote rank failed: {first}")
+            for t in phase:
+                t.set_state(TaskState.ERR, e)
+            raise e
+
+        if shuffled:
+            recv = comm.exchange_buckets(send, exemplar.schema)
+            # group received frames by (task, partition)
+            grouped: Dict[Tuple[str, int], List] = {}
+            for (tname, p, f) in recv:
+                grouped.setdefault((tname, p), []).append(f)
+            for (tname, p), frames in grouped.items():
+                if self.device != "cpu":
+                    frames = [f.to(self.device) for f in frames]
+                rows = sum(len(f) for f in frames)
+                self.store.put(tname, p, frames, rows)
+            # mark empty partitions we own so readers don't KeyError
+            for t in phase:
+                for p in range(t.num_partitions):
+                    if owner_rank(p, world) == comm.rank and \
+                            not self.store.has(t.name, p):
+                        self.store.put(t.name, p, [], 0)
+        for t in phase:
+            t.set_state(TaskState.OK)
+
+    def _run_task(self, task: Task):
+        """Run one task; returns per-partition frame lists (or None for
+        terminal tasks)."""
+        task.set_state(TaskState.RUNNING)
+        ctx = TaskContext(device=self.device)
+        dep_readers = []
+        for dep in task.deps:
+            readers = [self.store.open(h.name, dep.partition,
+                                       device=self.device)
+                       for h in dep.head_tasks
+                       if self.store.has(h.name, dep.partition)]
+            if dep.expand:
+                dep_readers.append(readers)
+            else:
+                dep_readers.append(MultiReader(readers))
+        out = task.do(dep_readers, ctx)
+        if task.num_out_columns == 0:
+            for _ in out:
+                pass
+            self.store.put(task.name, 0, [], 0)
+            return None
+        w = PartitionWriter(task.num_partitions, task.partitioner,
+                            task.combiner, task.schema, self.device,
+                            ctx.chunk)
+        for f in out:
+            w.add(f)
+        return w.finish()
+
+    # -- Executor interface (driver-side readback) ------------------------
+
+    def run(self, task: Task) -> None:
+        raise RuntimeError("DistExecutor schedules phases itself")
+
+    def reader(self, task: Task, partition: int) -> Reader:
+        if self.store.has(task.name, partition):
+            return self.store.open(task.name, partition, device="cpu")
+        from ..sliceio import EmptyReader
+        return EmptyReader()
+
+    def discard(self, task: Task) -> None:
+        self.store.discard_task(task.name)
+
+    def gather_result(self, tasks: Sequence[Task], schema):
+        """Collective: gather all root-task outputs to rank 0."""
+        frames = []
+        for t in tasks:
+            if owner_rank(t.shard, self.comm.world) == self.comm.rank:
+                r = self.reader(t, 0)
+                frames.extend(list(r))
+        return self.comm.gather_frames(frames, schema)

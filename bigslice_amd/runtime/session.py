@@ -124,7 +124,14 @@ class Result(Slice):
         return self.session.executor.reader(self.tasks[shard], 0)
 
     def open(self) -> Reader:
-        """Reader over all shards' outputs."""
+        """Reader over all shards' outputs.  With a distributed executor
+        this is a collective call: every rank must call it; rank 0 sees
+        all rows, other ranks see none."""
+        gather = getattr(self.session.executor, "gather_result", None)
+        if gather is not None:
+            frames = gather(self.tasks, self.schema)
+            from ..sliceio import IterReader
+            return IterReader(iter(frames))
         return MultiReader([
             self.session.executor.reader(t, 0) for t in self.tasks])
 
@@ -176,7 +183,11 @@ class Session:
             compiler = Compiler(inv_index, self.env)
             tasks = compiler.compile(slice_)
             self.env.freeze()
-            evaluate(self.executor, tasks)
+            ev = getattr(self.executor, "evaluate", None)
+            if ev is not None:
+                ev(tasks)
+            else:
+                evaluate(self.executor, tasks)
             self.env = CompileEnv()  # fresh cache decisions per run
             return Result(self, slice_, tasks)
         finally:
@@ -192,10 +203,25 @@ class Session:
 
 
 def start(parallelism: int = None, device: str = None,
-          executor: Executor = None) -> Session:
-    """Create a session (exec.Start analog).  Defaults to the local
-    executor on cuda:0 when a GPU is visible, else CPU."""
+          executor: Executor = None, distributed: bool = None) -> Session:
+    """Create a session (exec.Start analog).
+
+    distributed=True (or WORLD_SIZE>1 in the environment) starts the SPMD
+    executor: one process per GPU over RCCL; every rank must call the
+    same Session methods in the same order.  Otherwise the local executor
+    runs on cuda:0 when a GPU is visible, else CPU.
+    """
+    import os
+    if distributed is None:
+        distributed = int(os.environ.get("WORLD_SIZE", "1")) > 1
     if executor is None:
-        from .local import LocalExecutor
-        executor = LocalExecutor(parallelism=parallelism, device=device)
+        if distributed:
+            from ..parallel.comm import init_comm
+            from .dist import DistExecutor
+            comm = init_comm(device=device)
+            comm.check_registry(registry_digest())
+            executor = DistExecutor(comm)
+        else:
+            from .local import LocalExecutor
+            executor = LocalExecutor(parallelism=parallelism, device=device)
     return Session(executor, parallelism)

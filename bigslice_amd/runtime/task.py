@@ -46,7 +46,8 @@ class Task:
     def __init__(self, name: str, invocation_index: int,
                  do: Callable, deps: List[TaskDep],
                  num_partitions: int = 1, partitioner=None,
-                 combiner=None, group: Optional[List["Task"]] = None,
+                 combiner=None, shuffle_out: bool = False,
+                 group: Optional[List["Task"]] = None,
                  num_out_columns: int = 1, pragma=None,
                  schema=None, shard: int = 0, num_shards: int = 1):
         self.name = name
@@ -56,6 +57,7 @@ class Task:
         self.num_partitions = num_partitions
         self.partitioner = partitioner
         self.combiner = combiner
+        self.shuffle_out = shuffle_out
         self.group: List[Task] = group if group is not None else [self]
         self.num_out_columns = num_out_columns
         self.pragma = pragma
