@@ -61,8 +61,16 @@ hipStream_t current_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
-constexpr int64_t ROWS_PER_BLOCK = 16384;
 constexpr int32_t MAX_LDS_PARTS = 4096;
+
+// Rows per block for the partition kernels: target ~4096 workgroups so
+// the grid comfortably fills 256 CUs across 8 XCDs, with a floor that
+// keeps per-block LDS histogram work worthwhile.
+int64_t rows_per_block(int64_t n) {
+  int64_t rpb = (n + 4095) / 4096;
+  if (rpb < 1024) rpb = 1024;
+  return rpb;
+}
 
 // ---------------------------------------------------------------- hash
 
@@ -83,7 +91,8 @@ torch::Tensor hash_columns(std::vector<torch::Tensor> cols, int64_t seed) {
 
 std::tuple<std::vector<torch::Tensor>, torch::Tensor> partition_common(
     const std::vector<torch::Tensor>& cols, torch::Tensor pids,
-    torch::Tensor block_hist, int64_t n, int64_t nparts, int64_t nblocks) {
+    torch::Tensor block_hist, int64_t n, int64_t nparts, int64_t nblocks,
+    int64_t rpb) {
   // Destination offsets: per-partition base + this block's running sum
   // (tiny tensors; stays on device).
   auto hist = block_hist.view({nblocks, nparts}).to(torch::kInt64);
@@ -107,7 +116,7 @@ std::tuple<std::vector<torch::Tensor>, torch::Tensor> partition_common(
   hipLaunchKernelGGL(k_part_scatter, dim3((uint32_t)nblocks), dim3(THREADS),
                      lds, current_stream(), sc,
                      pids.data_ptr<int32_t>(), n, (int32_t)nparts,
-                     ROWS_PER_BLOCK, block_off.data_ptr<int64_t>());
+                     rpb, block_off.data_ptr<int64_t>());
   HIP_CHECK(hipGetLastError());
   return {dst, part_counts.cpu()};
 }
@@ -119,18 +128,19 @@ std::tuple<std::vector<torch::Tensor>, torch::Tensor> hash_partition(
               "nparts must be in [1, ", MAX_LDS_PARTS, "]");
   KeyCols k = make_key_cols(key_cols);
   int64_t n = cols.at(0).size(0);
-  int64_t nblocks = (n + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+  int64_t rpb = rows_per_block(n);
+  int64_t nblocks = (n + rpb - 1) / rpb;
   auto pids = torch::empty({n}, cols[0].options().dtype(torch::kInt32));
   auto block_hist = torch::empty({nblocks * nparts},
                                  cols[0].options().dtype(torch::kUInt32));
   size_t lds = (size_t)nparts * sizeof(uint32_t);
   hipLaunchKernelGGL(k_part_hist, dim3((uint32_t)nblocks), dim3(THREADS),
                      lds, current_stream(), k, n, (int32_t)nparts,
-                     (uint32_t)seed, ROWS_PER_BLOCK,
+                     (uint32_t)seed, rpb,
                      pids.data_ptr<int32_t>(),
                      block_hist.data_ptr<uint32_t>());
   HIP_CHECK(hipGetLastError());
-  return partition_common(cols, pids, block_hist, n, nparts, nblocks);
+  return partition_common(cols, pids, block_hist, n, nparts, nblocks, rpb);
 }
 
 std::tuple<std::vector<torch::Tensor>, torch::Tensor> scatter_by_partition(
@@ -139,16 +149,17 @@ std::tuple<std::vector<torch::Tensor>, torch::Tensor> scatter_by_partition(
   TORCH_CHECK(pids.scalar_type() == torch::kInt32 && pids.is_cuda());
   pids = pids.contiguous();
   int64_t n = cols.at(0).size(0);
-  int64_t nblocks = (n + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+  int64_t rpb = rows_per_block(n);
+  int64_t nblocks = (n + rpb - 1) / rpb;
   auto block_hist = torch::empty({nblocks * nparts},
                                  cols[0].options().dtype(torch::kUInt32));
   size_t lds = (size_t)nparts * sizeof(uint32_t);
   hipLaunchKernelGGL(k_pids_hist, dim3((uint32_t)nblocks), dim3(THREADS),
                      lds, current_stream(), pids.data_ptr<int32_t>(), n,
-                     (int32_t)nparts, ROWS_PER_BLOCK,
+                     (int32_t)nparts, rpb,
                      block_hist.data_ptr<uint32_t>());
   HIP_CHECK(hipGetLastError());
-  return partition_common(cols, pids, block_hist, n, nparts, nblocks);
+  return partition_common(cols, pids, block_hist, n, nparts, nblocks, rpb);
 }
 
 // ------------------------------------------------------------- groupby
@@ -179,67 +190,73 @@ torch::Tensor agg_identity(torch::Tensor like, int32_t agg, int64_t size) {
   }
 }
 
-std::vector<torch::Tensor> groupby(torch::Tensor keys,
-                                   std::vector<torch::Tensor> vals,
-                                   std::vector<int64_t> aggs) {
-  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64,
-              "groupby keys must be int64 device tensors");
-  TORCH_CHECK(vals.size() == aggs.size());
+ValCols make_val_cols(const std::vector<torch::Tensor>& vals,
+                      const std::vector<torch::Tensor>& tabs,
+                      const std::vector<int64_t>& aggs) {
+  ValCols vc;
+  vc.n = (int)vals.size();
+  TORCH_CHECK(vals.size() == tabs.size() && vals.size() == aggs.size());
   TORCH_CHECK((int)vals.size() <= MAX_COLS);
+  for (size_t c = 0; c < vals.size(); ++c) {
+    TORCH_CHECK(vals[c].is_cuda() && vals[c].is_contiguous());
+    TORCH_CHECK(tabs[c].scalar_type() == vals[c].scalar_type());
+    vc.src[c] = {vals[c].data_ptr(), dtype_code(vals[c])};
+    vc.tab[c] = {tabs[c].data_ptr(), dtype_code(tabs[c])};
+    vc.agg[c] = (int32_t)aggs[c];
+  }
+  return vc;
+}
+
+// Streaming insert into a caller-owned table (tkeys: cap+1 slots filled
+// with the sentinel; tabs[c]: cap+1 slots at the agg identity; flags:
+// int32[2] = {sentinel_seen, overflow}).  Multiple insert calls
+// accumulate into the same table; the host checks flags[1] after
+// compaction and regrows on overflow.
+void groupby_insert(torch::Tensor keys, std::vector<torch::Tensor> vals,
+                    std::vector<int64_t> aggs, torch::Tensor tkeys,
+                    std::vector<torch::Tensor> tabs, torch::Tensor flags,
+                    int64_t max_probes) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
   keys = keys.contiguous();
   int64_t n = keys.size(0);
-
-  int64_t cap = 1024;
-  while (cap < 2 * n && cap < (1ll << 26)) cap <<= 1;
-
+  if (n == 0) return;
+  int64_t cap = tkeys.size(0) - 1;
+  TORCH_CHECK(cap > 0 && (cap & (cap - 1)) == 0, "capacity must be 2^k");
+  ValCols vc = make_val_cols(vals, tabs, aggs);
   const uint32_t seed = 0x9acb0442u;  // reference combiner hashSeed
-  for (;;) {
-    auto tkeys = torch::full({cap + 1}, (int64_t)GB_SENTINEL,
-                             keys.options());
-    ValCols vc;
-    vc.n = (int)vals.size();
-    std::vector<torch::Tensor> tabs;
-    for (size_t c = 0; c < vals.size(); ++c) {
-      TORCH_CHECK(vals[c].is_cuda() && vals[c].is_contiguous());
-      tabs.push_back(agg_identity(vals[c], (int32_t)aggs[c], cap + 1));
-      vc.src[c] = {vals[c].data_ptr(), dtype_code(vals[c])};
-      vc.tab[c] = {tabs[c].data_ptr(), dtype_code(vals[c])};
-      vc.agg[c] = (int32_t)aggs[c];
-    }
-    auto flags = torch::zeros({2}, keys.options().dtype(torch::kInt32));
-    int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
-    if (n > 0) {
-      hipLaunchKernelGGL(k_groupby_insert, dim3(blocks), dim3(THREADS), 0,
-                         current_stream(), keys.data_ptr<int64_t>(), n, vc,
-                         tkeys.data_ptr<int64_t>(), cap, seed,
-                         flags.data_ptr<int32_t>(),
-                         flags.data_ptr<int32_t>() + 1);
-      HIP_CHECK(hipGetLastError());
-    }
-    auto flags_h = flags.cpu();
-    if (flags_h[1].item<int>() != 0) {  // overflow: grow x2 and retry
-      TORCH_CHECK(cap < (1ll << 30), "groupby table overflow at cap ", cap);
-      cap <<= 1;
-      continue;
-    }
-    bool sentinel_seen = flags_h[0].item<int>() != 0;
-    auto body_keys = tkeys.narrow(0, 0, cap);
-    auto mask = body_keys.ne((int64_t)GB_SENTINEL);
-    std::vector<torch::Tensor> out;
-    auto out_keys = body_keys.masked_select(mask);
-    if (sentinel_seen) {
-      out_keys = torch::cat(
-          {out_keys, torch::full({1}, (int64_t)GB_SENTINEL, keys.options())});
-    }
-    out.push_back(out_keys);
-    for (size_t c = 0; c < vals.size(); ++c) {
-      auto v = tabs[c].narrow(0, 0, cap).masked_select(mask);
-      if (sentinel_seen)
-        v = torch::cat({v, tabs[c].narrow(0, cap, 1)});
-      out.push_back(v);
-    }
-    return out;
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
+  hipLaunchKernelGGL(k_groupby_insert, dim3(blocks), dim3(THREADS), 0,
+                     current_stream(), keys.data_ptr<int64_t>(), n, vc,
+                     tkeys.data_ptr<int64_t>(), cap, seed,
+                     flags.data_ptr<int32_t>(),
+                     flags.data_ptr<int32_t>() + 1, max_probes);
+  HIP_CHECK(hipGetLastError());
+}
+
+// Compact used slots into freshly-allocated output arrays; returns
+// [out_keys(cap), out_vals...(cap)]; cursor (u64[1], zeroed by caller)
+// receives the used-slot count — the caller narrows after reading it.
+std::vector<torch::Tensor> groupby_compact(torch::Tensor tkeys,
+                                           std::vector<torch::Tensor> tabs,
+                                           torch::Tensor cursor) {
+  int64_t cap = tkeys.size(0) - 1;
+  CompactCols cc;
+  cc.n = (int)tabs.size();
+  std::vector<torch::Tensor> out;
+  auto out_keys = torch::empty({cap}, tkeys.options());
+  out.push_back(out_keys);
+  for (size_t c = 0; c < tabs.size(); ++c) {
+    out.push_back(torch::empty({cap}, tabs[c].options()));
+    cc.tab[c] = {tabs[c].data_ptr(), dtype_code(tabs[c])};
+    cc.out[c] = {out[c + 1].data_ptr(), dtype_code(tabs[c])};
   }
+  int blocks = (int)std::min<int64_t>((cap + THREADS - 1) / THREADS, 16384);
+  hipLaunchKernelGGL(k_groupby_compact, dim3(blocks), dim3(THREADS), 0,
+                     current_stream(), tkeys.data_ptr<int64_t>(), cap, cc,
+                     out_keys.data_ptr<int64_t>(),
+                     (unsigned long long*)cursor.data_ptr());
+  HIP_CHECK(hipGetLastError());
+  return out;
 }
 
 // ---------------------------------------------------------------- sort
@@ -280,6 +297,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused hash+histogram+scatter partitioner (K4)");
   m.def("scatter_by_partition", &scatter_by_partition,
         "scatter rows by precomputed partition ids");
-  m.def("groupby", &groupby, "hash-aggregate group-by (K9)");
+  m.def("groupby_insert", &groupby_insert,
+        "hash-aggregate insert pass (K9)");
+  m.def("groupby_compact", &groupby_compact,
+        "hash-aggregate table compaction (K9)");
+  m.def("agg_identity", &agg_identity, "aggregation identity fill");
   m.def("radix_argsort", &radix_argsort, "device radix argsort (K6)");
 }

@@ -87,9 +87,53 @@ __device__ __forceinline__ void accum(const MutColDesc& t, int64_t slot,
   }
 }
 
+// Compaction: scan the table, appending used slots to the output arrays.
+// Wave-aggregated cursor: one global atomicAdd per wave (64 lanes), not per
+// element — a single hot counter saturates at ~88 atomics/us on this chip.
+struct CompactCols {
+  ColDesc tab[MAX_COLS];
+  MutColDesc out[MAX_COLS];
+  int n;
+};
+
+__device__ __forceinline__ void copy_elt(const MutColDesc& dst, int64_t di,
+                                         const ColDesc& src, int64_t si) {
+  switch (elt_size(src.code)) {
+    case 1: ((uint8_t*)dst.ptr)[di] = ((const uint8_t*)src.ptr)[si]; break;
+    case 2: ((uint16_t*)dst.ptr)[di] = ((const uint16_t*)src.ptr)[si]; break;
+    case 4: ((uint32_t*)dst.ptr)[di] = ((const uint32_t*)src.ptr)[si]; break;
+    default: ((uint64_t*)dst.ptr)[di] = ((const uint64_t*)src.ptr)[si];
+  }
+}
+
+extern "C" __global__ void k_groupby_compact(
+    const int64_t* tkeys, int64_t cap, CompactCols cols, int64_t* out_keys,
+    unsigned long long* cursor) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int lane = threadIdx.x & 63;
+  for (; i - lane < cap; i += stride) {
+    bool used = (i < cap) && (tkeys[i] != GB_SENTINEL);
+    uint64_t mask = __ballot(used);
+    if (mask == 0) continue;
+    unsigned long long base = 0;
+    if (lane == __ffsll((long long)mask) - 1)
+      base = atomicAdd(cursor, (unsigned long long)__popcll(mask));
+    base = __shfl(base, __ffsll((long long)mask) - 1);
+    if (used) {
+      int64_t pos =
+          (int64_t)base + __popcll(mask & ((1ull << lane) - 1));
+      out_keys[pos] = tkeys[i];
+      for (int c = 0; c < cols.n; ++c)
+        copy_elt(cols.out[c], pos, cols.tab[c], i);
+    }
+  }
+}
+
 extern "C" __global__ void k_groupby_insert(
     const int64_t* keys, int64_t n, ValCols vals, int64_t* tkeys,
-    int64_t cap, uint32_t seed, int32_t* sentinel_seen, int32_t* overflow) {
+    int64_t cap, uint32_t seed, int32_t* sentinel_seen, int32_t* overflow,
+    int64_t max_probes) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   uint64_t mask = (uint64_t)cap - 1;
@@ -109,7 +153,9 @@ extern "C" __global__ void k_groupby_insert(
                                    (unsigned long long)k);
         if (prev == GB_SENTINEL || prev == k) break;
         h = (h + 1) & mask;
-        if (++probes >= cap) {
+        // A long probe chain means the table is too loaded: signal the
+        // host to grow x2 and re-insert (reference combiner grow policy).
+        if (++probes >= max_probes) {
           atomicOr(overflow, 1);
           return;
         }

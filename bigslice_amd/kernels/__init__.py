@@ -83,8 +83,11 @@ def partition_frame(frame, num_partitions: int, partitioner):
 
 _AGG_CODES = {"sum": 0, "min": 1, "max": 2, "prod": 3}
 
-_GB_KEY_DTYPES = (torch.int64, torch.int32, torch.uint32, torch.uint64)
+_GB_KEY_DTYPES = (torch.int64,)
 _GB_VAL_DTYPES = (torch.int64, torch.int32, torch.float32, torch.float64)
+
+GB_SENTINEL = -(1 << 63)
+MAX_PROBES = 128
 
 
 def groupby_supported(keys: torch.Tensor, vals: List[torch.Tensor],
@@ -98,12 +101,101 @@ def groupby_supported(keys: torch.Tensor, vals: List[torch.Tensor],
     return all(a in _AGG_CODES for a in aggs)
 
 
+def _next_pow2(n: int) -> int:
+    c = 1024
+    while c < n:
+        c <<= 1
+    return c
+
+
+class GroupTable:
+    """Streaming device hash-aggregate (K9): an open-addressing table in
+    HBM that multiple batches insert into; finish() compacts used slots.
+    On probe-chain overflow the table doubles and re-inserts all retained
+    batches (the reference combiner's grow-x2, exec/combiner.go:47)."""
+
+    def __init__(self, val_dtypes, aggs: List[str], device,
+                 cap_hint: int = None):
+        self.aggs = aggs
+        self.codes = [_AGG_CODES[a] for a in aggs]
+        self.device = device
+        self.val_dtypes = list(val_dtypes)
+        self.cap = None
+        self.cap_hint = cap_hint
+        self.batches: List = []  # retained for overflow re-insert
+        self.rows = 0
+
+    def _alloc(self, cap: int):
+        self.cap = cap
+        dev = self.device
+        self.tkeys = torch.full((cap + 1,), GB_SENTINEL,
+                                dtype=torch.int64, device=dev)
+        self.tabs = [
+            _C.agg_identity(torch.empty(0, dtype=dt, device=dev), code,
+                            cap + 1)
+            for dt, code in zip(self.val_dtypes, self.codes)]
+        self.flags = torch.zeros(2, dtype=torch.int32, device=dev)
+
+    def insert(self, keys: torch.Tensor, vals: List[torch.Tensor]):
+        n = keys.shape[0]
+        if n == 0:
+            return
+        self.rows += n
+        if self.cap is None:
+            from .. import config
+            hint = self.cap_hint or min(
+                2 * n, getattr(config, "GROUPBY_INITIAL_CAP", 1 << 23))
+            self._alloc(_next_pow2(hint))
+        self.batches.append((keys, vals))
+        _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
+                          self.tabs, self.flags, MAX_PROBES)
+
+    def finish(self):
+        """Returns (keys, [vals]) of the aggregated groups."""
+        if self.cap is None:
+            empty = torch.empty(0, dtype=torch.int64, device=self.device)
+            return empty, [torch.empty(0, dtype=dt, device=self.device)
+                           for dt in self.val_dtypes]
+        while True:
+            cursor = torch.zeros(1, dtype=torch.int64, device=self.device)
+            outs = _C.groupby_compact(self.tkeys, self.tabs, cursor)
+            host = torch.cat([cursor,
+                              self.flags.to(torch.int64)]).cpu()
+            nkeys, sentinel_seen, overflow = (int(host[0]), int(host[1]),
+                                              int(host[2]))
+            if overflow:
+                cap = self.cap
+                while cap < 4 * self.rows and cap < (1 << 30):
+                    cap <<= 1
+                if cap == self.cap:
+                    cap <<= 1
+                if cap > (1 << 31):
+                    raise RuntimeError("groupby table overflow")
+                batches = self.batches
+                self._alloc(cap)
+                self.batches = []
+                self.rows = 0
+                for keys, vals in batches:
+                    self.insert(keys, vals)
+                continue
+            keys = outs[0][:nkeys]
+            vals = [o[:nkeys] for o in outs[1:]]
+            if sentinel_seen:
+                keys = torch.cat([keys, torch.full(
+                    (1,), GB_SENTINEL, dtype=torch.int64,
+                    device=self.device)])
+                vals = [torch.cat([v, t[self.cap:self.cap + 1]])
+                        for v, t in zip(vals, self.tabs)]
+            self.batches = []
+            return keys, vals
+
+
 def groupby(keys: torch.Tensor, vals: List[torch.Tensor],
             aggs: List[str]):
-    """Hash-aggregate: returns (unique_keys, [combined values])."""
-    codes = [_AGG_CODES[a] for a in aggs]
-    out = _C.groupby(keys, list(vals), codes)
-    return out[0], list(out[1:])
+    """One-shot hash-aggregate: returns (unique_keys, [combined vals])."""
+    t = GroupTable([v.dtype for v in vals], aggs, keys.device)
+    t.insert(keys, list(vals))
+    return t.finish()
 
 
 # -- sort (K6) -------------------------------------------------------------

@@ -91,12 +91,27 @@ class TensorAggregator:
         self.vals: List[torch.Tensor] = []
         self._pending: List[Frame] = []
         self._pending_rows = 0
+        # Device fast path: stream batches straight into the HIP
+        # GroupTable (no host sync until finish).
+        self._table = None
+        if device.startswith("cuda"):
+            from .. import kernels
+            val_dts = schema.dtypes[schema.prefix:]
+            if kernels.have_extension() and all(
+                    dt in kernels._GB_VAL_DTYPES for dt in val_dts) and \
+                    schema.dtypes[0] in kernels._GB_KEY_DTYPES and \
+                    all(isinstance(a, str) for a in agg.aggs):
+                self._table = kernels.GroupTable(val_dts, agg.aggs, device)
 
     def add(self, frame: Frame) -> None:
         if len(frame) == 0:
             return
         if frame.device != self.device:
             frame = frame.to(self.device)
+        if self._table is not None:
+            self._table.insert(frame.columns[0].contiguous(),
+                               [c.contiguous() for c in frame.columns[1:]])
+            return
         self._pending.append(frame)
         self._pending_rows += len(frame)
         if self._pending_rows >= config.COMBINER_TARGET_KEYS:
@@ -116,7 +131,12 @@ class TensorAggregator:
         self.keys, self.vals = _combine_once(keys, vals, self.agg)
 
     def result_frames(self, chunk: int):
-        self._flush()
+        if self._table is not None:
+            keys, vals = self._table.finish()
+            if keys.shape[0]:
+                self.keys, self.vals = keys, vals
+        else:
+            self._flush()
         if self.keys is None:
             return
         n = self.keys.shape[0]
@@ -127,6 +147,8 @@ class TensorAggregator:
                         self.schema.prefix)
 
     def num_keys(self) -> int:
+        if self._table is not None:
+            return self._table.rows and self._table.finish()[0].shape[0]
         self._flush()
         return 0 if self.keys is None else self.keys.shape[0]
 

@@ -67,18 +67,31 @@ class PartitionWriter:
         self.device = device
         self.chunk = chunk
         self.rows = 0
+        self.agg = None
+        self.aggs = None
+        self.buckets: List[List[Frame]] = \
+            [[] for _ in range(num_partitions)]
         if combiner is not None:
             from ..ops.aggregate import make_aggregator
-            self.aggs = [make_aggregator(schema, combiner, device)
-                         for _ in range(num_partitions)]
-            self.buckets = None
-        else:
-            self.aggs = None
-            self.buckets: List[List[Frame]] = \
-                [[] for _ in range(num_partitions)]
+            if partitioner is None:
+                # Combine-first: the default partitioner is a pure key
+                # function, so pre-combining the whole shard THEN
+                # splitting the (much smaller) combined result is
+                # equivalent to the reference's per-partition combiners
+                # and turns 8 small hash tables + scatters into one big
+                # streaming insert — the GPU-friendly order.
+                self.agg = make_aggregator(schema, combiner, device)
+            else:
+                # Custom partitioners may not be key-pure: keep the
+                # partition-then-combine order.
+                self.aggs = [make_aggregator(schema, combiner, device)
+                             for _ in range(num_partitions)]
 
     def add(self, frame: Frame) -> None:
         self.rows += len(frame)
+        if self.agg is not None:
+            self.agg.add(frame)
+            return
         parts = split_frame(frame, self.num_partitions, self.partitioner)
         for pi, pf in enumerate(parts):
             if pf is None or len(pf) == 0:
@@ -90,6 +103,13 @@ class PartitionWriter:
 
     def finish(self) -> List[List[Frame]]:
         """Per-partition frame lists."""
+        if self.agg is not None:
+            for f in self.agg.result_frames(self.chunk * 4):
+                parts = split_frame(f, self.num_partitions, None)
+                for pi, pf in enumerate(parts):
+                    if pf is not None and len(pf):
+                        self.buckets[pi].append(pf)
+            return self.buckets
         if self.aggs is not None:
             return [list(a.result_frames(self.chunk)) for a in self.aggs]
         return self.buckets

@@ -130,3 +130,26 @@ def test_end_to_end_gpu_matches_cpu():
     cpu = bs.start(parallelism=2, device="cpu").run(fv, 4)
     gpu = bs.start(parallelism=2, device="cuda:0").run(fv, 4)
     assert sorted(cpu.scan()) == sorted(gpu.scan())
+
+
+def test_grouptable_regrow_on_overflow(kernels):
+    # Force the overflow path: tiny initial capacity, many distinct keys.
+    n = 3_000_000
+    keys = torch.arange(n, dtype=torch.int64, device="cuda:0")
+    vals = torch.ones(n, dtype=torch.int64, device="cuda:0")
+    t = kernels.GroupTable([torch.int64], ["sum"], "cuda:0", cap_hint=1024)
+    t.insert(keys, [vals])
+    out_keys, out_vals = t.finish()
+    assert out_keys.shape[0] == n
+    assert int(out_vals[0].sum()) == n
+
+
+def test_grouptable_streaming_inserts(kernels):
+    t = kernels.GroupTable([torch.int64], ["sum"], "cuda:0")
+    for i in range(5):
+        keys = torch.randint(0, 1000, (200_000,), dtype=torch.int64,
+                             device="cuda:0")
+        t.insert(keys, [torch.ones_like(keys)])
+    out_keys, out_vals = t.finish()
+    assert int(out_vals[0].sum()) == 1_000_000
+    assert out_keys.shape[0] <= 1000
