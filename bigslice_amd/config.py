@@ -37,7 +37,7 @@ SORT_SPILL_TARGET_BYTES = _env_int("BIGSLICE_SORT_SPILL_BYTES", 4 << 30)
 
 # Initial hash-aggregate table capacity (slots); the table grows x2 on
 # probe-chain overflow (reference combiner grow policy, exec/combiner.go:47).
-GROUPBY_INITIAL_CAP = _env_int("BIGSLICE_GROUPBY_INITIAL_CAP", 1 << 23)
+GROUPBY_INITIAL_CAP = _env_int("BIGSLICE_GROUPBY_INITIAL_CAP", 1 << 22)
 
 # Maximum consecutive losses of a single task before giving up
 # (reference: exec/eval.go:30).

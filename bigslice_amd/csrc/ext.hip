@@ -250,10 +250,13 @@ std::vector<torch::Tensor> groupby_compact(torch::Tensor tkeys,
     cc.tab[c] = {tabs[c].data_ptr(), dtype_code(tabs[c])};
     cc.out[c] = {out[c + 1].data_ptr(), dtype_code(tabs[c])};
   }
-  int blocks = (int)std::min<int64_t>((cap + THREADS - 1) / THREADS, 16384);
+  // ~2048 blocks: fills the chip while keeping cursor atomics rare.
+  int64_t spb = (cap + 2047) / 2048;
+  if (spb < THREADS) spb = THREADS;
+  int blocks = (int)((cap + spb - 1) / spb);
   hipLaunchKernelGGL(k_groupby_compact, dim3(blocks), dim3(THREADS), 0,
-                     current_stream(), tkeys.data_ptr<int64_t>(), cap, cc,
-                     out_keys.data_ptr<int64_t>(),
+                     current_stream(), tkeys.data_ptr<int64_t>(), cap, spb,
+                     cc, out_keys.data_ptr<int64_t>(),
                      (unsigned long long*)cursor.data_ptr());
   HIP_CHECK(hipGetLastError());
   return out;

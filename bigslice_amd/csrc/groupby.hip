@@ -106,27 +106,44 @@ __device__ __forceinline__ void copy_elt(const MutColDesc& dst, int64_t di,
   }
 }
 
+// Two-pass block-chunk compaction: each block owns a contiguous slot
+// range; pass 1 counts its used slots (block reduce -> ONE cursor atomic
+// per block), pass 2 writes them out at block_base + thread prefix.
+// A single hot cursor saturates at ~88 atomics/us on this chip, so the
+// atomic count must scale with blocks (~2k), not waves (~100k).
 extern "C" __global__ void k_groupby_compact(
-    const int64_t* tkeys, int64_t cap, CompactCols cols, int64_t* out_keys,
-    unsigned long long* cursor) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  int lane = threadIdx.x & 63;
-  for (; i - lane < cap; i += stride) {
-    bool used = (i < cap) && (tkeys[i] != GB_SENTINEL);
-    uint64_t mask = __ballot(used);
-    if (mask == 0) continue;
-    unsigned long long base = 0;
-    if (lane == __ffsll((long long)mask) - 1)
-      base = atomicAdd(cursor, (unsigned long long)__popcll(mask));
-    base = __shfl(base, __ffsll((long long)mask) - 1);
-    if (used) {
-      int64_t pos =
-          (int64_t)base + __popcll(mask & ((1ull << lane) - 1));
-      out_keys[pos] = tkeys[i];
-      for (int c = 0; c < cols.n; ++c)
-        copy_elt(cols.out[c], pos, cols.tab[c], i);
+    const int64_t* tkeys, int64_t cap, int64_t slots_per_block,
+    CompactCols cols, int64_t* out_keys, unsigned long long* cursor) {
+  __shared__ uint32_t counts[THREADS];
+  __shared__ unsigned long long block_base;
+  int64_t start = (int64_t)blockIdx.x * slots_per_block;
+  int64_t end = min(start + slots_per_block, cap);
+  // pass 1: per-thread count over its strided slots
+  uint32_t mine = 0;
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    mine += (tkeys[i] != GB_SENTINEL);
+  counts[threadIdx.x] = mine;
+  __syncthreads();
+  // exclusive scan of per-thread counts (simple LDS scan, 256 wide)
+  if (threadIdx.x == 0) {
+    uint32_t run = 0;
+    for (int t = 0; t < (int)blockDim.x; ++t) {
+      uint32_t c = counts[t];
+      counts[t] = run;
+      run += c;
     }
+    block_base = run ? atomicAdd(cursor, (unsigned long long)run) : 0;
+  }
+  __syncthreads();
+  // pass 2: write used slots at base + prefix
+  int64_t pos = (int64_t)block_base + counts[threadIdx.x];
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+    int64_t k = tkeys[i];
+    if (k == GB_SENTINEL) continue;
+    out_keys[pos] = k;
+    for (int c = 0; c < cols.n; ++c)
+      copy_elt(cols.out[c], pos, cols.tab[c], i);
+    ++pos;
   }
 }
 
@@ -148,10 +165,17 @@ extern "C" __global__ void k_groupby_insert(
       uint64_t h = mm3_u64((uint64_t)k, seed) & mask;
       int64_t probes = 0;
       for (;;) {
-        long long prev = atomicCAS((unsigned long long*)&tkeys[h],
-                                   (unsigned long long)GB_SENTINEL,
-                                   (unsigned long long)k);
-        if (prev == GB_SENTINEL || prev == k) break;
+        // Plain read first: once the table is warm most probes land on
+        // an already-claimed matching key, and a load is far cheaper
+        // than an atomicCAS.
+        long long cur = ((volatile long long*)tkeys)[h];
+        if (cur == k) break;
+        if (cur == GB_SENTINEL) {
+          long long prev = atomicCAS((unsigned long long*)&tkeys[h],
+                                     (unsigned long long)GB_SENTINEL,
+                                     (unsigned long long)k);
+          if (prev == GB_SENTINEL || prev == k) break;
+        }
         h = (h + 1) & mask;
         // A long probe chain means the table is too loaded: signal the
         // host to grow x2 and re-insert (reference combiner grow policy).
