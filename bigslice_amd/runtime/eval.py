@@ -1,16 +1,21 @@
 """Evaluator: schedules the task graph to an executor.
 
 Role-parity: exec/eval.go:80-176 — enqueue roots, run runnable tasks
-concurrently, watch running tasks, resubmit LOST tasks together with their
-now-missing dependencies (:112-115, :352-376), give up after
-maxConsecutiveLost (:30, :139-159).  Multiple concurrent Evals coordinate
-through task state: a task already WAITING/RUNNING is watched, not re-run.
+concurrently, resubmit LOST tasks together with their now-missing
+dependencies (:112-115, :352-376), give up after maxConsecutiveLost
+(:30, :139-159).  Multiple concurrent Evals coordinate through task
+state: a task already WAITING/RUNNING is watched, not re-run.
+
+Implementation: callback-driven over a persistent thread pool (no
+watcher-thread-per-task, no polling): completion of a task decrements its
+consumers' pending counts and submits newly-runnable tasks.
 """
 
 from __future__ import annotations
 
 import threading
-from typing import List, Sequence
+from concurrent.futures import ThreadPoolExecutor
+from typing import Dict, List, Sequence, Set
 
 from .. import config
 from .task import Task, TaskState
@@ -29,7 +34,7 @@ class Executor:
 
     def run(self, task: Task) -> None:
         """Run the task, setting its state to RUNNING then OK/ERR/LOST.
-        Called from an evaluator goroutine; may block."""
+        Called from an evaluator pool thread; may block."""
         raise NotImplementedError
 
     def reader(self, task: Task, partition: int):
@@ -40,128 +45,138 @@ class Executor:
         pass
 
 
+def _reachable(roots: Sequence[Task]) -> List[Task]:
+    out: List[Task] = []
+    seen: Set[int] = set()
+
+    def visit(t: Task):
+        if id(t) in seen:
+            return
+        seen.add(id(t))
+        for dep in t.deps:
+            for h in dep.head_tasks:
+                visit(h)
+        out.append(t)
+
+    for r in roots:
+        visit(r)
+    return out
+
+
 def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
     """Evaluate all root tasks to OK (or raise)."""
-    pending_lock = threading.Lock()
-    done_event = threading.Event()
+    tasks = _reachable(roots)
+    consumers: Dict[int, List[Task]] = {}
+    for t in tasks:
+        for dep in t.deps:
+            for h in dep.head_tasks:
+                consumers.setdefault(id(h), []).append(t)
+
+    lock = threading.Lock()
+    done = threading.Condition(lock)
     errors: List[BaseException] = []
-    # Tasks this evaluation is responsible for watching.
-    watched = set()
-    inflight = [0]
+    # tasks this evaluation has claimed (submitted or is watching)
+    active: Set[int] = set()
 
-    def note_error(e: BaseException):
-        with pending_lock:
+    pool = getattr(executor, "pool", None)
+    own_pool = None
+    if pool is None:
+        own_pool = ThreadPoolExecutor(
+            max_workers=getattr(executor, "parallelism", None)
+            or config.DEFAULT_PARALLELISM)
+        pool = own_pool
+
+    def deps_ok(t: Task) -> bool:
+        return all(h.state == TaskState.OK
+                   for dep in t.deps for h in dep.head_tasks)
+
+    def fail(e: BaseException):
+        with lock:
             errors.append(e)
-        done_event.set()
+            done.notify_all()
 
-    def runnable(task: Task) -> bool:
-        return all(t.state == TaskState.OK
-                   for dep in task.deps for t in dep.head_tasks)
-
-    def enqueue(task: Task):
-        """Walk the graph; run tasks whose deps are satisfied; recurse
-        into deps otherwise (phase-aware recomputation: a LOST dep is
-        re-enqueued; exec/eval.go:255-451 semantics, simplified to
-        O(edges))."""
-        with pending_lock:
-            if task in watched and task.state in (
-                    TaskState.WAITING, TaskState.RUNNING):
-                return
-        st = task.state
+    def submit(t: Task):
+        """Called with lock held.  Schedule t if runnable, else make its
+        missing deps runnable first."""
+        if id(t) in active:
+            return
+        st = t.state
         if st == TaskState.OK:
             return
         if st == TaskState.ERR:
-            note_error(task.error or RuntimeError(f"{task.name} failed"))
+            errors.append(t.error or RuntimeError(f"{t.name} failed"))
+            done.notify_all()
             return
-        if task.consecutive_lost >= config.MAX_CONSECUTIVE_LOST:
-            note_error(TooManyTriesError(task))
+        if t.consecutive_lost >= config.MAX_CONSECUTIVE_LOST:
+            errors.append(TooManyTriesError(t))
+            done.notify_all()
             return
-        if runnable(task):
-            start(task)
+        if st in (TaskState.WAITING, TaskState.RUNNING):
+            # claimed by a concurrent evaluation: watch it
+            active.add(id(t))
+            threading.Thread(target=_watch, args=(t,), daemon=True).start()
+            return
+        if deps_ok(t):
+            active.add(id(t))
+            t.set_state(TaskState.WAITING)
+            pool.submit(_run, t)
         else:
-            # watch deps; when they complete we revisit this task
-            for dep in task.deps:
+            for dep in t.deps:
                 for h in dep.head_tasks:
-                    enqueue(h)
-            watch_until_deps_ready(task)
+                    if h.state != TaskState.OK:
+                        submit(h)
 
-    def start(task: Task):
-        with pending_lock:
-            if task in watched and task.state in (
-                    TaskState.WAITING, TaskState.RUNNING):
-                return
-            watched.add(task)
-            inflight[0] += 1
-        task.set_state(TaskState.WAITING)
+    def _watch(t: Task):
+        t.wait_state(TaskState.OK)
+        _finish(t)
 
-        def runner():
-            try:
-                executor.run(task)
-            except BaseException as e:  # executor bug; treat as ERR
-                task.set_state(TaskState.ERR, e)
-            finally:
-                finish(task)
+    def _run(t: Task):
+        try:
+            executor.run(t)
+        except BaseException as e:
+            t.set_state(TaskState.ERR, e)
+        _finish(t)
 
-        threading.Thread(target=runner, daemon=True).start()
+    def _finish(t: Task):
+        st = t.state
+        with lock:
+            active.discard(id(t))
+            if st == TaskState.OK:
+                for c in consumers.get(id(t), ()):  # wake consumers
+                    if id(c) not in active and c.state in (
+                            TaskState.INIT, TaskState.LOST) and deps_ok(c):
+                        submit(c)
+                # a consumer may be waiting only on this task at root
+                done.notify_all()
+            elif st == TaskState.LOST:
+                submit(t)  # resubmit (re-walks lost deps)
+                done.notify_all()
+            else:
+                errors.append(t.error or RuntimeError(f"{t.name} failed"))
+                done.notify_all()
 
-    def watch_until_deps_ready(task: Task):
-        with pending_lock:
-            if task in watched:
-                return
-            watched.add(task)
-            inflight[0] += 1
-
-        def waiter():
-            try:
-                for dep in task.deps:
-                    for h in dep.head_tasks:
-                        h.wait_state(TaskState.OK)
-                with pending_lock:
-                    watched.discard(task)
-                enqueue(task)
-            finally:
-                finish_watch()
-        threading.Thread(target=waiter, daemon=True).start()
-
-    def finish_watch():
-        with pending_lock:
-            inflight[0] -= 1
-            if inflight[0] == 0:
-                done_event.set()
-
-    def finish(task: Task):
-        st = task.state
-        with pending_lock:
-            watched.discard(task)
-            inflight[0] -= 1
-        if st == TaskState.OK:
-            pass
-        elif st == TaskState.LOST:
-            # resubmit: deps may also be lost; enqueue re-walks
-            enqueue(task)
-        elif st == TaskState.ERR:
-            note_error(task.error or RuntimeError(f"{task.name} failed"))
-        with pending_lock:
-            if inflight[0] == 0:
-                done_event.set()
-
-    for r in roots:
-        enqueue(r)
-    while True:
-        if all(r.state == TaskState.OK for r in roots):
-            return
-        done_event.wait(timeout=0.05)
-        with pending_lock:
-            if errors:
-                raise errors[0]
-            if inflight[0] == 0:
+    try:
+        with lock:
+            for r in roots:
+                submit(r)
+            while not errors:
                 if all(r.state == TaskState.OK for r in roots):
                     return
-                # quiescent but incomplete: re-enqueue (lost deps)
-                done_event.clear()
-                needs = [r for r in roots if r.state != TaskState.OK]
-            else:
-                done_event.clear()
-                continue
-        for r in needs:
-            enqueue(r)
+                if not active:
+                    # quiescent but incomplete: re-walk from roots
+                    # (lost interior tasks)
+                    progressed = False
+                    for r in roots:
+                        if r.state != TaskState.OK:
+                            submit(r)
+                            progressed = True
+                    if not progressed or not active:
+                        if errors:
+                            break
+                        if all(r.state == TaskState.OK for r in roots):
+                            return
+                done.wait(timeout=1.0)
+            raise errors[0]
+    finally:
+        if own_pool is not None:
+            own_pool.shutdown(wait=False)

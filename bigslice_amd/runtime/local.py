@@ -36,8 +36,26 @@ class LocalExecutor(Executor):
         self.device = device
         self.store = store or MemoryStore()
         self._sem = threading.Semaphore(self.parallelism)
+        # persistent worker pool reused across evaluations (one thread
+        # per concurrent shard task; GPU work overlaps via the HIP
+        # stream while threads trade the GIL)
+        from concurrent.futures import ThreadPoolExecutor
+        self.pool = ThreadPoolExecutor(max_workers=self.parallelism)
         # fault injection hook for chaos tests: fn(task) -> None or raise
         self.fault_hook = None
+        self._tls = threading.local()
+
+    def _stream(self):
+        """Per-worker-thread HIP stream: concurrent shard tasks overlap
+        on the device and one task's sync does not convoy behind
+        another's queued kernels."""
+        if not self.device.startswith("cuda"):
+            return None
+        s = getattr(self._tls, "stream", None)
+        if s is None:
+            s = torch.cuda.Stream(device=self.device)
+            self._tls.stream = s
+        return s
 
     # -- Executor ---------------------------------------------------------
 
@@ -49,7 +67,15 @@ class LocalExecutor(Executor):
         try:
             if self.fault_hook is not None:
                 self.fault_hook(task)
-            self._run_inner(task)
+            stream = self._stream()
+            if stream is not None:
+                with torch.cuda.stream(stream):
+                    self._run_inner(task)
+                # stored frames must be materialized before consumers
+                # (possibly on other streams) read them
+                stream.synchronize()
+            else:
+                self._run_inner(task)
             task.set_state(TaskState.OK)
         except TaskLost as e:
             task.set_state(TaskState.LOST)
