@@ -1,0 +1,154 @@
+"""Secondary BASELINE.json measurement configs (bench.py covers the
+north-star config 3):
+
+  2: Map+Filter scan over a 100M-row frame, 1 GPU (per-row kernel path)
+  4: Cogroup two-slice join, 2 x keyed rows (partitioned sort-merge join)
+  5: External sort with spill to host DRAM
+
+Usage: python benchmarks/configs.py --config 2 [--rows N] [--steps K]
+Prints one JSON line per run (same shape as bench.py).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+import bigslice_amd as bs
+
+_DATA = {}
+
+
+def build_mapfilter(nshard):
+    def gen(shard, ctx):
+        x = _DATA[("mf", shard)]
+        for off in range(0, x.shape[0], ctx.chunk):
+            yield (x[off:off + ctx.chunk],)
+    src = bs.ReaderFunc(nshard, gen, bs.schema_of(int))
+    mapped = bs.Map(src, lambda x: (x * 3 + 1,))
+    return bs.Filter(mapped, lambda x: (x & 7) != 0)
+
+
+def build_cogroup(nshard):
+    def gen_a(shard, ctx):
+        yield _DATA[("cga", shard)]
+
+    def gen_b(shard, ctx):
+        yield _DATA[("cgb", shard)]
+    a = bs.ReaderFunc(nshard, gen_a, bs.schema_of(int, int))
+    b = bs.ReaderFunc(nshard, gen_b, bs.schema_of(int, int))
+    return bs.Cogroup(a, b)
+
+
+def build_sort(nshard):
+    def gen(shard, ctx):
+        keys, vals = _DATA[("sort", shard)]
+        for off in range(0, keys.shape[0], ctx.chunk):
+            yield (keys[off:off + ctx.chunk], vals[off:off + ctx.chunk])
+    src = bs.ReaderFunc(nshard, gen, bs.schema_of(int, int))
+
+    # terminal external sort per shard via WriterFunc-free custom slice
+    class SortSlice(bs.Slice):
+        def reader(self, shard, dep_readers, ctx):
+            from bigslice_amd.sortio import SortReader
+            from bigslice_amd import config as cfg
+            return SortReader(dep_readers[0],
+                              run_bytes=int(os.environ.get(
+                                  "SORT_RUN_BYTES", 2 << 30)),
+                              device=ctx.device, chunk=ctx.chunk)
+
+    return SortSlice(src.schema, nshard, deps=[bs.Dep(src)])
+
+
+FV_MF = bs.func(build_mapfilter)
+FV_CG = bs.func(build_cogroup)
+FV_SORT = bs.func(build_sort)
+
+
+def emit(metric, rows, elapsed_s, steps, warmup, cfg, device):
+    ms = elapsed_s * 1000 / steps
+    print(json.dumps({
+        "metric": metric, "value": rows / (ms / 1000), "unit": "rows/sec",
+        "n_gpus": 1, "steps": steps, "warmup": warmup,
+        "ms_per_step": ms, "higher_is_better": True, "scaling": "weak",
+        "vs_baseline": None, "dtype": "int64", "data": "synthetic",
+        "config": cfg | {"device": device},
+    }))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--config", type=int, required=True, choices=[2, 4, 5])
+    ap.add_argument("--rows", type=int, default=None)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--shards", type=int, default=8)
+    args = ap.parse_args()
+    device = "cuda:0" if torch.cuda.is_available() else "cpu"
+    nshard = args.shards
+    sess = bs.start(parallelism=nshard, device=device)
+
+    def g(shape_rows, hi, seed):
+        gen = torch.Generator(device=device)
+        gen.manual_seed(seed)
+        return torch.randint(0, hi, (shape_rows,), dtype=torch.int64,
+                             device=device, generator=gen)
+
+    if args.config == 2:
+        rows = args.rows or 100_000_000
+        per = rows // nshard
+        for s in range(nshard):
+            _DATA[("mf", s)] = g(per, 1 << 40, s)
+        fv, cfg = FV_MF, {"model": "Map+Filter scan (BASELINE config 2)",
+                          "rows_total": per * nshard,
+                          "global_batch": per * nshard,
+                          "shards": nshard, "parallelism": "dp1"}
+        metric = "rows/sec Map+Filter scan"
+    elif args.config == 4:
+        rows = args.rows or 500_000_000
+        per = rows // nshard
+        for s in range(nshard):
+            _DATA[("cga", s)] = (g(per, 1 << 20, 100 + s), g(per, 1 << 30, 200 + s))
+            _DATA[("cgb", s)] = (g(per, 1 << 20, 300 + s), g(per, 1 << 30, 400 + s))
+        fv, cfg = FV_CG, {"model": "Cogroup join (BASELINE config 4)",
+                          "rows_total": 2 * per * nshard,
+                          "global_batch": 2 * per * nshard,
+                          "shards": nshard, "parallelism": "dp1"}
+        rows = 2 * per * nshard
+        metric = "rows/sec Cogroup join"
+    else:
+        rows = args.rows or 1_000_000_000
+        per = rows // nshard
+        for s in range(nshard):
+            _DATA[("sort", s)] = (g(per, 1 << 62, 500 + s),
+                                  g(per, 1 << 30, 600 + s))
+        fv, cfg = FV_SORT, {"model": "External sort (BASELINE config 5)",
+                            "rows_total": per * nshard,
+                            "global_batch": per * nshard,
+                            "shards": nshard, "parallelism": "dp1"}
+        metric = "rows/sec external sort"
+
+    def step():
+        res = sess.run(fv, nshard)
+        res.discard()
+
+    for _ in range(args.warmup):
+        step()
+    if device.startswith("cuda"):
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if device.startswith("cuda"):
+        torch.cuda.synchronize()
+    emit(metric, rows, time.perf_counter() - t0, args.steps, args.warmup,
+         cfg, device)
+
+
+if __name__ == "__main__":
+    main()

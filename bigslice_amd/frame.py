@@ -24,11 +24,59 @@ from . import hashing
 from .schema import OBJECT, Schema, infer_dtype, is_object
 
 
-Column = Union[torch.Tensor, list]
+class SegmentedColumn:
+    """A device-resident list-valued column: row i is
+    values[starts[i]:ends[i]].  This is the MI355X representation of the
+    reference's slice-valued Cogroup output columns ([]T, cogroup.go:94):
+    the segments stay in HBM as one values tensor + per-row offset
+    tensors, and only materialize as Python lists at scan time."""
+
+    __slots__ = ("values", "starts", "ends")
+
+    def __init__(self, values: torch.Tensor, starts: torch.Tensor,
+                 ends: torch.Tensor):
+        self.values = values
+        self.starts = starts
+        self.ends = ends
+
+    def __len__(self):
+        return self.starts.shape[0]
+
+    def __getitem__(self, idx):
+        if isinstance(idx, slice):
+            return SegmentedColumn(self.values, self.starts[idx],
+                                   self.ends[idx])
+        raise TypeError("SegmentedColumn supports slice indexing only")
+
+    def select(self, indices: torch.Tensor) -> "SegmentedColumn":
+        idx = indices.to(self.starts.device)
+        return SegmentedColumn(self.values, self.starts[idx],
+                               self.ends[idx])
+
+    def to(self, device, non_blocking=False) -> "SegmentedColumn":
+        return SegmentedColumn(
+            self.values.to(device, non_blocking=non_blocking),
+            self.starts.to(device, non_blocking=non_blocking),
+            self.ends.to(device, non_blocking=non_blocking))
+
+    def nbytes(self) -> int:
+        return (self.values.numel() * self.values.element_size() +
+                self.starts.numel() * 16)
+
+    def tolists(self) -> list:
+        v = self.values.cpu()
+        s = self.starts.cpu().tolist()
+        e = self.ends.cpu().tolist()
+        return [v[a:b].tolist() for a, b in zip(s, e)]
+
+
+Column = Union[torch.Tensor, list, SegmentedColumn]
 
 
 def _col_len(col: Column) -> int:
-    return col.shape[0] if isinstance(col, torch.Tensor) else len(col)
+    if isinstance(col, torch.Tensor):
+        return col.shape[0]
+    return len(col)
 
 
 def _col_dtype(col: Column):
@@ -103,13 +151,18 @@ class Frame:
 
     @property
     def has_objects(self) -> bool:
-        return any(not isinstance(c, torch.Tensor) for c in self.columns)
+        """True when a column is host-only (Python lists); segmented
+        device columns are NOT host-only."""
+        return any(not isinstance(c, (torch.Tensor, SegmentedColumn))
+                   for c in self.columns)
 
     def nbytes(self) -> int:
         total = 0
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 total += c.numel() * c.element_size()
+            elif isinstance(c, SegmentedColumn):
+                total += c.nbytes()
             else:
                 total += sum(len(str(x)) for x in c)  # rough
         return total
@@ -127,6 +180,8 @@ class Frame:
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 cols.append(c[indices.to(c.device)])
+            elif isinstance(c, SegmentedColumn):
+                cols.append(c.select(indices))
             else:
                 idx = indices.cpu().tolist()
                 cols.append([c[i] for i in idx])
@@ -139,6 +194,8 @@ class Frame:
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 cols.append(c[keep.to(c.device)])
+            elif isinstance(c, SegmentedColumn):
+                cols.append(c.select(keep.nonzero().flatten()))
             else:
                 km = keep.cpu().numpy()
                 cols.append([x for x, k in zip(c, km) if k])
@@ -158,6 +215,16 @@ class Frame:
             parts = [f.columns[i] for f in frames]
             if isinstance(parts[0], torch.Tensor):
                 cols.append(torch.cat(parts))
+            elif isinstance(parts[0], SegmentedColumn):
+                vals, starts, ends, off = [], [], [], 0
+                for p in parts:
+                    vals.append(p.values)
+                    starts.append(p.starts + off)
+                    ends.append(p.ends + off)
+                    off += p.values.shape[0]
+                cols.append(SegmentedColumn(torch.cat(vals),
+                                            torch.cat(starts),
+                                            torch.cat(ends)))
             else:
                 merged: list = []
                 for p in parts:
@@ -168,7 +235,7 @@ class Frame:
     def to(self, device: str, non_blocking: bool = False) -> "Frame":
         cols: List[Column] = []
         for c in self.columns:
-            if isinstance(c, torch.Tensor):
+            if isinstance(c, (torch.Tensor, SegmentedColumn)):
                 cols.append(c.to(device, non_blocking=non_blocking))
             else:
                 if device != "cpu":
@@ -178,9 +245,17 @@ class Frame:
         return Frame(cols, self.prefix)
 
     def clone(self) -> "Frame":
-        return Frame(
-            [c.clone() if isinstance(c, torch.Tensor) else list(c)
-             for c in self.columns], self.prefix)
+        cols = []
+        for c in self.columns:
+            if isinstance(c, torch.Tensor):
+                cols.append(c.clone())
+            elif isinstance(c, SegmentedColumn):
+                cols.append(SegmentedColumn(c.values.clone(),
+                                            c.starts.clone(),
+                                            c.ends.clone()))
+            else:
+                cols.append(list(c))
+        return Frame(cols, self.prefix)
 
     def with_prefix(self, prefix: int) -> "Frame":
         if not (0 < prefix <= self.num_columns):
@@ -227,6 +302,8 @@ class Frame:
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 out.append(c.cpu().tolist())
+            elif isinstance(c, SegmentedColumn):
+                out.append(c.tolists())
             else:
                 out.append(list(c))
         return out

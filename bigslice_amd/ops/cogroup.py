@@ -46,6 +46,14 @@ class Cogroup(Slice):
     def reader(self, shard, dep_readers, ctx: TaskContext) -> Reader:
         nkey = self.schema.prefix
         val_counts = self._val_counts
+        # Device fast path: single int64 key, numeric value columns ->
+        # sort-merge cogroup on device with SegmentedColumn outputs.
+        import torch as _t
+        if (ctx.device != "cpu" and nkey == 1
+                and self.deps[0].slice.schema.dtypes[0] == _t.int64
+                and all(dt != OBJECT
+                        for d in self.deps for dt in d.slice.schema.dtypes)):
+            return IterReader(self._device_gen(dep_readers, ctx))
 
         def gen():
             # Group each dep by key on the host: key -> [lists per column]
@@ -84,3 +92,59 @@ class Cogroup(Slice):
 
 def _key_sort(k):
     return k
+
+
+def _device_cogroup_columns(dep_frames, union_keys):
+    """Sort one dep's rows by key; return per-value-column
+    SegmentedColumns aligned with union_keys (searchsorted segment
+    boundaries over the key-sorted values)."""
+    import torch
+
+    from ..frame import Frame, SegmentedColumn
+    from .. import kernels
+    from ..sortio import sort_frame
+
+    cols = []
+    if dep_frames:
+        f = Frame.concat(dep_frames)
+        f = sort_frame(f)
+        keys = f.columns[0].contiguous()
+        starts = torch.searchsorted(keys, union_keys, right=False)
+        ends = torch.searchsorted(keys, union_keys, right=True)
+        for v in f.columns[1:]:
+            cols.append(SegmentedColumn(v.contiguous(), starts, ends))
+        return cols, keys
+    return cols, None
+
+
+# Attach the device generator to Cogroup (kept separate for readability).
+def _cogroup_device_gen(self, dep_readers, ctx):
+    import torch
+
+    from ..frame import Frame, SegmentedColumn
+
+    dep_frames = [[f for f in r] for r in dep_readers]
+    device = ctx.device
+    all_keys = [f.columns[0] for frames in dep_frames for f in frames]
+    if not all_keys:
+        return
+    union_keys = torch.unique(torch.cat(all_keys))  # sorted unique
+    out_cols = [union_keys]
+    for di, frames in enumerate(dep_frames):
+        nv = self._val_counts[di]
+        segs, _ = _device_cogroup_columns(frames, union_keys)
+        if segs:
+            out_cols.extend(segs)
+        else:
+            empty_vals = [torch.empty(0, dtype=dt, device=device)
+                          for dt in self.deps[di].slice.schema.dtypes[1:]]
+            zeros = torch.zeros(union_keys.shape[0], dtype=torch.int64,
+                                device=device)
+            out_cols.extend(SegmentedColumn(ev, zeros, zeros)
+                            for ev in empty_vals)
+    frame = Frame(out_cols, prefix=1)
+    for off in range(0, len(frame), ctx.chunk):
+        yield frame.slice(off, min(off + ctx.chunk, len(frame)))
+
+
+Cogroup._device_gen = _cogroup_device_gen

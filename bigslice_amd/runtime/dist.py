@@ -22,6 +22,7 @@ import torch
 
 from ..ops.slice_base import TaskContext
 from ..parallel.comm import Comm
+from ..utils import metrics
 from ..sliceio import MultiReader, Reader
 from .eval import Executor
 from .partition import PartitionWriter
@@ -42,6 +43,8 @@ class DistExecutor(Executor):
         self.comm = comm
         self.store = MemoryStore()
         self.device = comm.device
+        self.scopes = {}
+        self.tracer = None
 
     # The session calls executor.evaluate when present (instead of the
     # generic task-pull evaluator).
@@ -139,6 +142,13 @@ class DistExecutor(Executor):
         """Run one task; returns per-partition frame lists (or None for
         terminal tasks)."""
         task.set_state(TaskState.RUNNING)
+        scope = metrics.Scope()
+        self.scopes[task.name] = scope
+        self._scope_ctx = metrics.scoped(scope)
+        self._scope_ctx.__enter__()
+        if self.tracer:
+            self._span = self.tracer.span(task.name, pid=self.comm.rank)
+            self._span.__enter__()
         ctx = TaskContext(device=self.device)
         dep_readers = []
         for dep in task.deps:
@@ -155,13 +165,20 @@ class DistExecutor(Executor):
             for _ in out:
                 pass
             self.store.put(task.name, 0, [], 0)
+            self._scope_ctx.__exit__(None, None, None)
+            if self.tracer:
+                self._span.__exit__(None, None, None)
             return None
         w = PartitionWriter(task.num_partitions, task.partitioner,
                             task.combiner, task.schema, self.device,
                             ctx.chunk)
         for f in out:
             w.add(f)
-        return w.finish()
+        buckets = w.finish()
+        self._scope_ctx.__exit__(None, None, None)
+        if self.tracer:
+            self._span.__exit__(None, None, None)
+        return buckets
 
     # -- Executor interface (driver-side readback) ------------------------
 
@@ -176,6 +193,28 @@ class DistExecutor(Executor):
 
     def discard(self, task: Task) -> None:
         self.store.discard_task(task.name)
+
+    def merged_scope(self, tasks):
+        """Collective: merge task scopes across ranks."""
+        local = metrics.Scope()
+        seen = set()
+
+        def visit(t):
+            if t.name in seen:
+                return
+            seen.add(t.name)
+            s = self.scopes.get(t.name)
+            if s is not None:
+                local.merge(s)
+            for dep in t.deps:
+                for h in dep.head_tasks:
+                    visit(h)
+        for t in tasks:
+            visit(t)
+        out = metrics.Scope()
+        for d in self.comm.all_gather_obj(local.to_dict()):
+            out.merge(metrics.Scope.from_dict(d))
+        return out
 
     def gather_result(self, tasks: Sequence[Task], schema):
         """Collective: gather all root-task outputs to rank 0."""

@@ -16,6 +16,7 @@ import torch
 
 from .. import config
 from ..ops.slice_base import TaskContext
+from ..utils import metrics
 from ..sliceio import MultiReader, Reader
 from .eval import Executor
 from .partition import PartitionWriter
@@ -44,6 +45,8 @@ class LocalExecutor(Executor):
         # fault injection hook for chaos tests: fn(task) -> None or raise
         self.fault_hook = None
         self._tls = threading.local()
+        self.scopes = {}  # task name -> metrics.Scope
+        self.tracer = None  # utils.trace.Tracer, set by the session
 
     def _stream(self):
         """Per-worker-thread HIP stream: concurrent shard tasks overlap
@@ -67,15 +70,24 @@ class LocalExecutor(Executor):
         try:
             if self.fault_hook is not None:
                 self.fault_hook(task)
-            stream = self._stream()
-            if stream is not None:
-                with torch.cuda.stream(stream):
+            scope = metrics.Scope()
+            span = (self.tracer.span(task.name, pid=0)
+                    if self.tracer else None)
+            with metrics.scoped(scope):
+                if span:
+                    span.__enter__()
+                stream = self._stream()
+                if stream is not None:
+                    with torch.cuda.stream(stream):
+                        self._run_inner(task)
+                    # stored frames must be materialized before consumers
+                    # (possibly on other streams) read them
+                    stream.synchronize()
+                else:
                     self._run_inner(task)
-                # stored frames must be materialized before consumers
-                # (possibly on other streams) read them
-                stream.synchronize()
-            else:
-                self._run_inner(task)
+                if span:
+                    span.__exit__(None, None, None)
+            self.scopes[task.name] = scope
             task.set_state(TaskState.OK)
         except TaskLost as e:
             task.set_state(TaskState.LOST)
@@ -90,9 +102,18 @@ class LocalExecutor(Executor):
         ctx = TaskContext(device=self.device)
         dep_readers = []
         for dep in task.deps:
-            readers = [self.store.open(h.name, dep.partition,
-                                       device=self.device)
-                       for h in dep.head_tasks]
+            readers = []
+            for h in dep.head_tasks:
+                try:
+                    readers.append(self.store.open(
+                        h.name, dep.partition, device=self.device))
+                except KeyError:
+                    # The dep's stored output vanished (machine-loss
+                    # analog): mark the producer LOST so the evaluator
+                    # recomputes it, and resubmit this task
+                    # (exec/eval.go:352-376 semantics).
+                    h.set_state(TaskState.LOST)
+                    raise TaskLost(f"missing dep output {h.name}")
             if dep.expand:
                 dep_readers.append(readers)
             else:
@@ -119,3 +140,23 @@ class LocalExecutor(Executor):
 
     def discard(self, task: Task) -> None:
         self.store.discard_task(task.name)
+
+    def merged_scope(self, tasks):
+        """Merge task scopes across the reachable graph
+        (Result.Scope semantics, exec/session.go:418-426)."""
+        out = metrics.Scope()
+        seen = set()
+
+        def visit(t):
+            if t.name in seen:
+                return
+            seen.add(t.name)
+            s = self.scopes.get(t.name)
+            if s is not None:
+                out.merge(s)
+            for dep in t.deps:
+                for h in dep.head_tasks:
+                    visit(h)
+        for t in tasks:
+            visit(t)
+        return out

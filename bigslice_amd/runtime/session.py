@@ -142,6 +142,16 @@ class Result(Slice):
         """Iterate result rows."""
         return self.scanner().rows()
 
+    def scope(self):
+        """Merged user-metric scope over this result's task graph
+        (Result.Scope, exec/session.go:418-426).  Collective with a
+        distributed executor."""
+        ms = getattr(self.session.executor, "merged_scope", None)
+        if ms is None:
+            from ..utils import metrics
+            return metrics.Scope()
+        return ms(self.tasks)
+
     def discard(self) -> None:
         """Free stored task outputs (session.go:231-251)."""
         seen = set()
@@ -163,12 +173,21 @@ class Result(Slice):
 class Session:
     """An execution session bound to an executor (exec/session.go)."""
 
-    def __init__(self, executor: Executor, parallelism: int = None):
+    def __init__(self, executor: Executor, parallelism: int = None,
+                 trace_path: str = None):
         self.executor = executor
         self.parallelism = parallelism
         self._inv_counter = 0
         self._lock = threading.Lock()
         self.env = CompileEnv()
+        self.trace_path = trace_path
+        if trace_path is not None:
+            from ..utils.trace import Tracer
+            self.tracer = Tracer()
+            if hasattr(executor, "tracer"):
+                executor.tracer = self.tracer
+        else:
+            self.tracer = None
 
     def run(self, funcv: FuncValue, *args) -> Result:
         if not isinstance(funcv, FuncValue):
@@ -197,13 +216,21 @@ class Session:
         return self.run(funcv, *args)
 
     def shutdown(self):
+        if self.tracer is not None and self.trace_path:
+            self.tracer.write(self.trace_path)
         sd = getattr(self.executor, "shutdown", None)
         if sd is not None:
             sd()
 
+    def status(self):
+        """Task-state counts by invocation (the status display's
+        rollup, exec/slicestatus.go)."""
+        return dict(self._status) if hasattr(self, "_status") else {}
+
 
 def start(parallelism: int = None, device: str = None,
-          executor: Executor = None, distributed: bool = None) -> Session:
+          executor: Executor = None, distributed: bool = None,
+          trace_path: str = None) -> Session:
     """Create a session (exec.Start analog).
 
     distributed=True (or WORLD_SIZE>1 in the environment) starts the SPMD
@@ -224,4 +251,4 @@ def start(parallelism: int = None, device: str = None,
         else:
             from .local import LocalExecutor
             executor = LocalExecutor(parallelism=parallelism, device=device)
-    return Session(executor, parallelism)
+    return Session(executor, parallelism, trace_path=trace_path)

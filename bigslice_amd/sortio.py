@@ -1,0 +1,243 @@
+"""External sort and merge machinery.
+
+Role-parity: sortio/sort.go (SortReader: read -> sort in-core chunk ->
+spill -> k-way merge; NewMergeReader heap merge) and sortio/reader.go
+(Reduce: merge + adjacent-equal-key combine).
+
+MI355X redesign: in-core runs are whole device batches sorted with the
+radix-sort kernel (K6); spilled runs tier to pinned host DRAM through the
+Spiller; the k-way merge is batch-granular on device: take a window from
+each run, cut at the minimum window-max key, and merge the cut rows with
+one device sort (no per-row heap).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from . import config
+from .frame import Frame
+from .sliceio import Reader, IterReader
+from .sliceio.spiller import Spiller
+
+
+def sort_frame(frame: Frame) -> Frame:
+    """Sort one frame by its key prefix (device radix sort for single
+    numeric keys, lexicographic stable sort otherwise)."""
+    if len(frame) <= 1:
+        return frame
+    first = frame.columns[0]
+    if (frame.prefix == 1 and isinstance(first, torch.Tensor)
+            and first.is_cuda):
+        from . import kernels
+        if kernels.sort_pairs_supported(first):
+            perm = kernels.radix_argsort(first.contiguous())
+            return frame.select(perm)
+    return frame.sort_by_prefix()
+
+
+class SortReader(Reader):
+    """Drains a reader, producing key-sorted output: in-core runs of up
+    to `run_bytes` are sorted and spilled; read() merges the runs
+    (sortio/sort.go:31-76)."""
+
+    def __init__(self, source: Reader, run_bytes: int = None,
+                 device: str = "cpu", chunk: int = None,
+                 host_budget: int = 64 << 30):
+        self.source = source
+        self.run_bytes = run_bytes or config.SORT_SPILL_TARGET_BYTES
+        self.device = device
+        self.chunk = chunk or config.chunk_rows(device)
+        self.host_budget = host_budget
+        self._merged: Optional[Reader] = None
+
+    def _build(self) -> Reader:
+        spiller = Spiller(host_budget_bytes=self.host_budget)
+        pending: List[Frame] = []
+        pending_bytes = 0
+        runs: List[List[Frame]] = []
+
+        def flush_run():
+            nonlocal pending, pending_bytes
+            if not pending:
+                return
+            run = sort_frame(Frame.concat(pending))
+            # keep the run on-device if it is the only one so far and
+            # small; otherwise spill to host DRAM
+            start = spiller.num_batches()
+            spiller.spill(run)
+            runs.append(list(range(start, spiller.num_batches())))
+            pending = []
+            pending_bytes = 0
+
+        for f in self.source:
+            pending.append(f)
+            pending_bytes += f.nbytes()
+            if pending_bytes >= self.run_bytes:
+                flush_run()
+        flush_run()
+        if not runs:
+            return IterReader(iter(()))
+        readers = [_SpillRunReader(spiller, idxs, self.device)
+                   for idxs in runs]
+        if len(readers) == 1:
+            return readers[0]
+        return MergeReader(readers, chunk=self.chunk)
+
+    def read(self) -> Optional[Frame]:
+        if self._merged is None:
+            self._merged = self._build()
+        return self._merged.read()
+
+
+class _SpillRunReader(Reader):
+    def __init__(self, spiller: Spiller, batch_idxs: List[int],
+                 device: str):
+        self.spiller = spiller
+        self.idxs = batch_idxs
+        self.device = device
+        self.i = 0
+
+    def read(self) -> Optional[Frame]:
+        if self.i >= len(self.idxs):
+            return None
+        b = self.spiller.batches[self.idxs[self.i]]
+        self.i += 1
+        from .sliceio.spiller import _DiskBatch
+        if isinstance(b, _DiskBatch):
+            return b.load(self.device)
+        f: Frame = b
+        if self.device != "cpu":
+            f = f.to(self.device, non_blocking=True)
+        return f
+
+
+class _RunCursor:
+    """Buffered cursor over one sorted run."""
+
+    def __init__(self, reader: Reader):
+        self.reader = reader
+        self.buf: Optional[Frame] = None
+        self.eof = False
+
+    def fill(self, want_rows: int) -> None:
+        while not self.eof and (self.buf is None or len(self.buf) <
+                                want_rows):
+            f = self.reader.read()
+            if f is None:
+                self.eof = True
+                return
+            self.buf = f if self.buf is None else Frame.concat(
+                [self.buf, f])
+
+    def max_buffered_key(self):
+        c = self.buf.columns[0]
+        if isinstance(c, torch.Tensor):
+            return c[-1].item() if self.eof else c[-1]
+        return c[-1]
+
+    def take_upto(self, cutoff) -> Optional[Frame]:
+        """Rows with key <= cutoff (buffer is sorted)."""
+        c = self.buf.columns[0]
+        if isinstance(c, torch.Tensor):
+            n = int(torch.searchsorted(c, cutoff, right=True).item()) \
+                if isinstance(cutoff, torch.Tensor) else \
+                int(torch.searchsorted(
+                    c, torch.tensor(cutoff, dtype=c.dtype,
+                                    device=c.device), right=True).item())
+        else:
+            import bisect
+            n = bisect.bisect_right(c, cutoff)
+        if n == 0:
+            return None
+        out = self.buf.slice(0, n)
+        self.buf = self.buf.slice(n, len(self.buf)) \
+            if n < len(self.buf) else None
+        return out
+
+
+class MergeReader(Reader):
+    """Batch-granular k-way merge of sorted runs (sortio/sort.go:161-222
+    redesigned): each read() pulls a window per run, cuts at the minimum
+    window-max key (guaranteeing global order), and sorts the union of
+    the cut rows on device."""
+
+    def __init__(self, readers: List[Reader], chunk: int = None):
+        self.cursors = [_RunCursor(r) for r in readers]
+        self.chunk = chunk or 1 << 20
+
+    def read(self) -> Optional[Frame]:
+        per_run = max(self.chunk // max(len(self.cursors), 1), 1024)
+        live = []
+        for c in self.cursors:
+            c.fill(per_run)
+            if c.buf is not None and len(c.buf) > 0:
+                live.append(c)
+        if not live:
+            return None
+        if len(live) == 1:
+            out = live[0].buf
+            live[0].buf = None
+            return out
+        # cutoff = min over runs of the last buffered first-key.  Rows
+        # with first-key <= cutoff are globally mergeable ONLY once every
+        # run whose buffer ends exactly at the cutoff either has more
+        # buffered beyond it or is at EOF — otherwise a later batch of
+        # that run could continue the same first-key with smaller
+        # secondary key columns.
+        while True:
+            maxes = [c.max_buffered_key() for c in live]
+            if isinstance(maxes[0], torch.Tensor):
+                cutoff = torch.stack(list(maxes)).min()
+                at_cut = [bool((m == cutoff).item()) for m in maxes]
+            else:
+                cutoff = min(maxes)
+                at_cut = [m == cutoff for m in maxes]
+            grew = False
+            for c, ac in zip(live, at_cut):
+                if ac and not c.eof:
+                    want = len(c.buf) * 2
+                    c.fill(want)
+                    if len(c.buf) > want // 2 or c.eof:
+                        grew = True
+            if not grew:
+                break
+        parts = []
+        for c in live:
+            p = c.take_upto(cutoff)
+            if p is not None:
+                parts.append(p)
+        merged = sort_frame(Frame.concat(parts))
+        return merged
+
+
+def reduce_reader(readers: List[Reader], schema, agg,
+                  device: str = "cpu", chunk: int = None) -> Reader:
+    """Merge sorted per-producer streams, combining equal keys
+    (sortio/reader.go:36-130).  Used by sorted-reduce consumers; the
+    hash-aggregate path (ops.aggregate) is the default for Reduce."""
+    merged = MergeReader(readers, chunk=chunk)
+
+    def gen():
+        from .ops.aggregate import make_aggregator
+        carry: Optional[Frame] = None
+        for f in merged:
+            if carry is not None:
+                f = Frame.concat([carry, f])
+            # combine adjacent equal keys; hold back the last key group
+            # (it may continue in the next batch)
+            agg_ = make_aggregator(schema, agg, device)
+            agg_.add(f)
+            combined = Frame.concat(
+                list(agg_.result_frames(1 << 62)))
+            combined = sort_frame(combined)
+            if len(combined) > 1:
+                carry = combined.slice(len(combined) - 1, len(combined))
+                yield combined.slice(0, len(combined) - 1)
+            else:
+                carry = combined
+        if carry is not None and len(carry):
+            yield carry
+    return IterReader(gen())

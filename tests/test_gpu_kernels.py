@@ -153,3 +153,24 @@ def test_grouptable_streaming_inserts(kernels):
     out_keys, out_vals = t.finish()
     assert int(out_vals[0].sum()) == 1_000_000
     assert out_keys.shape[0] <= 1000
+
+
+def test_cogroup_device_matches_cpu():
+    import bigslice_amd as bs
+
+    def build(nshard):
+        a = bs.Const(nshard,
+                     torch.tensor([1, 2, 1, 3], dtype=torch.int64),
+                     torch.tensor([10, 20, 30, 40], dtype=torch.int64))
+        b = bs.Const(nshard,
+                     torch.tensor([2, 3, 3], dtype=torch.int64),
+                     torch.tensor([5, 6, 7], dtype=torch.int64))
+        return bs.Cogroup(a, b)
+
+    fv = bs.func(build)
+    cpu = bs.start(parallelism=2, device="cpu").run(fv, 2)
+    gpu = bs.start(parallelism=2, device="cuda:0").run(fv, 2)
+
+    def norm(rows):
+        return sorted((k, sorted(va), sorted(vb)) for k, va, vb in rows)
+    assert norm(cpu.scan()) == norm(gpu.scan())

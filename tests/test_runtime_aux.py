@@ -1,0 +1,172 @@
+"""Auxiliary runtime subsystems: metrics, tracing, chaos/fault injection,
+file store, partial cache, slicetest harness."""
+
+import json
+import os
+import random
+
+import pytest
+import torch
+
+import bigslice_amd as bs
+from bigslice_amd import slicetest
+from bigslice_amd.runtime.local import LocalExecutor, TaskLost
+from bigslice_amd.runtime.session import Session
+from bigslice_amd.runtime.store import FileStore
+from bigslice_amd.utils import metrics
+
+
+def test_metrics_scope_merged():
+    filtered = metrics.counter("filtered-rows")
+
+    def build():
+        def count_filter(x):
+            keep = x % 2 == 0
+            filtered.incr(int((~keep).sum()))
+            return keep
+        return bs.Filter(bs.Const(4, torch.arange(100, dtype=torch.int64)),
+                         count_filter)
+
+    res = slicetest.run(build)
+    assert len(list(res.scan())) == 50
+    snap = res.scope().snapshot()
+    assert snap.get("filtered-rows") == 50
+
+
+def test_tracer_writes_chrome_trace(tmp_path):
+    path = str(tmp_path / "trace.json")
+    fv = bs.func(lambda: bs.Const(2, torch.arange(4, dtype=torch.int64)))
+    sess = bs.start(parallelism=2, device="cpu", trace_path=path)
+    sess.run(fv)
+    sess.shutdown()
+    with open(path) as fp:
+        data = json.load(fp)
+    names = [e["name"] for e in data["traceEvents"]]
+    assert any("const" in n for n in names)
+
+
+def test_chaos_random_task_loss():
+    # chaosmonkey_test.go analog: random transient losses; job completes
+    # with the right answer.
+    rng = random.Random(7)
+
+    def build():
+        keys = torch.arange(1000, dtype=torch.int64) % 13
+        vals = torch.ones(1000, dtype=torch.int64)
+        return bs.Reduce(bs.Const(6, keys, vals), "sum")
+
+    ex = LocalExecutor(parallelism=4, device="cpu")
+
+    def chaos(task):
+        if task.consecutive_lost < 2 and rng.random() < 0.4:
+            # also lose the task's stored output (machine-loss analog)
+            ex.store.discard_task(task.name)
+            raise TaskLost(task.name)
+    ex.fault_hook = chaos
+
+    sess = Session(ex)
+    fv = bs.func(build)
+    res = sess.run(fv)
+    got = dict(res.scan())
+    want = {}
+    for k in (torch.arange(1000) % 13).tolist():
+        want[k] = want.get(k, 0) + 1
+    assert got == want
+
+
+def test_chaos_interior_output_loss_recomputes():
+    # A consumer fails because its dep's output vanished (machine loss
+    # after OK): the evaluator must re-run the producer.
+    calls = {"n": 0}
+
+    def build():
+        keys = torch.arange(100, dtype=torch.int64) % 5
+        vals = torch.ones(100, dtype=torch.int64)
+        return bs.Reduce(bs.Const(2, keys, vals), "sum")
+
+    ex = LocalExecutor(parallelism=2, device="cpu")
+    lost_once = {"done": False}
+
+    def chaos(task):
+        # when the first consumer (reduce) task starts, drop one
+        # producer's output once
+        if task.deps and not lost_once["done"]:
+            lost_once["done"] = True
+            prod = task.deps[0].head_tasks[0]
+            ex.store.discard_task(prod.name)
+            from bigslice_amd.runtime.task import TaskState
+            prod.set_state(TaskState.LOST)
+            raise TaskLost("dep output lost")
+    ex.fault_hook = chaos
+    res = Session(ex).run(bs.func(build))
+    got = dict(res.scan())
+    assert got == {k: 20 for k in range(5)}
+
+
+def test_file_store_roundtrip(tmp_path):
+    from bigslice_amd.frame import Frame
+    st = FileStore(str(tmp_path / "store"))
+    f = Frame([torch.arange(10, dtype=torch.int64)])
+    st.put("taskA", 0, [f], 10)
+    assert st.has("taskA", 0)
+    size, rows = st.stat("taskA", 0)
+    assert rows == 10
+    frames = list(st.open("taskA", 0))
+    assert len(frames) == 1 and frames[0].columns[0].tolist() == \
+        list(range(10))
+    st.discard_task("taskA")
+    assert not st.has("taskA", 0)
+
+
+def test_file_store_executor_end_to_end(tmp_path):
+    # local executor with a file-backed store: every task output is a
+    # durable checkpoint (reference fileStore semantics).
+    ex = LocalExecutor(parallelism=2, device="cpu",
+                       store=FileStore(str(tmp_path / "store")))
+    fv = bs.func(lambda: bs.Reduce(
+        bs.Const(2, torch.tensor([1, 2, 1], dtype=torch.int64),
+                 torch.tensor([1, 1, 1], dtype=torch.int64)), "sum"))
+    res = Session(ex).run(fv)
+    assert sorted(res.scan()) == [(1, 2), (2, 1)]
+
+
+def test_partial_cache_recomputes_missing(tmp_path):
+    prefix = str(tmp_path / "pc")
+    computed = []
+
+    def gen(shard, ctx):
+        computed.append(shard)
+        yield (torch.arange(2, dtype=torch.int64) + shard * 10,)
+
+    def build():
+        return bs.Cache(bs.ReaderFunc(3, gen, bs.schema_of(int)), prefix,
+                        partial=True)
+
+    fv = bs.func(build)
+    r1 = bs.start(parallelism=2, device="cpu").run(fv)
+    assert len(computed) == 3
+    # drop one shard's cache file: only that shard recomputes
+    os.remove(prefix + "-0001-of-0003")
+    computed.clear()
+    r2 = bs.start(parallelism=2, device="cpu").run(fv)
+    assert sorted(r2.scan()) == [0, 1, 10, 11, 20, 21]
+    assert computed == [1]
+
+
+def test_slicetest_harness():
+    rows = slicetest.scan_all(
+        lambda: bs.Const(2, torch.arange(5, dtype=torch.int64)))
+    assert sorted(rows) == [0, 1, 2, 3, 4]
+    err = slicetest.run_err(
+        lambda: bs.Map(bs.Const(1, torch.arange(2, dtype=torch.int64)),
+                       lambda x: 1 / 0, out_schema=(int,)))
+    assert err is not None
+
+
+def test_error_propagates_from_udf():
+    def boom(x):
+        raise ValueError("user boom")
+    err = slicetest.run_err(
+        lambda: bs.Map(bs.Const(2, torch.arange(4, dtype=torch.int64)),
+                       boom, out_schema=(int,)))
+    assert isinstance(err, ValueError)

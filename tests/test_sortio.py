@@ -1,0 +1,102 @@
+"""External sort / merge tests over randomized frames (reference
+sortio/sort_test.go shapes, fuzzFrame analog)."""
+
+import operator
+
+import pytest
+import torch
+
+from bigslice_amd.frame import Frame
+from bigslice_amd.schema import Schema
+from bigslice_amd.sliceio import IterReader, read_all
+from bigslice_amd.sortio import MergeReader, SortReader, reduce_reader
+
+
+def frames_of(tensors, prefix=1, batch=1000):
+    def gen():
+        n = tensors[0].shape[0]
+        for off in range(0, n, batch):
+            yield Frame([t[off:off + batch] for t in tensors], prefix)
+    return IterReader(gen())
+
+
+def test_sortreader_sorts_across_spilled_runs():
+    g = torch.Generator().manual_seed(0)
+    keys = torch.randint(0, 10_000, (50_000,), generator=g,
+                         dtype=torch.int64)
+    vals = torch.arange(50_000, dtype=torch.int64)
+    # tiny run budget forces many spilled runs
+    sr = SortReader(frames_of([keys, vals]), run_bytes=100_000)
+    out = read_all(sr)
+    assert len(out) == 50_000
+    ok = out.columns[0]
+    assert bool((ok[1:] >= ok[:-1]).all())
+    # rows intact
+    got = sorted(zip(ok.tolist(), out.columns[1].tolist()))
+    want = sorted(zip(keys.tolist(), vals.tolist()))
+    assert got == want
+
+
+def test_merge_reader_two_runs():
+    a = torch.arange(0, 100, 2, dtype=torch.int64)
+    b = torch.arange(1, 101, 2, dtype=torch.int64)
+    m = MergeReader([frames_of([a.sort().values], batch=7),
+                     frames_of([b.sort().values], batch=13)], chunk=16)
+    out = read_all(m)
+    assert out.columns[0].tolist() == list(range(100))
+
+
+def test_merge_reader_multi_key_ties_across_batches():
+    # run 1 ends a batch mid-way through first-key 5; secondary keys of
+    # the next batch are smaller than run 2's: the merge must wait.
+    k1 = torch.tensor([1, 5, 5, 5, 5], dtype=torch.int64)
+    s1 = torch.tensor([9, 0, 1, 2, 3], dtype=torch.int64)
+    k2 = torch.tensor([5, 6], dtype=torch.int64)
+    s2 = torch.tensor([1, 0], dtype=torch.int64)
+    m = MergeReader([frames_of([k1, s1], prefix=2, batch=2),
+                     frames_of([k2, s2], prefix=2, batch=2)], chunk=2)
+    out = read_all(m)
+    rows = list(zip(out.columns[0].tolist(), out.columns[1].tolist()))
+    assert rows == sorted(rows)
+    assert sorted(rows) == sorted(
+        list(zip(k1.tolist(), s1.tolist())) +
+        list(zip(k2.tolist(), s2.tolist())))
+
+
+def test_reduce_reader_combines_across_streams():
+    schema = Schema([torch.int64, torch.int64], 1)
+    a_k = torch.tensor([1, 1, 2, 3], dtype=torch.int64)
+    a_v = torch.tensor([1, 2, 3, 4], dtype=torch.int64)
+    b_k = torch.tensor([2, 3, 3, 9], dtype=torch.int64)
+    b_v = torch.tensor([10, 20, 30, 40], dtype=torch.int64)
+    from bigslice_amd.ops.aggregate import Aggregation
+    r = reduce_reader([frames_of([a_k, a_v], batch=2),
+                       frames_of([b_k, b_v], batch=2)],
+                      schema, Aggregation(["sum"]), chunk=4)
+    out = read_all(r)
+    got = dict(zip(out.columns[0].tolist(), out.columns[1].tolist()))
+    assert got == {1: 3, 2: 13, 3: 54, 9: 40}
+
+
+def test_sortreader_object_keys():
+    keys = ["banana", "apple", "cherry", "apple"]
+    vals = torch.tensor([1, 2, 3, 4], dtype=torch.int64)
+    sr = SortReader(IterReader(iter([Frame([list(keys), vals])])),
+                    run_bytes=10)
+    out = read_all(sr)
+    assert out.columns[0] == ["apple", "apple", "banana", "cherry"]
+
+
+@pytest.mark.gpu
+def test_sortreader_gpu_large():
+    g = torch.Generator(device="cuda:0").manual_seed(1)
+    keys = torch.randint(0, 1 << 40, (5_000_000,), dtype=torch.int64,
+                         device="cuda:0", generator=g)
+    vals = torch.arange(5_000_000, dtype=torch.int64, device="cuda:0")
+    sr = SortReader(frames_of([keys, vals], batch=1_000_000),
+                    run_bytes=20_000_000, device="cuda:0")
+    out = read_all(sr)
+    assert len(out) == 5_000_000
+    ok = out.columns[0]
+    assert bool((ok[1:] >= ok[:-1]).all())
+    assert int(out.columns[1].sum()) == int(vals.sum())
