@@ -21,7 +21,12 @@ from .frame import Frame  # noqa
 from .ops import (Aggregation, Cache, Cogroup, Const, Dep, Filter,  # noqa
                   Flatmap, Fold, Head, Map, Pragma, Prefixed, ReadCache,
                   ReaderFunc, Reduce, Repartition, Reshard, Reshuffle, Scan,
-                  ScanReader, Slice, WriterFunc, schema_of, unwrap)
+                  ScanReader, Slice, WriterFunc, exclusive, materialize,
+                  procs, schema_of, unwrap)
+from . import slicetest  # noqa
+from . import sliceconfig  # noqa
+from . import sortio  # noqa
+from .utils import metrics  # noqa
 from .runtime import (Result, Session, func, registry_digest, start)  # noqa
 from . import sliceio  # noqa
 from . import config  # noqa

@@ -1,6 +1,7 @@
 """Slice operator exports."""
 
-from .slice_base import Dep, Name, Pragma, Slice, TaskContext, unwrap  # noqa
+from .slice_base import (Dep, Name, Pragma, Slice, TaskContext,  # noqa
+                         exclusive, materialize, procs, unwrap)
 from .sources import Const, ReaderFunc, ScanReader  # noqa
 from .elementwise import (Filter, Flatmap, Head, Map, Prefixed, Scan,  # noqa
                           WriterFunc, schema_of)

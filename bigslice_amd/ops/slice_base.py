@@ -78,6 +78,28 @@ class Pragma:
 DEFAULT_PRAGMA = Pragma()
 
 
+def exclusive(slice_: "Slice") -> "Slice":
+    """Mark a slice's tasks Exclusive (slice.go Exclusive pragma)."""
+    slice_.pragma = Pragma(procs=slice_.pragma.procs, exclusive=True,
+                           materialize=slice_.pragma.materialize)
+    return slice_
+
+
+def procs(slice_: "Slice", n: int) -> "Slice":
+    """Reserve n procs per task (slice.go Procs pragma)."""
+    slice_.pragma = Pragma(procs=n, exclusive=slice_.pragma.exclusive,
+                           materialize=slice_.pragma.materialize)
+    return slice_
+
+
+def materialize(slice_: "Slice") -> "Slice":
+    """Break pipelining below this slice (ExperimentalMaterialize)."""
+    slice_.pragma = Pragma(procs=slice_.pragma.procs,
+                           exclusive=slice_.pragma.exclusive,
+                           materialize=True)
+    return slice_
+
+
 class TaskContext:
     """Per-task execution context handed to readers: target device,
     batch size, metrics scope and (on GPU) the HIP stream."""

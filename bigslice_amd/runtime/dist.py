@@ -84,7 +84,7 @@ class DistExecutor(Executor):
         my_tasks = [t for t in phase
                     if owner_rank(t.shard, world) == comm.rank]
         exemplar = phase[0]
-        shuffled = exemplar.num_partitions > 1
+        shuffled = exemplar.shuffle_out
         send: List[List[Tuple[str, int, object]]] = \
             [[] for _ in range(world)]
         err: Optional[BaseException] = None

@@ -36,7 +36,10 @@ class LocalExecutor(Executor):
             device = "cuda:0" if torch.cuda.is_available() else "cpu"
         self.device = device
         self.store = store or MemoryStore()
-        self._sem = threading.Semaphore(self.parallelism)
+        # proc accounting (Pragma Procs/Exclusive, slice.go:109-200):
+        # a task takes pragma.procs permits; Exclusive takes them all.
+        self._avail = self.parallelism
+        self._res_cond = threading.Condition()
         # persistent worker pool reused across evaluations (one thread
         # per concurrent shard task; GPU work overlaps via the HIP
         # stream while threads trade the GIL)
@@ -63,9 +66,13 @@ class LocalExecutor(Executor):
     # -- Executor ---------------------------------------------------------
 
     def run(self, task: Task) -> None:
-        procs = task.pragma.procs if task.pragma else 1
-        acquired = 1
-        self._sem.acquire()
+        pragma = task.pragma
+        want = self.parallelism if (pragma and pragma.exclusive) else \
+            min(pragma.procs if pragma else 1, self.parallelism)
+        want = max(want, 1)
+        with self._res_cond:
+            self._res_cond.wait_for(lambda: self._avail >= want)
+            self._avail -= want
         task.set_state(TaskState.RUNNING)
         try:
             if self.fault_hook is not None:
@@ -95,8 +102,9 @@ class LocalExecutor(Executor):
             e.task_traceback = traceback.format_exc()
             task.set_state(TaskState.ERR, e)
         finally:
-            for _ in range(acquired):
-                self._sem.release()
+            with self._res_cond:
+                self._avail += want
+                self._res_cond.notify_all()
 
     def _run_inner(self, task: Task) -> None:
         ctx = TaskContext(device=self.device)

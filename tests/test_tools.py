@@ -1,0 +1,129 @@
+"""Tools, sliceconfig, tarslice, debug HTTP and pragma tests."""
+
+import io
+import json
+import subprocess
+import sys
+import tarfile
+import urllib.request
+
+import pytest
+import torch
+
+import bigslice_amd as bs
+
+
+def test_slicer_reduce_cpu():
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.slicer", "reduce",
+         "--nshard", "4", "--nkey", "1000", "--device", "cpu"],
+        capture_output=True, text=True)
+    assert p.returncode == 0, p.stderr
+    assert "reduce OK" in p.stdout
+
+
+def test_slicer_cogroup_cpu():
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.slicer", "cogroup",
+         "--nshard", "3", "--nkey", "200", "--device", "cpu"],
+        capture_output=True, text=True)
+    assert p.returncode == 0, p.stderr
+    assert "cogroup OK" in p.stdout
+
+
+def test_slicer_memiter_cpu():
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.slicer", "memiter",
+         "--iters", "3", "--device", "cpu"],
+        capture_output=True, text=True)
+    assert p.returncode == 0, p.stderr
+
+
+def test_badfuncs_late_registration():
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.badfuncs", "late"],
+        capture_output=True, text=True)
+    assert p.returncode == 0, p.stderr
+    assert "caught expected" in p.stdout
+
+
+def test_wordcount_tool(tmp_path):
+    f = tmp_path / "in.txt"
+    f.write_text("a b a\nc a\n")
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.wordcount", str(f),
+         "--shards", "2", "--local"],
+        capture_output=True, text=True)
+    assert p.returncode == 0, p.stderr
+    assert p.stdout.splitlines()[0].split() == ["3", "a"]
+
+
+def test_sliceconfig_profile(tmp_path):
+    prof = tmp_path / "config"
+    prof.write_text("parallelism = 3\ndevice = cpu\n")
+    sess, rest = bs.sliceconfig.parse(
+        ["--extra", "1"], profile_path=str(prof))
+    assert sess.executor.parallelism == 3
+    assert rest == ["--extra", "1"]
+
+
+def test_tar_reader(tmp_path):
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w") as tf:
+        for i in range(5):
+            data = f"content{i}".encode()
+            info = tarfile.TarInfo(name=f"f{i}.txt")
+            info.size = len(data)
+            tf.addfile(info, io.BytesIO(data))
+    raw = buf.getvalue()
+
+    from bigslice_amd.ops.archive import TarReader
+    fv = bs.func(lambda: TarReader(2, lambda: io.BytesIO(raw)))
+    res = bs.start(parallelism=2, device="cpu").run(fv)
+    rows = sorted(res.scan())
+    assert [r[0] for r in rows] == [f"f{i}.txt" for i in range(5)]
+    assert rows[0][1] == b"content0"
+
+
+def test_debug_http_endpoints():
+    from bigslice_amd.utils.debug_http import serve_session
+    sess = bs.start(parallelism=2, device="cpu")
+    server = serve_session(sess, 0)
+    port = server.server_address[1]
+    fv = bs.func(lambda: bs.Const(2, torch.arange(4, dtype=torch.int64)))
+    sess.run(fv)
+    with urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/debug/tasks") as r:
+        tasks = json.loads(r.read())
+    assert any("const" in t["task"] for t in tasks)
+    with urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/debug/tasks/graph") as r:
+        g = json.loads(r.read())
+    assert g["nodes"]
+    server.shutdown()
+
+
+def test_exclusive_pragma_serializes():
+    import threading
+    active = []
+    lock = threading.Lock()
+    peak = [0]
+
+    def track(x):
+        with lock:
+            active.append(1)
+            peak[0] = max(peak[0], len(active))
+        import time
+        time.sleep(0.02)
+        with lock:
+            active.pop()
+        return (x,)
+
+    def build():
+        s = bs.Map(bs.Const(4, torch.arange(8, dtype=torch.int64)),
+                   track, out_schema=(int,))
+        return bs.exclusive(s)
+
+    res = bs.slicetest.run(build, parallelism=4)
+    assert len(list(res.scan())) == 8
+    assert peak[0] == 1  # exclusive tasks never overlap
