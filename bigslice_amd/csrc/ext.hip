@@ -233,6 +233,57 @@ void groupby_insert(torch::Tensor keys, std::vector<torch::Tensor> vals,
   HIP_CHECK(hipGetLastError());
 }
 
+// Packed-slot fast path (single int64 SUM value): see groupby.hip.
+void groupby_insert_packed(torch::Tensor keys, torch::Tensor vals,
+                           torch::Tensor table, torch::Tensor flags,
+                           int64_t max_probes) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
+  TORCH_CHECK(vals.scalar_type() == torch::kInt64);
+  keys = keys.contiguous();
+  vals = vals.contiguous();
+  int64_t n = keys.size(0);
+  if (n == 0) return;
+  int64_t cap = table.size(0) / 2 - 1;
+  TORCH_CHECK(cap > 0 && (cap & (cap - 1)) == 0);
+  const uint32_t seed = 0x9acb0442u;
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
+  hipLaunchKernelGGL(k_groupby_insert_packed_sum_i64, dim3(blocks),
+                     dim3(THREADS), 0, current_stream(),
+                     keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(), n,
+                     table.data_ptr<int64_t>(), cap, seed,
+                     flags.data_ptr<int32_t>(),
+                     flags.data_ptr<int32_t>() + 1, max_probes);
+  HIP_CHECK(hipGetLastError());
+}
+
+std::vector<torch::Tensor> groupby_compact_packed(torch::Tensor table,
+                                                  torch::Tensor cursor) {
+  int64_t cap = table.size(0) / 2 - 1;
+  auto out_keys = torch::empty({cap}, table.options());
+  auto out_vals = torch::empty({cap}, table.options());
+  int64_t spb = (cap + 2047) / 2048;
+  if (spb < THREADS) spb = THREADS;
+  int blocks = (int)((cap + spb - 1) / spb);
+  hipLaunchKernelGGL(k_groupby_compact_packed, dim3(blocks), dim3(THREADS),
+                     0, current_stream(), table.data_ptr<int64_t>(), cap,
+                     spb, out_keys.data_ptr<int64_t>(),
+                     out_vals.data_ptr<int64_t>(),
+                     (unsigned long long*)cursor.data_ptr());
+  HIP_CHECK(hipGetLastError());
+  return {out_keys, out_vals};
+}
+
+torch::Tensor alloc_packed_table(int64_t cap, torch::Tensor like) {
+  auto table = torch::empty({2 * (cap + 1)}, like.options());
+  int64_t nslots = cap + 1;
+  int blocks = (int)std::min<int64_t>((nslots + THREADS - 1) / THREADS,
+                                      16384);
+  hipLaunchKernelGGL(k_fill_packed_slots, dim3(blocks), dim3(THREADS), 0,
+                     current_stream(), table.data_ptr<int64_t>(), nslots);
+  HIP_CHECK(hipGetLastError());
+  return table;
+}
+
 // Compact used slots into freshly-allocated output arrays; returns
 // [out_keys(cap), out_vals...(cap)]; cursor (u64[1], zeroed by caller)
 // receives the used-slot count — the caller narrows after reading it.
@@ -302,6 +353,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "scatter rows by precomputed partition ids");
   m.def("groupby_insert", &groupby_insert,
         "hash-aggregate insert pass (K9)");
+  m.def("groupby_insert_packed", &groupby_insert_packed,
+        "packed-slot insert (int64 sum fast path)");
+  m.def("groupby_compact_packed", &groupby_compact_packed,
+        "packed-slot compaction");
+  m.def("alloc_packed_table", &alloc_packed_table,
+        "allocate+init packed table");
   m.def("groupby_compact", &groupby_compact,
         "hash-aggregate table compaction (K9)");
   m.def("agg_identity", &agg_identity, "aggregation identity fill");

@@ -147,6 +147,91 @@ extern "C" __global__ void k_groupby_compact(
   }
 }
 
+// Packed-slot variant for the hottest shape (single int64 value, SUM):
+// slot i = table[2i]=key, table[2i+1]=sum.  Key and accumulator share a
+// 16-byte-aligned pair, so each row touches ONE cache line instead of
+// two — the insert is atomic-latency-bound, so halving touched lines
+// matters.  Table size 2*(cap+1); extra slot for the sentinel key.
+extern "C" __global__ void k_groupby_insert_packed_sum_i64(
+    const int64_t* keys, const int64_t* vals, int64_t n, int64_t* table,
+    int64_t cap, uint32_t seed, int32_t* sentinel_seen, int32_t* overflow,
+    int64_t max_probes) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t mask = (uint64_t)cap - 1;
+  for (; i < n; i += stride) {
+    int64_t k = keys[i];
+    int64_t slot;
+    if (k == GB_SENTINEL) {
+      atomicOr(sentinel_seen, 1);
+      slot = cap;
+    } else {
+      uint64_t h = mm3_u64((uint64_t)k, seed) & mask;
+      int64_t probes = 0;
+      for (;;) {
+        long long cur = ((volatile long long*)table)[2 * h];
+        if (cur == k) break;
+        if (cur == GB_SENTINEL) {
+          long long prev = atomicCAS((unsigned long long*)&table[2 * h],
+                                     (unsigned long long)GB_SENTINEL,
+                                     (unsigned long long)k);
+          if (prev == GB_SENTINEL || prev == k) break;
+        }
+        h = (h + 1) & mask;
+        if (++probes >= max_probes) {
+          atomicOr(overflow, 1);
+          return;
+        }
+      }
+      slot = (int64_t)h;
+    }
+    atomicAdd((unsigned long long*)&table[2 * slot + 1],
+              (unsigned long long)vals[i]);
+  }
+}
+
+extern "C" __global__ void k_groupby_compact_packed(
+    const int64_t* table, int64_t cap, int64_t slots_per_block,
+    int64_t* out_keys, int64_t* out_vals, unsigned long long* cursor) {
+  __shared__ uint32_t counts[THREADS];
+  __shared__ unsigned long long block_base;
+  int64_t start = (int64_t)blockIdx.x * slots_per_block;
+  int64_t end = min(start + slots_per_block, cap);
+  uint32_t mine = 0;
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    mine += (table[2 * i] != GB_SENTINEL);
+  counts[threadIdx.x] = mine;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint32_t run = 0;
+    for (int t = 0; t < (int)blockDim.x; ++t) {
+      uint32_t c = counts[t];
+      counts[t] = run;
+      run += c;
+    }
+    block_base = run ? atomicAdd(cursor, (unsigned long long)run) : 0;
+  }
+  __syncthreads();
+  int64_t pos = (int64_t)block_base + counts[threadIdx.x];
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+    int64_t k = table[2 * i];
+    if (k == GB_SENTINEL) continue;
+    out_keys[pos] = k;
+    out_vals[pos] = table[2 * i + 1];
+    ++pos;
+  }
+}
+
+extern "C" __global__ void k_fill_packed_slots(int64_t* table,
+                                               int64_t nslots) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < nslots; i += stride) {
+    table[2 * i] = GB_SENTINEL;
+    table[2 * i + 1] = 0;
+  }
+}
+
 extern "C" __global__ void k_groupby_insert(
     const int64_t* keys, int64_t n, ValCols vals, int64_t* tkeys,
     int64_t cap, uint32_t seed, int32_t* sentinel_seen, int32_t* overflow,

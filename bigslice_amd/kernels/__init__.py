@@ -125,9 +125,24 @@ class GroupTable:
         self.batches: List = []  # retained for overflow re-insert
         self.rows = 0
 
+    @property
+    def _packed(self) -> bool:
+        """Packed key+sum slots for the hottest shape: one cache line
+        touched per row instead of two."""
+        import os
+        return (len(self.val_dtypes) == 1
+                and self.val_dtypes[0] == torch.int64
+                and self.aggs == ["sum"]
+                and os.environ.get("BIGSLICE_GB_PACKED", "1") == "1")
+
     def _alloc(self, cap: int):
         self.cap = cap
         dev = self.device
+        if self._packed:
+            like = torch.empty(0, dtype=torch.int64, device=dev)
+            self.table = _C.alloc_packed_table(cap, like)
+            self.flags = torch.zeros(2, dtype=torch.int32, device=dev)
+            return
         self.tkeys = torch.full((cap + 1,), GB_SENTINEL,
                                 dtype=torch.int64, device=dev)
         self.tabs = [
@@ -147,8 +162,12 @@ class GroupTable:
                 2 * n, getattr(config, "GROUPBY_INITIAL_CAP", 1 << 23))
             self._alloc(_next_pow2(hint))
         self.batches.append((keys, vals))
-        _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
-                          self.tabs, self.flags, MAX_PROBES)
+        if self._packed:
+            _C.groupby_insert_packed(keys, vals[0], self.table,
+                                     self.flags, MAX_PROBES)
+        else:
+            _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
+                              self.tabs, self.flags, MAX_PROBES)
 
     def finish(self):
         """Returns (keys, [vals]) of the aggregated groups."""
@@ -158,7 +177,10 @@ class GroupTable:
                            for dt in self.val_dtypes]
         while True:
             cursor = torch.zeros(1, dtype=torch.int64, device=self.device)
-            outs = _C.groupby_compact(self.tkeys, self.tabs, cursor)
+            if self._packed:
+                outs = _C.groupby_compact_packed(self.table, cursor)
+            else:
+                outs = _C.groupby_compact(self.tkeys, self.tabs, cursor)
             host = torch.cat([cursor,
                               self.flags.to(torch.int64)]).cpu()
             nkeys, sentinel_seen, overflow = (int(host[0]), int(host[1]),
@@ -184,8 +206,13 @@ class GroupTable:
                 keys = torch.cat([keys, torch.full(
                     (1,), GB_SENTINEL, dtype=torch.int64,
                     device=self.device)])
-                vals = [torch.cat([v, t[self.cap:self.cap + 1]])
-                        for v, t in zip(vals, self.tabs)]
+                if self._packed:
+                    c = self.cap
+                    vals = [torch.cat([vals[0],
+                                       self.table[2 * c + 1:2 * c + 2]])]
+                else:
+                    vals = [torch.cat([v, t[self.cap:self.cap + 1]])
+                            for v, t in zip(vals, self.tabs)]
             self.batches = []
             return keys, vals
 

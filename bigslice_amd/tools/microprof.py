@@ -1,0 +1,88 @@
+"""Kernel micro-benchmarks (for rocprofv3/PMC runs and A/B timing).
+
+  python -m bigslice_amd.tools.microprof groupby --rows 125000000
+  python -m bigslice_amd.tools.microprof partition --rows 125000000
+  python -m bigslice_amd.tools.microprof hash --rows 125000000
+
+Times each kernel phase with hipEvents over --iters iterations.  Keep
+runs short: this is the target for `rocprofv3 --pmc ... -- ...`.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+
+import torch
+
+from bigslice_amd import kernels
+from bigslice_amd.frame import Frame
+
+
+def timeit(fn, iters):
+    s = torch.cuda.Event(enable_timing=True)
+    e = torch.cuda.Event(enable_timing=True)
+    fn()  # warmup
+    torch.cuda.synchronize()
+    s.record()
+    for _ in range(iters):
+        fn()
+    e.record()
+    torch.cuda.synchronize()
+    return s.elapsed_time(e) / iters
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("which", choices=["groupby", "partition", "hash",
+                                      "compact"])
+    ap.add_argument("--rows", type=int, default=125_000_000)
+    ap.add_argument("--nkeys", type=int, default=1_000_000)
+    ap.add_argument("--nparts", type=int, default=8)
+    ap.add_argument("--iters", type=int, default=3)
+    args = ap.parse_args()
+    assert torch.cuda.is_available() and kernels.have_extension()
+    dev = "cuda:0"
+    g = torch.Generator(device=dev)
+    g.manual_seed(1)
+    keys = torch.randint(0, args.nkeys, (args.rows,), dtype=torch.int64,
+                         device=dev, generator=g)
+    vals = torch.ones(args.rows, dtype=torch.int64, device=dev)
+
+    out = {"which": args.which, "rows": args.rows, "nkeys": args.nkeys,
+           "packed": os.environ.get("BIGSLICE_GB_PACKED", "1")}
+    if args.which == "groupby":
+        def run():
+            t = kernels.GroupTable([torch.int64], ["sum"], dev)
+            t.insert(keys, [vals])
+            t.finish()
+        ms = timeit(run, args.iters)
+    elif args.which == "compact":
+        t = kernels.GroupTable([torch.int64], ["sum"], dev)
+        t.insert(keys, [vals])
+
+        def run():
+            cursor = torch.zeros(1, dtype=torch.int64, device=dev)
+            if t._packed:
+                kernels._C.groupby_compact_packed(t.table, cursor)
+            else:
+                kernels._C.groupby_compact(t.tkeys, t.tabs, cursor)
+        ms = timeit(run, args.iters)
+    elif args.which == "partition":
+        f = Frame([keys, vals], prefix=1)
+
+        def run():
+            kernels.partition_frame(f, args.nparts, None)
+        ms = timeit(run, args.iters)
+    else:
+        def run():
+            kernels.hash_columns_device([keys], 0)
+        ms = timeit(run, args.iters)
+    out["ms"] = ms
+    out["grows_per_sec"] = args.rows / ms / 1e6
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
