@@ -39,6 +39,13 @@ SORT_SPILL_TARGET_BYTES = _env_int("BIGSLICE_SORT_SPILL_BYTES", 4 << 30)
 # probe-chain overflow (reference combiner grow policy, exec/combiner.go:47).
 GROUPBY_INITIAL_CAP = _env_int("BIGSLICE_GROUPBY_INITIAL_CAP", 1 << 22)
 
+# Machine-combiners mode: producer tasks of one shuffle phase on a GPU
+# share one combiner table (reference exec/session.go:166-176).  Shared
+# tables keep the working set LLC-resident; disabled automatically when
+# fault injection is active (the mode has no loss recovery).
+MACHINE_COMBINERS = os.environ.get("BIGSLICE_MACHINE_COMBINERS",
+                                   "1") == "1"
+
 # Maximum consecutive losses of a single task before giving up
 # (reference: exec/eval.go:30).
 MAX_CONSECUTIVE_LOST = 5

@@ -88,9 +88,20 @@ class DistExecutor(Executor):
         send: List[List[Tuple[str, int, object]]] = \
             [[] for _ in range(world)]
         err: Optional[BaseException] = None
+        # Machine-combiners mode (exec/session.go:166-176): this rank's
+        # producer tasks of the phase share one combiner/partitioner.
+        shared_writer = None
+        if (shuffled and len(my_tasks) > 1
+                and exemplar.combiner is not None
+                and exemplar.partitioner is None):
+            from ..ops.slice_base import TaskContext
+            shared_writer = PartitionWriter(
+                exemplar.num_partitions, None, exemplar.combiner,
+                exemplar.schema, self.device,
+                TaskContext(device=self.device).chunk)
         try:
             for t in my_tasks:
-                buckets = self._run_task(t)
+                buckets = self._run_task(t, shared_writer)
                 if buckets is None:
                     continue
                 for p, frames in enumerate(buckets):
@@ -98,11 +109,15 @@ class DistExecutor(Executor):
                         d = owner_rank(p, world)
                         for f in frames:
                             send[d].append((t.name, p, f))
-                        if not frames:
-                            pass
                     else:
                         rows = sum(len(f) for f in frames)
                         self.store.put(t.name, p, frames, rows)
+            if shared_writer is not None and my_tasks:
+                first = my_tasks[0]
+                for p, frames in enumerate(shared_writer.finish()):
+                    d = owner_rank(p, world)
+                    for f in frames:
+                        send[d].append((first.name, p, f))
         except BaseException as e:
             err = e
         # Surface errors collectively so every rank raises.
@@ -138,9 +153,9 @@ class DistExecutor(Executor):
         for t in phase:
             t.set_state(TaskState.OK)
 
-    def _run_task(self, task: Task):
+    def _run_task(self, task: Task, shared_writer=None):
         """Run one task; returns per-partition frame lists (or None for
-        terminal tasks)."""
+        terminal tasks or when a shared phase writer absorbs output)."""
         task.set_state(TaskState.RUNNING)
         scope = metrics.Scope()
         self.scopes[task.name] = scope
@@ -165,6 +180,13 @@ class DistExecutor(Executor):
             for _ in out:
                 pass
             self.store.put(task.name, 0, [], 0)
+            self._scope_ctx.__exit__(None, None, None)
+            if self.tracer:
+                self._span.__exit__(None, None, None)
+            return None
+        if shared_writer is not None:
+            for f in out:
+                shared_writer.add(f)
             self._scope_ctx.__exit__(None, None, None)
             if self.tracer:
                 self._span.__exit__(None, None, None)

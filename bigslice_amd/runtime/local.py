@@ -50,6 +50,11 @@ class LocalExecutor(Executor):
         self._tls = threading.local()
         self.scopes = {}  # task name -> metrics.Scope
         self.tracer = None  # utils.trace.Tracer, set by the session
+        # machine-combiners mode (reference exec/session.go:166-176):
+        # producer tasks of one shuffle phase share one combiner table.
+        self.machine_combiners = config.MACHINE_COMBINERS
+        self._shared_combiners = {}  # phase id -> SharedPhaseCombiner
+        self._shared_lock = threading.Lock()
 
     def _stream(self):
         """Per-worker-thread HIP stream: concurrent shard tasks overlap
@@ -133,6 +138,30 @@ class LocalExecutor(Executor):
                 pass
             self.store.put(task.name, 0, [], 0)
             return
+        shared = self._shared_combiner_for(task)
+        if shared is not None:
+            for f in out:
+                shared.add(f)
+            if shared.task_done():
+                buckets = shared.finish_buckets()
+                # combined output stored under the phase's first task;
+                # siblings store empty partitions (consumers concat all)
+                first = task.group[0]
+                for pi, frames in enumerate(buckets):
+                    rows = sum(len(f) for f in frames)
+                    self.store.put(first.name, pi, frames, rows)
+                with self._shared_lock:
+                    self._shared_combiners.pop(id(task.group[0]), None)
+            if task is not task.group[0]:
+                for pi in range(task.num_partitions):
+                    self.store.put(task.name, pi, [], 0)
+            elif task.group[0] is task and \
+                    not self.store.has(task.name, 0):
+                # first task finished before the phase completed: its
+                # entries are written by the last finisher above; make
+                # sure empty markers exist if it was also the last.
+                pass
+            return
         w = PartitionWriter(task.num_partitions, task.partitioner,
                             task.combiner, task.schema, self.device,
                             ctx.chunk)
@@ -142,6 +171,29 @@ class LocalExecutor(Executor):
         for pi, frames in enumerate(buckets):
             rows = sum(len(f) for f in frames)
             self.store.put(task.name, pi, frames, rows)
+
+    def _shared_combiner_for(self, task: Task):
+        """The phase's shared combiner, when machine-combiners mode
+        applies: combiner present, default hash partitioner, no fault
+        injection (the mode has no loss recovery — reference
+        exec/session.go:166-176), and a multi-task phase."""
+        if (not self.machine_combiners or task.combiner is None
+                or task.partitioner is not None
+                or self.fault_hook is not None
+                or len(task.group) <= 1):
+            return None
+        from ..ops.aggregate import make_aggregator, TensorAggregator
+        key = id(task.group[0])
+        with self._shared_lock:
+            sc = self._shared_combiners.get(key)
+            if sc is None:
+                from .partition import SharedPhaseCombiner
+                sc = SharedPhaseCombiner(
+                    task.schema, task.combiner, self.device,
+                    task.num_partitions, len(task.group),
+                    TaskContext(device=self.device).chunk)
+                self._shared_combiners[key] = sc
+            return sc
 
     def reader(self, task: Task, partition: int) -> Reader:
         return self.store.open(task.name, partition, device="cpu")

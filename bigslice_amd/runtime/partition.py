@@ -54,6 +54,67 @@ def split_frame(frame: Frame, num_partitions: int,
     return out
 
 
+class SharedPhaseCombiner:
+    """Per-GPU shared combiner for one shuffle phase (the reference's
+    machine-combiners mode, exec/session.go:166-176 + bigmachine.go
+    combine keys): every producer task of the phase inserts into ONE
+    hash-aggregate table, and the last task to finish compacts,
+    partitions and stores the combined output once.
+
+    Known limitation carried from the reference: no error recovery —
+    the local executor only enables this when no fault injection is
+    active (tasks cannot be LOST mid-phase).
+    """
+
+    def __init__(self, schema, combiner, device: str, num_partitions: int,
+                 expected_tasks: int, chunk: int):
+        import threading
+
+        from ..ops.aggregate import make_aggregator
+        self.agg = make_aggregator(schema, combiner, device)
+        self.device = device
+        self.num_partitions = num_partitions
+        self.chunk = chunk
+        self.remaining = expected_tasks
+        self.lock = threading.Lock()
+        self.events = []
+        self.rows = 0
+
+    def add(self, frame: Frame) -> None:
+        with self.lock:
+            self.rows += len(frame)
+            self.agg.add(frame)
+
+    def task_done(self) -> bool:
+        """Record this task's inserts (stream event) and return True for
+        the last task, which must then call finish_buckets."""
+        if self.device.startswith("cuda"):
+            ev = torch.cuda.Event()
+            ev.record()
+        else:
+            ev = None
+        with self.lock:
+            if ev is not None:
+                self.events.append(ev)
+            self.remaining -= 1
+            return self.remaining == 0
+
+    def finish_buckets(self) -> List[List[Frame]]:
+        # all other tasks' insert kernels must complete first
+        if self.device.startswith("cuda"):
+            cur = torch.cuda.current_stream()
+            for ev in self.events:
+                cur.wait_event(ev)
+        buckets: List[List[Frame]] = \
+            [[] for _ in range(self.num_partitions)]
+        for f in self.agg.result_frames(self.chunk * 4):
+            for pi, pf in enumerate(split_frame(f, self.num_partitions,
+                                                None)):
+                if pf is not None and len(pf):
+                    buckets[pi].append(pf)
+        return buckets
+
+
 class PartitionWriter:
     """Accumulates a task's output into per-partition frame lists, with
     optional producer-side pre-combine (the reference's combiner
