@@ -233,6 +233,31 @@ void groupby_insert(torch::Tensor keys, std::vector<torch::Tensor> vals,
   HIP_CHECK(hipGetLastError());
 }
 
+// Two-level LDS insert (single int64 SUM value): see groupby.hip.
+void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
+                        torch::Tensor tkeys, torch::Tensor tab,
+                        torch::Tensor flags, int64_t max_probes) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
+  TORCH_CHECK(vals.scalar_type() == torch::kInt64);
+  keys = keys.contiguous();
+  vals = vals.contiguous();
+  int64_t n = keys.size(0);
+  if (n == 0) return;
+  int64_t cap = tkeys.size(0) - 1;
+  TORCH_CHECK(cap > 0 && (cap & (cap - 1)) == 0);
+  const uint32_t seed = 0x9acb0442u;
+  int64_t rpb = rows_per_block(n);
+  int64_t nblocks = (n + rpb - 1) / rpb;
+  hipLaunchKernelGGL(k_groupby_insert_sum_i64_lds, dim3((uint32_t)nblocks),
+                     dim3(THREADS), 0, current_stream(),
+                     keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(), n,
+                     tkeys.data_ptr<int64_t>(),
+                     (long long*)tab.data_ptr<int64_t>(), cap, seed,
+                     flags.data_ptr<int32_t>(),
+                     flags.data_ptr<int32_t>() + 1, max_probes, rpb);
+  HIP_CHECK(hipGetLastError());
+}
+
 // Packed-slot fast path (single int64 SUM value): see groupby.hip.
 void groupby_insert_packed(torch::Tensor keys, torch::Tensor vals,
                            torch::Tensor table, torch::Tensor flags,
@@ -353,6 +378,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "scatter rows by precomputed partition ids");
   m.def("groupby_insert", &groupby_insert,
         "hash-aggregate insert pass (K9)");
+  m.def("groupby_insert_lds", &groupby_insert_lds,
+        "two-level LDS+global insert (int64 sum)");
   m.def("groupby_insert_packed", &groupby_insert_packed,
         "packed-slot insert (int64 sum fast path)");
   m.def("groupby_compact_packed", &groupby_compact_packed,

@@ -126,14 +126,26 @@ class GroupTable:
         self.rows = 0
 
     @property
-    def _packed(self) -> bool:
-        """Packed key+sum slots for the hottest shape: one cache line
-        touched per row instead of two."""
-        import os
+    def _sum_i64(self) -> bool:
         return (len(self.val_dtypes) == 1
                 and self.val_dtypes[0] == torch.int64
-                and self.aggs == ["sum"]
-                and os.environ.get("BIGSLICE_GB_PACKED", "1") == "1")
+                and self.aggs == ["sum"])
+
+    @property
+    def _packed(self) -> bool:
+        """Packed key+sum slots (A/B'd neutral-to-slightly-worse than
+        split arrays; off by default, kept for experiments)."""
+        import os
+        return (self._sum_i64
+                and os.environ.get("BIGSLICE_GB_PACKED", "0") == "1")
+
+    @property
+    def _lds(self) -> bool:
+        """Two-level LDS pre-aggregation (K10): absorbs hot keys on-CU.
+        Default on for the int64-sum shape."""
+        import os
+        return (self._sum_i64 and not self._packed
+                and os.environ.get("BIGSLICE_GB_LDS", "1") == "1")
 
     def _alloc(self, cap: int):
         self.cap = cap
@@ -165,6 +177,9 @@ class GroupTable:
         if self._packed:
             _C.groupby_insert_packed(keys, vals[0], self.table,
                                      self.flags, MAX_PROBES)
+        elif self._lds:
+            _C.groupby_insert_lds(keys, vals[0], self.tkeys,
+                                  self.tabs[0], self.flags, MAX_PROBES)
         else:
             _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
                               self.tabs, self.flags, MAX_PROBES)
