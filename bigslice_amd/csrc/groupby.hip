@@ -188,46 +188,72 @@ __device__ __forceinline__ void gb_global_insert_sum(
   atomicAdd((unsigned long long*)&tab[slot], (unsigned long long)v);
 }
 
+// One row through the LDS table; returns false on LDS-table miss
+// (caller falls back to the global table).  hits counts rows absorbed
+// by an EXISTING LDS entry (the duplicate-rate signal).
+__device__ __forceinline__ bool gb_lds_try(
+    int64_t k, long long v, long long* lk, long long* lv, uint32_t seed,
+    uint32_t* hits) {
+  uint32_t h = mm3_u64((uint64_t)k, seed) & (GB_LDS_SLOTS - 1);
+  for (int p = 0; p < 4; ++p) {  // short LDS probe chain
+    long long cur = lk[h];
+    bool existed = true;
+    if (cur == GB_SENTINEL) {
+      long long prev = atomicCAS((unsigned long long*)&lk[h],
+                                 (unsigned long long)GB_SENTINEL,
+                                 (unsigned long long)k);
+      existed = (prev != GB_SENTINEL);
+      cur = existed ? prev : k;
+    }
+    if (cur == k) {
+      atomicAdd((unsigned long long*)&lv[h], (unsigned long long)v);
+      if (existed && hits) atomicAdd(hits, 1u);
+      return true;
+    }
+    h = (h + 1) & (GB_LDS_SLOTS - 1);
+  }
+  return false;
+}
+
 extern "C" __global__ void k_groupby_insert_sum_i64_lds(
     const int64_t* keys, const int64_t* vals, int64_t n, int64_t* tkeys,
     long long* tab, int64_t cap, uint32_t seed, int32_t* sentinel_seen,
     int32_t* overflow, int64_t max_probes, int64_t rows_per_block) {
   __shared__ long long lk[GB_LDS_SLOTS];
   __shared__ long long lv[GB_LDS_SLOTS];
+  __shared__ uint32_t lds_hits;
   for (int i = threadIdx.x; i < GB_LDS_SLOTS; i += blockDim.x) {
     lk[i] = GB_SENTINEL;
     lv[i] = 0;
   }
+  if (threadIdx.x == 0) lds_hits = 0;
   __syncthreads();
   uint64_t gmask = (uint64_t)cap - 1;
   int64_t start = (int64_t)blockIdx.x * rows_per_block;
   int64_t end = min(start + rows_per_block, n);
-  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+
+  // Phase 1 (sample): first rows go through the LDS table, counting how
+  // many were absorbed by existing entries.
+  const int64_t SAMPLE = 4 * GB_LDS_SLOTS;
+  int64_t mid = min(start + SAMPLE, end);
+  for (int64_t i = start + threadIdx.x; i < mid; i += blockDim.x) {
     int64_t k = keys[i];
     long long v = (long long)vals[i];
-    if (k == GB_SENTINEL) {
+    if (k == GB_SENTINEL ||
+        !gb_lds_try(k, v, lk, lv, seed, &lds_hits))
       gb_global_insert_sum(k, v, tkeys, tab, gmask, cap, seed,
                            sentinel_seen, overflow, max_probes);
-      continue;
-    }
-    uint32_t h = mm3_u64((uint64_t)k, seed) & (GB_LDS_SLOTS - 1);
-    bool done = false;
-    for (int p = 0; p < 4; ++p) {  // short LDS probe chain
-      long long cur = lk[h];
-      if (cur == GB_SENTINEL) {
-        long long prev = atomicCAS((unsigned long long*)&lk[h],
-                                   (unsigned long long)GB_SENTINEL,
-                                   (unsigned long long)k);
-        cur = (prev == GB_SENTINEL) ? k : prev;
-      }
-      if (cur == k) {
-        atomicAdd((unsigned long long*)&lv[h], (unsigned long long)v);
-        done = true;
-        break;
-      }
-      h = (h + 1) & (GB_LDS_SLOTS - 1);
-    }
-    if (!done)
+  }
+  __syncthreads();
+  // Duplicate-rate decision: keep the LDS tier only when >=1/8 of the
+  // sampled rows hit an existing entry — otherwise the key space is too
+  // wide for a 2048-slot table and probing it just costs latency.
+  bool use_lds = lds_hits * 8u >= (uint32_t)(mid - start);
+  for (int64_t i = mid + threadIdx.x; i < end; i += blockDim.x) {
+    int64_t k = keys[i];
+    long long v = (long long)vals[i];
+    if (k == GB_SENTINEL ||
+        !(use_lds && gb_lds_try(k, v, lk, lv, seed, nullptr)))
       gb_global_insert_sum(k, v, tkeys, tab, gmask, cap, seed,
                            sentinel_seen, overflow, max_probes);
   }
