@@ -236,7 +236,8 @@ void groupby_insert(torch::Tensor keys, std::vector<torch::Tensor> vals,
 // Two-level LDS insert (single int64 SUM value): see groupby.hip.
 void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
                         torch::Tensor tkeys, torch::Tensor tab,
-                        torch::Tensor flags, int64_t max_probes) {
+                        torch::Tensor flags, int64_t max_probes,
+                        int64_t force) {
   TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
   TORCH_CHECK(vals.scalar_type() == torch::kInt64);
   keys = keys.contiguous();
@@ -254,7 +255,8 @@ void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
                      tkeys.data_ptr<int64_t>(),
                      (long long*)tab.data_ptr<int64_t>(), cap, seed,
                      flags.data_ptr<int32_t>(),
-                     flags.data_ptr<int32_t>() + 1, max_probes, rpb);
+                     flags.data_ptr<int32_t>() + 1, max_probes, rpb,
+                     (int32_t)force);
   HIP_CHECK(hipGetLastError());
 }
 

@@ -218,7 +218,8 @@ __device__ __forceinline__ bool gb_lds_try(
 extern "C" __global__ void k_groupby_insert_sum_i64_lds(
     const int64_t* keys, const int64_t* vals, int64_t n, int64_t* tkeys,
     long long* tab, int64_t cap, uint32_t seed, int32_t* sentinel_seen,
-    int32_t* overflow, int64_t max_probes, int64_t rows_per_block) {
+    int32_t* overflow, int64_t max_probes, int64_t rows_per_block,
+    int32_t force) {
   __shared__ long long lk[GB_LDS_SLOTS];
   __shared__ long long lv[GB_LDS_SLOTS];
   __shared__ uint32_t lds_hits;
@@ -233,9 +234,10 @@ extern "C" __global__ void k_groupby_insert_sum_i64_lds(
   int64_t end = min(start + rows_per_block, n);
 
   // Phase 1 (sample): first rows go through the LDS table, counting how
-  // many were absorbed by existing entries.
+  // many were absorbed by existing entries.  force: -1 = sample-adaptive,
+  // 0 = global-only, 1 = always-LDS (A/B knobs).
   const int64_t SAMPLE = 4 * GB_LDS_SLOTS;
-  int64_t mid = min(start + SAMPLE, end);
+  int64_t mid = (force == -1) ? min(start + SAMPLE, end) : start;
   for (int64_t i = start + threadIdx.x; i < mid; i += blockDim.x) {
     int64_t k = keys[i];
     long long v = (long long)vals[i];
@@ -248,7 +250,9 @@ extern "C" __global__ void k_groupby_insert_sum_i64_lds(
   // Duplicate-rate decision: keep the LDS tier only when >=1/8 of the
   // sampled rows hit an existing entry — otherwise the key space is too
   // wide for a 2048-slot table and probing it just costs latency.
-  bool use_lds = lds_hits * 8u >= (uint32_t)(mid - start);
+  bool use_lds = (force == -1)
+      ? (lds_hits * 8u >= (uint32_t)(mid - start))
+      : (force == 1);
   for (int64_t i = mid + threadIdx.x; i < end; i += blockDim.x) {
     int64_t k = keys[i];
     long long v = (long long)vals[i];

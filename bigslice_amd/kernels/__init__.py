@@ -178,8 +178,10 @@ class GroupTable:
             _C.groupby_insert_packed(keys, vals[0], self.table,
                                      self.flags, MAX_PROBES)
         elif self._lds:
+            force = int(os.environ.get("BIGSLICE_GB_LDS_FORCE", "-1"))
             _C.groupby_insert_lds(keys, vals[0], self.tkeys,
-                                  self.tabs[0], self.flags, MAX_PROBES)
+                                  self.tabs[0], self.flags, MAX_PROBES,
+                                  force)
         else:
             _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
                               self.tabs, self.flags, MAX_PROBES)
