@@ -149,13 +149,15 @@ extern "C" __global__ void k_groupby_compact(
 
 // Two-level insert for the single-int64-SUM shape (reference K10,
 // exec/bigmachine.go:1144-1199 two-level combine): each workgroup first
-// aggregates into a 2048-slot LDS table, absorbing hot keys entirely
+// aggregates into a 1024-slot LDS table (16 KiB: small enough to keep
+// full 32-wave/CU occupancy for the global-atomic path), absorbing hot
+// keys entirely
 // on-CU (LDS atomics), and only LDS-table misses and the end-of-block
 // flush touch the global table.  At low key cardinality this removes
 // the global atomic contention that serializes the one-level kernel
 // (measured 4x); at high cardinality the LDS probe adds a small
 // constant.
-#define GB_LDS_SLOTS 2048
+#define GB_LDS_SLOTS 1024
 
 __device__ __forceinline__ void gb_global_insert_sum(
     int64_t k, long long v, int64_t* tkeys, long long* tab, uint64_t mask,
