@@ -237,7 +237,7 @@ void groupby_insert(torch::Tensor keys, std::vector<torch::Tensor> vals,
 void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
                         torch::Tensor tkeys, torch::Tensor tab,
                         torch::Tensor flags, int64_t max_probes,
-                        int64_t force) {
+                        int64_t force, int64_t target_blocks) {
   TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
   TORCH_CHECK(vals.scalar_type() == torch::kInt64);
   keys = keys.contiguous();
@@ -247,7 +247,9 @@ void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
   int64_t cap = tkeys.size(0) - 1;
   TORCH_CHECK(cap > 0 && (cap & (cap - 1)) == 0);
   const uint32_t seed = 0x9acb0442u;
-  int64_t rpb = rows_per_block(n);
+  if (target_blocks <= 0) target_blocks = 4096;
+  int64_t rpb = (n + target_blocks - 1) / target_blocks;
+  if (rpb < 1024) rpb = 1024;
   int64_t nblocks = (n + rpb - 1) / rpb;
   hipLaunchKernelGGL(k_groupby_insert_sum_i64_lds, dim3((uint32_t)nblocks),
                      dim3(THREADS), 0, current_stream(),

@@ -179,9 +179,10 @@ class GroupTable:
                                      self.flags, MAX_PROBES)
         elif self._lds:
             force = int(os.environ.get("BIGSLICE_GB_LDS_FORCE", "-1"))
+            blocks = int(os.environ.get("BIGSLICE_GB_LDS_BLOCKS", "4096"))
             _C.groupby_insert_lds(keys, vals[0], self.tkeys,
                                   self.tabs[0], self.flags, MAX_PROBES,
-                                  force)
+                                  force, blocks)
         else:
             _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
                               self.tabs, self.flags, MAX_PROBES)
