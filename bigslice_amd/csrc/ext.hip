@@ -237,7 +237,8 @@ void groupby_insert(torch::Tensor keys, std::vector<torch::Tensor> vals,
 void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
                         torch::Tensor tkeys, torch::Tensor tab,
                         torch::Tensor flags, int64_t max_probes,
-                        int64_t force, int64_t target_blocks) {
+                        int64_t force, int64_t target_blocks,
+                        torch::Tensor hits) {
   TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
   TORCH_CHECK(vals.scalar_type() == torch::kInt64);
   keys = keys.contiguous();
@@ -258,7 +259,8 @@ void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
                      (long long*)tab.data_ptr<int64_t>(), cap, seed,
                      flags.data_ptr<int32_t>(),
                      flags.data_ptr<int32_t>() + 1, max_probes, rpb,
-                     (int32_t)force);
+                     (int32_t)force,
+                     hits.numel() ? (uint32_t*)hits.data_ptr() : nullptr);
   HIP_CHECK(hipGetLastError());
 }
 

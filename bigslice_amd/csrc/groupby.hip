@@ -221,7 +221,7 @@ extern "C" __global__ void k_groupby_insert_sum_i64_lds(
     const int64_t* keys, const int64_t* vals, int64_t n, int64_t* tkeys,
     long long* tab, int64_t cap, uint32_t seed, int32_t* sentinel_seen,
     int32_t* overflow, int64_t max_probes, int64_t rows_per_block,
-    int32_t force) {
+    int32_t force, uint32_t* global_hits) {
   __shared__ long long lk[GB_LDS_SLOTS];
   __shared__ long long lv[GB_LDS_SLOTS];
   __shared__ uint32_t lds_hits;
@@ -270,6 +270,9 @@ extern "C" __global__ void k_groupby_insert_sum_i64_lds(
       gb_global_insert_sum(lk[i], lv[i], tkeys, tab, gmask, cap, seed,
                            sentinel_seen, overflow, max_probes);
   }
+  // absorption feedback for the host-side mode decision
+  if (global_hits && threadIdx.x == 0 && lds_hits)
+    atomicAdd(global_hits, lds_hits);
 }
 
 // Packed-slot variant for the hottest shape (single int64 value, SUM):
