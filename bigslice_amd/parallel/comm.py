@@ -44,7 +44,8 @@ def init_comm(backend: Optional[str] = None, device: Optional[str] = None):
     Comm.  backend defaults to nccl (=RCCL) when a GPU is visible."""
     if not is_initialized():
         if backend is None:
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            backend = os.environ.get("BIGSLICE_BACKEND") or (
+                "nccl" if torch.cuda.is_available() else "gloo")
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29517")
         dist.init_process_group(backend=backend)

@@ -114,6 +114,7 @@ class LocalExecutor(Executor):
     def _run_inner(self, task: Task) -> None:
         ctx = TaskContext(device=self.device)
         dep_readers = []
+        missing = []
         for dep in task.deps:
             readers = []
             for h in dep.head_tasks:
@@ -121,16 +122,20 @@ class LocalExecutor(Executor):
                     readers.append(self.store.open(
                         h.name, dep.partition, device=self.device))
                 except KeyError:
-                    # The dep's stored output vanished (machine-loss
-                    # analog): mark the producer LOST so the evaluator
-                    # recomputes it, and resubmit this task
-                    # (exec/eval.go:352-376 semantics).
-                    h.set_state(TaskState.LOST)
-                    raise TaskLost(f"missing dep output {h.name}")
+                    missing.append(h)
             if dep.expand:
                 dep_readers.append(readers)
             else:
                 dep_readers.append(MultiReader(readers))
+        if missing:
+            # Stored dep outputs vanished (machine-loss analog): mark
+            # every missing producer LOST in one pass so the evaluator
+            # recomputes them all before this task's single resubmit
+            # (exec/eval.go:352-376 semantics).
+            for h in missing:
+                h.set_state(TaskState.LOST)
+            raise TaskLost(
+                f"missing dep outputs: {[h.name for h in missing]}")
         out = task.do(dep_readers, ctx)
         if task.num_out_columns == 0:
             # terminal (Scan) task: drive it (exec/local.go:188-193)
@@ -182,11 +187,21 @@ class LocalExecutor(Executor):
                 or self.fault_hook is not None
                 or len(task.group) <= 1):
             return None
-        from ..ops.aggregate import make_aggregator, TensorAggregator
         key = id(task.group[0])
         with self._shared_lock:
             sc = self._shared_combiners.get(key)
             if sc is None:
+                if any(s.state == TaskState.OK for s in task.group
+                       if s is not task):
+                    # Partial phase re-run (output loss after the shared
+                    # combine completed): the mode cannot recover — fall
+                    # back to solo per-task combining and invalidate any
+                    # surviving sibling outputs so consumers recompute
+                    # them too (they re-enqueue via the missing-dep
+                    # TaskLost path).
+                    for s in task.group:
+                        self.store.discard_task(s.name)
+                    return None
                 from .partition import SharedPhaseCombiner
                 sc = SharedPhaseCombiner(
                     task.schema, task.combiner, self.device,

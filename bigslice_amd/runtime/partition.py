@@ -42,7 +42,7 @@ def split_frame(frame: Frame, num_partitions: int,
         if kernels.partition_supported(frame):
             return kernels.partition_frame(frame, num_partitions,
                                            partitioner)
-    p = partition_ids(frame, num_partitions)
+    p = partition_ids(frame, num_partitions, partitioner)
     order = torch.argsort(p, stable=True)
     sorted_f = frame.select(order)
     counts = torch.bincount(p, minlength=num_partitions)

@@ -132,8 +132,34 @@ class Result(Slice):
             frames = gather(self.tasks, self.schema)
             from ..sliceio import IterReader
             return IterReader(iter(frames))
-        return MultiReader([
-            self.session.executor.reader(t, 0) for t in self.tasks])
+        # Scan-time fault tolerance (reference evalOpenerAt,
+        # exec/bigmachine.go:1500-1513): a task whose stored output
+        # vanished since the run is re-evaluated before reading.
+        session = self.session
+
+        def open_task(t):
+            try:
+                return session.executor.reader(t, 0)
+            except KeyError:
+                from ..runtime.task import TaskState
+                from .eval import evaluate
+                t.set_state(TaskState.LOST)
+                evaluate(session.executor, [t])
+                return session.executor.reader(t, 0)
+
+        from ..sliceio import FuncReader, Reader
+
+        class _LazyTaskReader(Reader):
+            def __init__(self, t):
+                self.t = t
+                self.r = None
+
+            def read(self):
+                if self.r is None:
+                    self.r = open_task(self.t)
+                return self.r.read()
+
+        return MultiReader([_LazyTaskReader(t) for t in self.tasks])
 
     def scanner(self) -> Scanner:
         return Scanner(self.open())

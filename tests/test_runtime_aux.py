@@ -170,3 +170,46 @@ def test_error_propagates_from_udf():
         lambda: bs.Map(bs.Const(2, torch.arange(4, dtype=torch.int64)),
                        boom, out_schema=(int,)))
     assert isinstance(err, ValueError)
+
+
+def test_scan_after_output_loss_reevaluates():
+    # reference session_test.go:188-246 analog: output vanishes between
+    # run and scan; the scan re-evaluates the lost tasks.
+    fv = bs.func(lambda: bs.Reduce(
+        bs.Const(3, torch.arange(300, dtype=torch.int64) % 7,
+                 torch.ones(300, dtype=torch.int64)), "sum"))
+    sess = bs.start(parallelism=2, device="cpu")
+    res = sess.run(fv)
+    # lose every task's stored output
+    for t in res.tasks:
+        sess.executor.store.discard_task(t.name)
+        for dep in t.deps:
+            for h in dep.head_tasks:
+                sess.executor.store.discard_task(h.name)
+    got = dict(res.scan())
+    want = {}
+    for k in (torch.arange(300) % 7).tolist():
+        want[k] = want.get(k, 0) + 1
+    assert got == want
+
+
+def test_repartition_custom_partitioner_colocates():
+    # reference reshuffle_test.go lengthHashKey analog: partition by a
+    # custom function of the key (here key % 3) and assert co-location.
+    def part_fn(frame, nshard):
+        return (frame.columns[0] % 3) % nshard
+
+    keys = torch.arange(90, dtype=torch.int64)
+    fv = bs.func(lambda: bs.Repartition(bs.Const(3, keys), part_fn))
+    sess = bs.start(parallelism=2, device="cpu")
+    res = sess.run(fv)
+    shard_mods = []
+    for t in res.tasks:
+        mods = set()
+        for f in sess.executor.reader(t, 0):
+            mods.update((f.columns[0] % 3).tolist())
+        shard_mods.append(mods)
+    for i in range(len(shard_mods)):
+        for j in range(i + 1, len(shard_mods)):
+            assert not (shard_mods[i] & shard_mods[j])
+    assert sorted(res.scan()) == list(range(90))
