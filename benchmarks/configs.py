@@ -88,7 +88,12 @@ def main():
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--shards", type=int, default=8)
+    ap.add_argument("--run-bytes", type=int, default=None,
+                    help="external-sort run size (forces spill when "
+                         "shard bytes exceed it)")
     args = ap.parse_args()
+    if args.run_bytes:
+        os.environ["SORT_RUN_BYTES"] = str(args.run_bytes)
     device = "cuda:0" if torch.cuda.is_available() else "cpu"
     nshard = args.shards
     sess = bs.start(parallelism=nshard, device=device)

@@ -1,9 +1,10 @@
 """Spill tier: HBM -> pinned host DRAM -> disk.
 
 Role-parity with the reference's disk spiller (sliceio/spiller.go:27-127),
-redesigned for the MI355X memory hierarchy: a spilled device batch first moves
-to pinned host DRAM over hipMemcpyAsync on a side stream (non_blocking copy),
-and only overflows to disk past a host-memory budget.  Readback streams
+redesigned for the MI355X memory hierarchy: a spilled device batch moves
+to pinned host DRAM via hipMemcpyAsync on a dedicated copy stream that
+overlaps with compute on the task's stream (events order the handoff);
+past a host-memory budget batches overflow to disk.  Readback streams
 batches back to the device asynchronously.
 """
 
@@ -11,6 +12,7 @@ from __future__ import annotations
 
 import os
 import tempfile
+import threading
 from typing import List, Optional
 
 import torch
@@ -31,6 +33,42 @@ class _DiskBatch:
         return f
 
 
+class _HostBatch:
+    """A host-resident batch whose D2H copy may still be in flight."""
+
+    __slots__ = ("frame", "event", "src_refs")
+
+    def __init__(self, frame: Frame, event, src_refs):
+        self.frame = frame
+        self.event = event  # torch.cuda.Event or None
+        self.src_refs = src_refs  # keep device tensors alive until done
+
+    def ready(self, cpu_access: bool = False):
+        if self.event is not None:
+            if cpu_access:
+                # CPU will read the pinned buffers directly
+                self.event.synchronize()
+            else:
+                # device consumption: stream ordering suffices
+                torch.cuda.current_stream().wait_event(self.event)
+            if self.event.query():
+                self.src_refs = None
+        return self.frame
+
+
+_copy_streams = {}
+_copy_lock = threading.Lock()
+
+
+def _copy_stream(device):
+    with _copy_lock:
+        s = _copy_streams.get(device)
+        if s is None:
+            s = torch.cuda.Stream(device=device)
+            _copy_streams[device] = s
+        return s
+
+
 class Spiller:
     """Accumulates spilled batches, tiering host-DRAM -> disk."""
 
@@ -38,7 +76,7 @@ class Spiller:
                  dir: Optional[str] = None, pin: bool = None):
         self.host_budget = host_budget_bytes
         self.host_used = 0
-        self.batches: List[object] = []  # Frame (host) or _DiskBatch
+        self.batches: List[object] = []  # _HostBatch | Frame | _DiskBatch
         self.rows = 0
         self._dir = dir
         self._tmpdir = None
@@ -51,12 +89,12 @@ class Spiller:
         return self._dir
 
     def spill(self, frame: Frame) -> int:
-        """Spill one batch; returns bytes spilled."""
+        """Spill one batch; returns bytes spilled.  Device batches copy
+        out asynchronously on the copy stream."""
         nbytes = frame.nbytes()
         self.rows += len(frame)
         if self.host_used + nbytes <= self.host_budget:
-            host = self._to_host(frame)
-            self.batches.append(host)
+            self.batches.append(self._to_host(frame))
             self.host_used += nbytes
             return nbytes
         d = self._ensure_dir()
@@ -66,25 +104,33 @@ class Spiller:
         self.batches.append(_DiskBatch(path))
         return nbytes
 
-    def _to_host(self, frame: Frame) -> Frame:
+    def _to_host(self, frame: Frame):
         if frame.device == "cpu":
             return frame
-        cols = []
         pin = torch.cuda.is_available() if self._pin is None else self._pin
-        for c in frame.columns:
-            if isinstance(c, torch.Tensor) and c.is_cuda:
-                if pin:
+        if not pin:
+            return frame.to("cpu")
+        device = frame.device
+        cs = _copy_stream(device)
+        # the copy stream must see the producer's writes
+        ready = torch.cuda.Event()
+        ready.record()
+        cs.wait_event(ready)
+        cols = []
+        src_refs = []
+        with torch.cuda.stream(cs):
+            for c in frame.columns:
+                if isinstance(c, torch.Tensor) and c.is_cuda:
                     dst = torch.empty(c.shape, dtype=c.dtype, device="cpu",
                                       pin_memory=True)
                     dst.copy_(c, non_blocking=True)
                     cols.append(dst)
+                    src_refs.append(c)
                 else:
-                    cols.append(c.cpu())
-            else:
-                cols.append(c)
-        if frame.device != "cpu" and pin:
-            torch.cuda.synchronize()
-        return Frame(cols, frame.prefix)
+                    cols.append(c)
+            done = torch.cuda.Event()
+            done.record()
+        return _HostBatch(Frame(cols, frame.prefix), done, src_refs)
 
     def num_batches(self) -> int:
         return len(self.batches)
@@ -103,6 +149,18 @@ class Spiller:
             self._tmpdir = None
 
 
+def load_batch(b, device: str) -> Frame:
+    if isinstance(b, _DiskBatch):
+        return b.load(device)
+    if isinstance(b, _HostBatch):
+        f = b.ready(cpu_access=(device == "cpu"))
+    else:
+        f = b
+    if device != "cpu" and f.device == "cpu":
+        return f.to(device, non_blocking=True)
+    return f
+
+
 class SpillReader:
     def __init__(self, batches: List[object], device: str):
         self.batches = batches
@@ -114,12 +172,7 @@ class SpillReader:
             return None
         b = self.batches[self.i]
         self.i += 1
-        if isinstance(b, _DiskBatch):
-            return b.load(self.device)
-        f: Frame = b
-        if self.device != "cpu":
-            return f.to(self.device, non_blocking=True)
-        return f
+        return load_batch(b, self.device)
 
     def close(self) -> None:
         pass

@@ -58,19 +58,29 @@ class SortReader(Reader):
         pending: List[Frame] = []
         pending_bytes = 0
         runs: List[List[Frame]] = []
+        first_run: List[Optional[Frame]] = [None]
 
         def flush_run():
             nonlocal pending, pending_bytes
             if not pending:
                 return
             run = sort_frame(Frame.concat(pending))
-            # keep the run on-device if it is the only one so far and
-            # small; otherwise spill to host DRAM
+            pending = []
+            pending_bytes = 0
+            if first_run[0] is None and not runs:
+                # Defer spilling the first run: if the whole input is a
+                # single run it never leaves HBM (the host round trip is
+                # pure overhead when the data fits).
+                first_run[0] = run
+                return
+            if first_run[0] is not None:
+                start = spiller.num_batches()
+                spiller.spill(first_run[0])
+                runs.append(list(range(start, spiller.num_batches())))
+                first_run[0] = None
             start = spiller.num_batches()
             spiller.spill(run)
             runs.append(list(range(start, spiller.num_batches())))
-            pending = []
-            pending_bytes = 0
 
         for f in self.source:
             pending.append(f)
@@ -78,6 +88,9 @@ class SortReader(Reader):
             if pending_bytes >= self.run_bytes:
                 flush_run()
         flush_run()
+        if first_run[0] is not None:
+            from .sliceio import FrameReader
+            return FrameReader(first_run[0], self.chunk)
         if not runs:
             return IterReader(iter(()))
         readers = [_SpillRunReader(spiller, idxs, self.device)
@@ -105,13 +118,8 @@ class _SpillRunReader(Reader):
             return None
         b = self.spiller.batches[self.idxs[self.i]]
         self.i += 1
-        from .sliceio.spiller import _DiskBatch
-        if isinstance(b, _DiskBatch):
-            return b.load(self.device)
-        f: Frame = b
-        if self.device != "cpu":
-            f = f.to(self.device, non_blocking=True)
-        return f
+        from .sliceio.spiller import load_batch
+        return load_batch(b, self.device)
 
 
 class _RunCursor:
