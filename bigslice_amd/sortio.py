@@ -142,9 +142,7 @@ class _RunCursor:
 
     def max_buffered_key(self):
         c = self.buf.columns[0]
-        if isinstance(c, torch.Tensor):
-            return c[-1].item() if self.eof else c[-1]
-        return c[-1]
+        return c[-1]  # tensor scalar for tensor columns, value for lists
 
     def take_upto(self, cutoff) -> Optional[Frame]:
         """Rows with key <= cutoff (buffer is sorted)."""

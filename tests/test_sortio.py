@@ -100,3 +100,32 @@ def test_sortreader_gpu_large():
     ok = out.columns[0]
     assert bool((ok[1:] >= ok[:-1]).all())
     assert int(out.columns[1].sum()) == int(vals.sum())
+
+
+def test_merge_reader_mixed_eof_states():
+    # regression: one run at EOF while others still buffered must not
+    # mix scalar types in the cutoff computation.
+    a = torch.arange(0, 10, dtype=torch.int64)         # short run (EOF early)
+    b = torch.arange(5, 40, dtype=torch.int64)
+    c = torch.arange(20, 60, dtype=torch.int64)
+    m = MergeReader([frames_of([a], batch=3), frames_of([b], batch=4),
+                     frames_of([c], batch=5)], chunk=6)
+    out = read_all(m)
+    got = out.columns[0].tolist()
+    assert got == sorted(a.tolist() + b.tolist() + c.tolist())
+
+
+@pytest.mark.gpu
+def test_sortreader_gpu_forced_spill():
+    g = torch.Generator(device="cuda:0").manual_seed(3)
+    keys = torch.randint(0, 1 << 50, (8_000_000,), dtype=torch.int64,
+                         device="cuda:0", generator=g)
+    vals = torch.arange(8_000_000, dtype=torch.int64, device="cuda:0")
+    sr = SortReader(frames_of([keys, vals], batch=1_000_000),
+                    run_bytes=8_000_000,  # ~0.5M rows per run -> 16 runs
+                    device="cuda:0")
+    out = read_all(sr)
+    assert len(out) == 8_000_000
+    ok = out.columns[0]
+    assert bool((ok[1:] >= ok[:-1]).all())
+    assert int(out.columns[1].sum()) == int(vals.sum())
