@@ -259,11 +259,15 @@ class WriterFunc(Slice):
     def reader(self, shard, dep_readers, ctx: TaskContext) -> Reader:
         src = dep_readers[0]
         write_fn = self.write_fn
+        schema = self.schema
 
         def gen():
             for f in src:
                 write_fn(shard, f)
                 yield f
+            # the reference invokes the writer on the EOF read too
+            # (slice.go:516-539), letting writers flush
+            write_fn(shard, Frame.empty(schema))
         return IterReader(gen())
 
 

@@ -79,10 +79,22 @@ class ReaderFunc(Slice):
 
     def reader(self, shard: int, dep_readers, ctx: TaskContext) -> Reader:
         prefix = self.schema.prefix
+
         def gen():
             it = self.fn(shard, ctx)
+            empties = 0
             for item in it:
                 f = _columns_to_frame(item, prefix)
+                if len(f) == 0:
+                    # warn on busy-looping sources (slice.go:382-385)
+                    empties += 1
+                    if empties == 8:
+                        import logging
+                        logging.getLogger("bigslice_amd").warning(
+                            "reader func returned 8 consecutive empty "
+                            "batches (shard %d)", shard)
+                    continue
+                empties = 0
                 if ctx.device != "cpu" and not f.has_objects:
                     f = f.to(ctx.device, non_blocking=True)
                 yield f

@@ -6,9 +6,11 @@ dependencies (:112-115, :352-376), give up after maxConsecutiveLost
 (:30, :139-159).  Multiple concurrent Evals coordinate through task
 state: a task already WAITING/RUNNING is watched, not re-run.
 
-Implementation: callback-driven over a persistent thread pool (no
-watcher-thread-per-task, no polling): completion of a task decrements its
-consumers' pending counts and submits newly-runnable tasks.
+Implementation: callback-driven over a persistent thread pool with the
+reference's phase-aware dependency counting (exec/eval.go:255-451): a
+shuffle consumer depends on the producer PHASE, so bookkeeping is
+O(tasks), not O(edges) — a phase keeps one remaining-task counter and
+one consumer list, and the last task of a phase wakes all consumers.
 """
 
 from __future__ import annotations
@@ -48,35 +50,61 @@ class Executor:
 def _reachable(roots: Sequence[Task]) -> List[Task]:
     out: List[Task] = []
     seen: Set[int] = set()
-
-    def visit(t: Task):
+    stack = list(roots)
+    while stack:
+        t = stack.pop()
         if id(t) in seen:
-            return
+            continue
         seen.add(id(t))
-        for dep in t.deps:
-            for h in dep.head_tasks:
-                visit(h)
         out.append(t)
-
-    for r in roots:
-        visit(r)
+        for dep in t.deps:
+            stack.extend(dep.head_tasks)
     return out
 
 
 def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
     """Evaluate all root tasks to OK (or raise)."""
     tasks = _reachable(roots)
-    consumers: Dict[int, List[Task]] = {}
+
+    def gid(t: Task) -> int:
+        return id(t.group[0])
+
+    # Phase bookkeeping over reachable tasks (exec/eval.go state):
+    phase_tasks: Dict[int, List[Task]] = {}
     for t in tasks:
+        phase_tasks.setdefault(gid(t), []).append(t)
+    # consumers of each phase (tasks with a dep whose heads are in it)
+    phase_consumers: Dict[int, List[Task]] = {}
+    dep_groups: Dict[int, List[int]] = {}  # task -> dep phase ids
+    for t in tasks:
+        groups = []
+        seen_g = set()
         for dep in t.deps:
+            # heads normally share one group; scan defensively for
+            # hand-built graphs (tests) with ungrouped heads
             for h in dep.head_tasks:
-                consumers.setdefault(id(h), []).append(t)
+                g = gid(h)
+                if g not in seen_g:
+                    seen_g.add(g)
+                    groups.append(g)
+                    phase_consumers.setdefault(g, []).append(t)
+        dep_groups[id(t)] = groups
 
     lock = threading.Lock()
     done = threading.Condition(lock)
     errors: List[BaseException] = []
-    # tasks this evaluation has claimed (submitted or is watching)
-    active: Set[int] = set()
+    enqueued: Set[int] = set()   # visited, not yet finalized
+    ok_counted: Set[int] = set()  # tasks currently counted complete
+    walked_groups: Set[int] = set()  # phases whose members were walked
+    phase_remaining: Dict[int, int] = {}
+    for g, members in phase_tasks.items():
+        rem = 0
+        for m in members:
+            if m.state == TaskState.OK:
+                ok_counted.add(id(m))
+            else:
+                rem += 1
+        phase_remaining[g] = rem
 
     pool = getattr(executor, "pool", None)
     own_pool = None
@@ -86,19 +114,14 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
             or config.DEFAULT_PARALLELISM)
         pool = own_pool
 
-    def deps_ok(t: Task) -> bool:
-        return all(h.state == TaskState.OK
-                   for dep in t.deps for h in dep.head_tasks)
+    def deps_ready(t: Task) -> bool:
+        return all(phase_remaining.get(g, 0) == 0
+                   for g in dep_groups[id(t)])
 
-    def fail(e: BaseException):
-        with lock:
-            errors.append(e)
-            done.notify_all()
-
-    def submit(t: Task):
-        """Called with lock held.  Schedule t if runnable, else make its
-        missing deps runnable first."""
-        if id(t) in active:
+    def visit(t: Task):
+        """Lock held: walk t once; start if its dep phases are complete,
+        else walk incomplete dep tasks (their completion wakes t)."""
+        if id(t) in enqueued:
             return
         st = t.state
         if st == TaskState.OK:
@@ -111,20 +134,36 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
             errors.append(TooManyTriesError(t))
             done.notify_all()
             return
+        enqueued.add(id(t))
         if st in (TaskState.WAITING, TaskState.RUNNING):
             # claimed by a concurrent evaluation: watch it
-            active.add(id(t))
-            threading.Thread(target=_watch, args=(t,), daemon=True).start()
+            threading.Thread(target=_watch, args=(t,),
+                             daemon=True).start()
             return
-        if deps_ok(t):
-            active.add(id(t))
-            t.set_state(TaskState.WAITING)
-            pool.submit(_run, t)
+        if deps_ready(t):
+            start(t)
         else:
-            for dep in t.deps:
-                for h in dep.head_tasks:
+            # walk each incomplete dep PHASE once (O(tasks), not
+            # O(consumers x heads))
+            for g in dep_groups[id(t)]:
+                if phase_remaining.get(g, 0) == 0 or g in walked_groups:
+                    continue
+                walked_groups.add(g)
+                for h in phase_tasks.get(g, ()):
                     if h.state != TaskState.OK:
-                        submit(h)
+                        note_incomplete(h)
+                        visit(h)
+
+    def note_incomplete(h: Task):
+        """A task previously counted OK went LOST/stale: restore its
+        phase counter so consumers wait again."""
+        if id(h) in ok_counted and h.state != TaskState.OK:
+            ok_counted.discard(id(h))
+            phase_remaining[gid(h)] = phase_remaining.get(gid(h), 0) + 1
+
+    def start(t: Task):
+        t.set_state(TaskState.WAITING)
+        pool.submit(_run, t)
 
     def _watch(t: Task):
         t.wait_state(TaskState.OK)
@@ -140,16 +179,30 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
     def _finish(t: Task):
         st = t.state
         with lock:
-            active.discard(id(t))
             if st == TaskState.OK:
-                for c in consumers.get(id(t), ()):  # wake consumers
-                    if id(c) not in active and c.state in (
-                            TaskState.INIT, TaskState.LOST) and deps_ok(c):
-                        submit(c)
-                # a consumer may be waiting only on this task at root
+                enqueued.discard(id(t))
+                if id(t) not in ok_counted:
+                    ok_counted.add(id(t))
+                    g = gid(t)
+                    phase_remaining[g] -= 1
+                    if phase_remaining[g] == 0:
+                        for c in phase_consumers.get(g, ()):
+                            if id(c) in enqueued and c.state in (
+                                    TaskState.INIT, TaskState.LOST) and \
+                                    deps_ready(c):
+                                start(c)
                 done.notify_all()
             elif st == TaskState.LOST:
-                submit(t)  # resubmit (re-walks lost deps)
+                # resubmit, re-walking deps that were also lost:
+                # reconcile phase counters for dep members whose OK was
+                # revoked (marked LOST by the failed run)
+                enqueued.discard(id(t))
+                note_incomplete(t)
+                for g in dep_groups[id(t)]:
+                    walked_groups.discard(g)
+                    for h in phase_tasks.get(g, ()):
+                        note_incomplete(h)
+                visit(t)
                 done.notify_all()
             else:
                 errors.append(t.error or RuntimeError(f"{t.name} failed"))
@@ -158,23 +211,20 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
     try:
         with lock:
             for r in roots:
-                submit(r)
+                visit(r)
             while not errors:
                 if all(r.state == TaskState.OK for r in roots):
                     return
-                if not active:
+                if not enqueued:
                     # quiescent but incomplete: re-walk from roots
-                    # (lost interior tasks)
-                    progressed = False
+                    # (e.g. interior tasks lost after completion)
                     for r in roots:
                         if r.state != TaskState.OK:
-                            submit(r)
-                            progressed = True
-                    if not progressed or not active:
-                        if errors:
-                            break
-                        if all(r.state == TaskState.OK for r in roots):
-                            return
+                            visit(r)
+                    if errors:
+                        break
+                    if all(r.state == TaskState.OK for r in roots):
+                        return
                 done.wait(timeout=1.0)
             raise errors[0]
     finally:
