@@ -64,37 +64,37 @@ class Aggregation:
 
 def make_aggregator(schema: Schema, agg: Aggregation, device: str,
                     seed: int = config.COMBINER_HASH_SEED):
-    """Choose the right aggregator implementation for a schema."""
-    nkey = schema.prefix
-    key_dts = schema.dtypes[:nkey]
-    val_dts = schema.dtypes[nkey:]
+    """Choose the right aggregator implementation for a schema:
+    tensor path for any all-numeric schema (single- or multi-column
+    keys), host dict path for object keys / arbitrary combine fns."""
     numeric = (not any(is_object(d) for d in schema.dtypes))
-    if agg.all_builtin and numeric and nkey == 1:
+    if agg.all_builtin and numeric and schema.prefix >= 1:
         return TensorAggregator(schema, agg, device)
     return DictAggregator(schema, agg)
 
 
 class TensorAggregator:
-    """Device/CPU aggregation for a single numeric key column.
+    """Device/CPU aggregation for numeric key column(s).
 
-    Keeps a running (keys, values...) state; each added batch is combined
-    via segment-reduce over sorted-unique keys (torch fallback), or the
-    HIP hash-aggregate kernel on GPU.  State is re-compacted whenever it
-    doubles, bounding memory at O(distinct keys).
+    Single int64 keys on GPU stream through the HIP GroupTable;
+    everything else combines via sort-based segment reduce (multi-column
+    keys use a lexicographic sort + boundary detection).  State is
+    re-compacted on flush, bounding memory at O(distinct keys).
     """
 
     def __init__(self, schema: Schema, agg: Aggregation, device: str):
         self.schema = schema
+        self.nkey = schema.prefix
         self.agg = agg
         self.device = device
-        self.keys: Optional[torch.Tensor] = None
+        self.keys: Optional[List[torch.Tensor]] = None
         self.vals: List[torch.Tensor] = []
         self._pending: List[Frame] = []
         self._pending_rows = 0
         # Device fast path: stream batches straight into the HIP
         # GroupTable (no host sync until finish).
         self._table = None
-        if device.startswith("cuda"):
+        if device.startswith("cuda") and self.nkey == 1:
             from .. import kernels
             val_dts = schema.dtypes[schema.prefix:]
             if kernels.have_extension() and all(
@@ -123,10 +123,10 @@ class TensorAggregator:
         f = Frame.concat(self._pending)
         self._pending.clear()
         self._pending_rows = 0
-        keys = f.columns[0]
-        vals = list(f.columns[1:])
+        keys = [c for c in f.columns[: self.nkey]]
+        vals = list(f.columns[self.nkey:])
         if self.keys is not None:
-            keys = torch.cat([self.keys, keys])
+            keys = [torch.cat([sk, k]) for sk, k in zip(self.keys, keys)]
             vals = [torch.cat([sv, v]) for sv, v in zip(self.vals, vals)]
         self.keys, self.vals = _combine_once(keys, vals, self.agg)
 
@@ -134,15 +134,15 @@ class TensorAggregator:
         if self._table is not None:
             keys, vals = self._table.finish()
             if keys.shape[0]:
-                self.keys, self.vals = keys, vals
+                self.keys, self.vals = [keys], vals
         else:
             self._flush()
         if self.keys is None:
             return
-        n = self.keys.shape[0]
+        n = self.keys[0].shape[0]
         for off in range(0, n, chunk):
             stop = min(off + chunk, n)
-            yield Frame([self.keys[off:stop]] +
+            yield Frame([k[off:stop] for k in self.keys] +
                         [v[off:stop] for v in self.vals],
                         self.schema.prefix)
 
@@ -150,28 +150,54 @@ class TensorAggregator:
         if self._table is not None:
             return self._table.rows and self._table.finish()[0].shape[0]
         self._flush()
-        return 0 if self.keys is None else self.keys.shape[0]
+        return 0 if self.keys is None else self.keys[0].shape[0]
 
 
-def _combine_once(keys: torch.Tensor, vals: List[torch.Tensor],
+def _combine_once(keys: List[torch.Tensor], vals: List[torch.Tensor],
                   agg: Aggregation):
-    """One compaction: group rows by key, combining values.
+    """One compaction: group rows by key column(s), combining values.
 
-    GPU path uses the HIP hash-aggregate kernel when available for the
-    supported dtype combos; otherwise sort-based segment reduce.
+    Single-key GPU batches use the HIP hash-aggregate kernel when
+    supported; otherwise a sort-based segment reduce (multi-column keys:
+    lexicographic stable sort + boundary detection), which runs on both
+    CPU and GPU.
     """
-    if keys.is_cuda:
-        from .. import kernels
-        if kernels.groupby_supported(keys, vals, agg.aggs):
-            return kernels.groupby(keys, vals, agg.aggs)
-    uk, inv = torch.unique(keys, return_inverse=True)
+    if len(keys) == 1:
+        k = keys[0]
+        if k.is_cuda:
+            from .. import kernels
+            if kernels.groupby_supported(k, vals, agg.aggs):
+                uk, ov = kernels.groupby(k, vals, agg.aggs)
+                return [uk], ov
+        uk, inv = torch.unique(k, return_inverse=True)
+        nseg = uk.shape[0]
+        uks = [uk]
+    else:
+        n = keys[0].shape[0]
+        perm = torch.arange(n, device=keys[0].device)
+        for kc in reversed(keys):
+            _, o = torch.sort(kc[perm], stable=True)
+            perm = perm[o]
+        skeys = [kc[perm] for kc in keys]
+        boundary = torch.zeros(n, dtype=torch.bool,
+                               device=keys[0].device)
+        boundary[0] = True
+        for kc in skeys:
+            boundary[1:] |= kc[1:] != kc[:-1]
+        inv_sorted = boundary.cumsum(0) - 1
+        nseg = int(inv_sorted[-1].item()) + 1 if n else 0
+        uks = [kc[boundary] for kc in skeys]
+        # map back to original row order for scatter
+        inv = torch.empty(n, dtype=torch.int64, device=keys[0].device)
+        inv[perm] = inv_sorted
+        vals = list(vals)
     out_vals = []
     for v, a in zip(vals, agg.aggs):
-        init = _scatter_init(a, v.dtype, uk.shape[0], v.device)
+        init = _scatter_init(a, v.dtype, nseg, v.device)
         out = init.scatter_reduce_(0, inv, v, reduce=_SCATTER_OP[a],
                                    include_self=False)
         out_vals.append(out)
-    return uk, out_vals
+    return uks, out_vals
 
 
 def _scatter_init(aggname: str, dtype, n: int, device):

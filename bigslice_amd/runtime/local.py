@@ -50,6 +50,7 @@ class LocalExecutor(Executor):
         self._tls = threading.local()
         self.scopes = {}  # task name -> metrics.Scope
         self.tracer = None  # utils.trace.Tracer, set by the session
+        self.eventer = None  # utils.eventlog.Eventer, set by the session
         # machine-combiners mode (reference exec/session.go:166-176):
         # producer tasks of one shuffle phase share one combiner table.
         self.machine_combiners = config.MACHINE_COMBINERS
@@ -79,6 +80,8 @@ class LocalExecutor(Executor):
             self._res_cond.wait_for(lambda: self._avail >= want)
             self._avail -= want
         task.set_state(TaskState.RUNNING)
+        import time as _time
+        _t0 = _time.perf_counter()
         try:
             if self.fault_hook is not None:
                 self.fault_hook(task)
@@ -107,6 +110,10 @@ class LocalExecutor(Executor):
             e.task_traceback = traceback.format_exc()
             task.set_state(TaskState.ERR, e)
         finally:
+            if self.eventer is not None:
+                self.eventer.task_complete(
+                    task.name, _time.perf_counter() - _t0,
+                    task.state.name)
             with self._res_cond:
                 self._avail += want
                 self._res_cond.notify_all()

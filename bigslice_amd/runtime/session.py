@@ -200,13 +200,18 @@ class Session:
     """An execution session bound to an executor (exec/session.go)."""
 
     def __init__(self, executor: Executor, parallelism: int = None,
-                 trace_path: str = None):
+                 trace_path: str = None, eventlog_path: str = None):
         self.executor = executor
         self.parallelism = parallelism
         self._inv_counter = 0
         self._lock = threading.Lock()
         self.env = CompileEnv()
         self.trace_path = trace_path
+        from ..utils.eventlog import Eventer, NOP
+        self.eventer = Eventer(eventlog_path) if eventlog_path else NOP
+        if hasattr(executor, "eventer"):
+            executor.eventer = self.eventer
+        self.eventer.session_start()
         if trace_path is not None:
             from ..utils.trace import Tracer
             self.tracer = Tracer()
@@ -256,7 +261,7 @@ class Session:
 
 def start(parallelism: int = None, device: str = None,
           executor: Executor = None, distributed: bool = None,
-          trace_path: str = None) -> Session:
+          trace_path: str = None, eventlog_path: str = None) -> Session:
     """Create a session (exec.Start analog).
 
     distributed=True (or WORLD_SIZE>1 in the environment) starts the SPMD
@@ -277,4 +282,5 @@ def start(parallelism: int = None, device: str = None,
         else:
             from .local import LocalExecutor
             executor = LocalExecutor(parallelism=parallelism, device=device)
-    return Session(executor, parallelism, trace_path=trace_path)
+    return Session(executor, parallelism, trace_path=trace_path,
+                   eventlog_path=eventlog_path)

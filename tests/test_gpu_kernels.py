@@ -174,3 +174,27 @@ def test_cogroup_device_matches_cpu():
     def norm(rows):
         return sorted((k, sorted(va), sorted(vb)) for k, va, vb in rows)
     assert norm(cpu.scan()) == norm(gpu.scan())
+
+
+def test_multikey_reduce_gpu_matches_cpu():
+    import bigslice_amd as bs
+
+    def build(nshard):
+        def gen(shard, ctx):
+            g = torch.Generator()
+            g.manual_seed(shard)
+            k1 = torch.randint(0, 50, (200_000,), dtype=torch.int64,
+                               generator=g)
+            k2 = torch.randint(0, 20, (200_000,), dtype=torch.int64,
+                               generator=g)
+            v = torch.randint(0, 100, (200_000,), dtype=torch.int64,
+                              generator=g)
+            yield (k1, k2, v)
+        src = bs.ReaderFunc(nshard, gen,
+                            bs.schema_of(int, int, int, prefix=2))
+        return bs.Reduce(src, "sum")
+
+    fv = bs.func(build)
+    cpu = bs.start(parallelism=2, device="cpu").run(fv, 3)
+    gpu = bs.start(parallelism=2, device="cuda:0").run(fv, 3)
+    assert sorted(cpu.scan()) == sorted(gpu.scan())

@@ -127,3 +127,22 @@ def test_exclusive_pragma_serializes():
     res = bs.slicetest.run(build, parallelism=4)
     assert len(list(res.scan())) == 8
     assert peak[0] == 1  # exclusive tasks never overlap
+
+
+def test_typecheck_tool(tmp_path):
+    bad = tmp_path / "bad.py"
+    bad.write_text(
+        "import bigslice_amd as bs\n"
+        "import torch\n"
+        "def build(nshard, path):\n"
+        "    return bs.Const(nshard, torch.arange(3))\n"
+        "fv = bs.func(build)\n"
+        "sess = bs.start()\n"
+        "sess.run(fv, 4)\n"          # missing `path` arg
+        "sess.run(fv, 4, 'x')\n")    # correct
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.typecheck", str(bad)],
+        capture_output=True, text=True)
+    assert p.returncode == 1
+    assert "takes 2 argument(s)" in p.stdout
+    assert p.stdout.count("\n") == 1  # only the bad call flagged
