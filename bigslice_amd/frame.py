@@ -280,9 +280,11 @@ class Frame:
         n = len(self)
         cols = self.columns[: self.prefix]
         if any(not isinstance(c, torch.Tensor) for c in cols):
+            from .hashing import custom_less_key
             keys = list(zip(*[list(c) if not isinstance(c, torch.Tensor)
                               else c.cpu().tolist() for c in cols]))
-            order = sorted(range(n), key=lambda i: keys[i])
+            order = sorted(range(n), key=lambda i: tuple(
+                custom_less_key(v) for v in keys[i]))
             return torch.tensor(order, dtype=torch.int64)
         # Stable sorts applied from least- to most-significant column give
         # a lexicographic order.

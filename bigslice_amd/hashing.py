@@ -90,6 +90,27 @@ def murmur3_bytes(data: bytes, seed: int) -> int:
         return int(_fmix(h, n))
 
 
+# User-defined per-type ops (reference frame.RegisterOps,
+# frame/ops.go:31-96): custom hash/order for object column values.
+_custom_ops = {}
+
+
+def register_ops(py_type, hash_fn=None, less_key=None):
+    """Register custom ops for a Python value type appearing in object
+    columns: hash_fn(value, seed) -> uint32 int; less_key(value) -> a
+    sortable key.  Mirrors frame.RegisterOps (the reference's only
+    in-repo use is a key type hashed by string length,
+    reshuffle_test.go:86-94)."""
+    _custom_ops[py_type] = (hash_fn, less_key)
+
+
+def custom_less_key(value):
+    ops = _custom_ops.get(type(value))
+    if ops is not None and ops[1] is not None:
+        return ops[1](value)
+    return value
+
+
 def _hash_host_column(col, seed: int) -> np.ndarray:
     """Hash one host column -> uint32 numpy array."""
     with np.errstate(over="ignore"):
@@ -113,7 +134,10 @@ def _hash_host_column(col, seed: int) -> np.ndarray:
         # object column (strings / arbitrary python values)
         out = np.empty(len(col), dtype=np.uint32)
         for i, v in enumerate(col):
-            if isinstance(v, str):
+            ops = _custom_ops.get(type(v))
+            if ops is not None and ops[0] is not None:
+                out[i] = np.uint32(ops[0](v, seed) & 0xFFFFFFFF)
+            elif isinstance(v, str):
                 out[i] = murmur3_bytes(v.encode("utf-8"), seed)
             elif isinstance(v, bytes):
                 out[i] = murmur3_bytes(v, seed)

@@ -212,3 +212,39 @@ def test_result_reuse_iterative():
     r1 = sess.run(fv1)
     r2 = sess.run(fv2, r1)
     assert sorted(r2.scan()) == [(1, 20), (2, 10)]
+
+
+def test_custom_ops_length_hash_colocation():
+    # Parity with the reference's ONLY in-repo use of RegisterOps
+    # (reshuffle_test.go:86-94): a key type hashed by its string LENGTH
+    # must land every key of one length in exactly one shard while the
+    # full multiset of rows is preserved, for shard counts 1..6.
+    from bigslice_amd import hashing
+
+    class LengthKey(str):
+        pass
+
+    hashing.register_ops(LengthKey,
+                         hash_fn=lambda v, seed: seed + len(v),
+                         less_key=lambda v: (len(v), str(v)))
+
+    words = [LengthKey(w) for w in
+             ["a", "bb", "cc", "ddd", "e", "ffff", "gg", "hhh"]]
+    vals = torch.arange(len(words), dtype=torch.int64)
+    fv = bs.func(lambda m: bs.Reshuffle(
+        bs.Const(m, list(words), vals)))
+    for m in range(1, 7):
+        sess = bs.start(parallelism=2, device="cpu")
+        res = sess.run(fv, m)
+        rows = sorted((str(k), v) for k, v in res.scan())
+        assert rows == sorted((str(k), v.item())
+                              for k, v in zip(words, vals))
+        lengths_by_shard = []
+        for t in res.tasks:
+            ls = set()
+            for f in sess.executor.reader(t, 0):
+                ls.update(len(k) for k in f.columns[0])
+            lengths_by_shard.append(ls)
+        for i in range(len(lengths_by_shard)):
+            for j in range(i + 1, len(lengths_by_shard)):
+                assert not (lengths_by_shard[i] & lengths_by_shard[j])
