@@ -82,6 +82,18 @@ class Comm:
         dist.all_gather_object(out, obj)
         return out
 
+    def any_flag(self, flag: bool) -> bool:
+        """Cheap collective OR (one small all_reduce — used as the
+        per-phase error check so the common path avoids object
+        gathers)."""
+        if self.world == 1:
+            return flag
+        t = torch.tensor([1 if flag else 0], dtype=torch.int32)
+        if self.backend == "nccl":
+            t = t.to(self.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return bool(t.cpu().item())
+
     def check_registry(self, digest: str):
         """Cross-process Func-registry verification (the FuncLocations
         diff, exec/slicemachine.go:689-702): all ranks must have built
@@ -97,18 +109,24 @@ class Comm:
 
     def exchange_buckets(
             self, send: List[List[Tuple[str, int, Frame]]],
-            schema: Schema) -> List[Tuple[str, int, Frame]]:
+            schema: Schema,
+            name_index: Optional[Dict[str, int]] = None,
+            index_name: Optional[List[str]] = None
+    ) -> List[Tuple[str, int, Frame]]:
         """All-to-allv of partition buckets.
 
         send[d] = list of (task_name, partition, frame) for dest rank d.
         Returns the buckets destined to this rank (from all ranks,
-        including self).
+        including self).  When the caller provides the phase's
+        deterministic task-name index, the metadata plane also rides
+        tensor collectives (no object gathers on the hot path).
         """
         if self.world == 1:
             return list(send[0])
         if self.tensor_exchange_ok and not any(
                 is_object(dt) for dt in schema.dtypes):
-            return self._exchange_tensors(send, schema)
+            return self._exchange_tensors(send, schema, name_index,
+                                          index_name)
         return self._exchange_objects(send, schema)
 
     def _exchange_objects(self, send, schema: Schema
@@ -124,14 +142,22 @@ class Comm:
                 out.append((t, p, f))
         return out
 
-    def _exchange_tensors(self, send, schema: Schema
+    def _exchange_tensors(self, send, schema: Schema,
+                          name_index: Optional[Dict[str, int]] = None,
+                          index_name: Optional[List[str]] = None
                           ) -> List[Tuple[str, int, Frame]]:
         device = self.device
         ncols = schema.num_columns
         # metadata: per dest, list of (task, partition, rows)
         meta = [[(t, p, len(f)) for (t, p, f) in bucket]
                 for bucket in send]
-        all_meta = self.all_gather_obj(meta)
+        if name_index is not None:
+            my_meta = self._exchange_meta_tensors(meta, name_index,
+                                                  index_name)
+        else:
+            all_meta = self.all_gather_obj(meta)
+            my_meta = [all_meta[src][self.rank]
+                       for src in range(self.world)]
 
         # Build per-column send buffers in dest-rank order.
         send_cols: List[torch.Tensor] = []
@@ -147,7 +173,7 @@ class Comm:
                 send_cols.append(torch.empty(
                     0, dtype=schema.dtypes[c], device=device))
 
-        out_splits = [sum(r for (_, _, r) in all_meta[src][self.rank])
+        out_splits = [sum(r for (_, _, r) in my_meta[src])
                       for src in range(self.world)]
         total_out = sum(out_splits)
 
@@ -165,11 +191,44 @@ class Comm:
         out: List[Tuple[str, int, Frame]] = []
         off = 0
         for src in range(self.world):
-            for (t, p, r) in all_meta[src][self.rank]:
+            for (t, p, r) in my_meta[src]:
                 cols = [rc[off:off + r] for rc in recv_cols]
                 out.append((t, p, Frame(cols, schema.prefix)))
                 off += r
         return out
+
+    def _exchange_meta_tensors(self, meta, name_index, index_name):
+        """Metadata plane over tensor collectives: (task_idx, partition,
+        rows) triples per destination, preceded by a count exchange."""
+        device = self.device
+        ents = []
+        for bucket in meta:
+            e = torch.tensor(
+                [[name_index[t], p, r] for (t, p, r) in bucket],
+                dtype=torch.int64).reshape(-1, 3)
+            ents.append(e)
+        send_counts = torch.tensor([e.shape[0] for e in ents],
+                                   dtype=torch.int64, device=device)
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts)
+        rc = recv_counts.cpu().tolist()
+        flat = (torch.cat(ents).to(device).flatten() if ents else
+                torch.empty(0, dtype=torch.int64, device=device))
+        recv_flat = torch.empty(sum(rc) * 3, dtype=torch.int64,
+                                device=device)
+        dist.all_to_all_single(
+            recv_flat, flat,
+            output_split_sizes=[c * 3 for c in rc],
+            input_split_sizes=[e.shape[0] * 3 for e in ents])
+        recv_rows = recv_flat.cpu().view(-1, 3).tolist()
+        my_meta = []
+        off = 0
+        for src in range(self.world):
+            part = recv_rows[off:off + rc[src]]
+            my_meta.append([(index_name[int(ti)], int(p), int(r))
+                            for (ti, p, r) in part])
+            off += rc[src]
+        return my_meta
 
     def gather_frames(self, frames: List[Frame], schema: Schema,
                       root: int = 0) -> List[Frame]:

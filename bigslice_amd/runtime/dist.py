@@ -120,21 +120,22 @@ class DistExecutor(Executor):
                         send[d].append((first.name, p, f))
         except BaseException as e:
             err = e
-        # Surface errors collectively so every rank raises.
-        errs = comm.all_gather_obj(repr(err) if err else None)
-        if err is not None:
-            for t in phase:
-                t.set_state(TaskState.ERR, err)
-            raise err
-        first = next((e for e in errs if e), None)
-        if first:
-            e = RuntimeError(f"remote rank failed: {first}")
+        # Surface errors collectively so every rank raises; the common
+        # path costs one tiny all_reduce, details gather only on error.
+        if comm.any_flag(err is not None):
+            errs = comm.all_gather_obj(repr(err) if err else None)
+            e = err or RuntimeError(
+                "remote rank failed: "
+                f"{next((x for x in errs if x), '?')}")
             for t in phase:
                 t.set_state(TaskState.ERR, e)
             raise e
 
         if shuffled:
-            recv = comm.exchange_buckets(send, exemplar.schema)
+            index_name = [t.name for t in phase]
+            name_index = {n: i for i, n in enumerate(index_name)}
+            recv = comm.exchange_buckets(send, exemplar.schema,
+                                         name_index, index_name)
             # group received frames by (task, partition)
             grouped: Dict[Tuple[str, int], List] = {}
             for (tname, p, f) in recv:
