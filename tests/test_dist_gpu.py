@@ -37,8 +37,11 @@ def test_rccl_exchange_world1():
     f2 = Frame([torch.arange(7, dtype=torch.int64, device="cuda:0"),
                 torch.rand(7, dtype=torch.float64, device="cuda:0")])
     send = [[("taskA", 0, f1), ("taskB", 0, f2)]]
-    # call the tensor path directly (world==1 normally short-circuits)
-    out = comm._exchange_tensors(send, schema)
+    # call the tensor path directly (world==1 normally short-circuits),
+    # with the tensor metadata plane
+    out = comm._exchange_tensors(send, schema,
+                                 {"taskA": 0, "taskB": 1},
+                                 ["taskA", "taskB"])
     assert [(t, p, len(f)) for (t, p, f) in out] == \
         [("taskA", 0, 100), ("taskB", 0, 7)]
     got = out[0][2]
