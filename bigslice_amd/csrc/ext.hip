@@ -346,6 +346,30 @@ std::vector<torch::Tensor> groupby_compact(torch::Tensor tkeys,
 
 // ---------------------------------------------------------------- sort
 
+torch::Tensor radix_sort_keys(torch::Tensor keys) {
+  TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
+  int64_t n = keys.size(0);
+  auto keys_out = torch::empty_like(keys);
+  auto run = [&](auto fn) {
+    size_t temp_bytes = 0;
+    fn(keys.data_ptr(), keys_out.data_ptr(), n, nullptr, temp_bytes,
+       current_stream());
+    auto temp = torch::empty({(int64_t)temp_bytes},
+                             keys.options().dtype(torch::kUInt8));
+    fn(keys.data_ptr(), keys_out.data_ptr(), n, temp.data_ptr(),
+       temp_bytes, current_stream());
+  };
+  switch (keys.scalar_type()) {
+    case torch::kInt64: run(radix_sort_keys_int64_t); break;
+    case torch::kInt32: run(radix_sort_keys_int32_t); break;
+    case torch::kFloat32: run(radix_sort_keys_float); break;
+    case torch::kFloat64: run(radix_sort_keys_double); break;
+    default:
+      TORCH_CHECK(false, "radix_sort_keys: unsupported dtype");
+  }
+  return keys_out;
+}
+
 torch::Tensor radix_argsort(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
@@ -396,4 +420,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hash-aggregate table compaction (K9)");
   m.def("agg_identity", &agg_identity, "aggregation identity fill");
   m.def("radix_argsort", &radix_argsort, "device radix argsort (K6)");
+  m.def("radix_sort_keys", &radix_sort_keys, "device radix key sort");
 }

@@ -22,6 +22,18 @@ static void radix_sort_pairs_impl(const K* keys_in, K* keys_out,
                              hipGetErrorString(err));
 }
 
+template <typename K>
+static void radix_sort_keys_impl(const K* keys_in, K* keys_out, int64_t n,
+                                 void* temp, size_t& temp_bytes,
+                                 hipStream_t stream) {
+  hipError_t err = rocprim::radix_sort_keys(
+      temp, temp_bytes, keys_in, keys_out, (size_t)n, 0, sizeof(K) * 8,
+      stream);
+  if (err != hipSuccess)
+    throw std::runtime_error(std::string("rocprim::radix_sort_keys: ") +
+                             hipGetErrorString(err));
+}
+
 // Explicit type dispatch used by ext.hip.
 #define INSTANTIATE_SORT(K)                                               \
   void radix_sort_pairs_##K(const void* ki, void* ko, const int64_t* vi,  \
@@ -35,3 +47,15 @@ INSTANTIATE_SORT(int64_t)
 INSTANTIATE_SORT(int32_t)
 INSTANTIATE_SORT(float)
 INSTANTIATE_SORT(double)
+
+#define INSTANTIATE_SORT_KEYS(K)                                          \
+  void radix_sort_keys_##K(const void* ki, void* ko, int64_t n,           \
+                           void* temp, size_t& temp_bytes,                \
+                           hipStream_t s) {                               \
+    radix_sort_keys_impl<K>((const K*)ki, (K*)ko, n, temp, temp_bytes, s);\
+  }
+
+INSTANTIATE_SORT_KEYS(int64_t)
+INSTANTIATE_SORT_KEYS(int32_t)
+INSTANTIATE_SORT_KEYS(float)
+INSTANTIATE_SORT_KEYS(double)

@@ -288,3 +288,10 @@ def radix_argsort(keys: torch.Tensor) -> torch.Tensor:
     if _C is None:
         return torch.argsort(keys, stable=True)
     return _C.radix_argsort(keys)
+
+
+def radix_sort_keys(keys: torch.Tensor) -> torch.Tensor:
+    """Device radix key-only sort (no permutation)."""
+    if _C is None:
+        return torch.sort(keys).values
+    return _C.radix_sort_keys(keys)
