@@ -93,13 +93,17 @@ std::tuple<std::vector<torch::Tensor>, torch::Tensor> partition_common(
     const std::vector<torch::Tensor>& cols, torch::Tensor pids,
     torch::Tensor block_hist, int64_t n, int64_t nparts, int64_t nblocks,
     int64_t rpb) {
-  // Destination offsets: per-partition base + this block's running sum
-  // (tiny tensors; stays on device).
-  auto hist = block_hist.view({nblocks, nparts}).to(torch::kInt64);
-  auto part_counts = hist.sum(0);
-  auto part_start = part_counts.cumsum(0) - part_counts;
-  auto within = hist.cumsum(0) - hist;
-  auto block_off = (part_start.unsqueeze(0) + within).contiguous();
+  // Destination offsets: one small kernel computes per-block reserved
+  // ranges + partition totals (stays on device).
+  auto opts64 = cols[0].options().dtype(torch::kInt64);
+  auto block_off = torch::empty({nblocks * nparts}, opts64);
+  auto part_counts = torch::empty({nparts}, opts64);
+  hipLaunchKernelGGL(k_part_offsets, dim3(1), dim3(THREADS), 0,
+                     current_stream(), block_hist.data_ptr<uint32_t>(),
+                     nblocks, (int32_t)nparts,
+                     block_off.data_ptr<int64_t>(),
+                     part_counts.data_ptr<int64_t>());
+  HIP_CHECK(hipGetLastError());
 
   ScatterCols sc;
   sc.n = (int)cols.size();

@@ -67,6 +67,41 @@ extern "C" __global__ void k_pids_hist(const int32_t* pids, int64_t n,
   for (int p = threadIdx.x; p < nparts; p += blockDim.x) gh[p] = lhist[p];
 }
 
+// Destination-offset computation: one small kernel replaces a chain of
+// torch cumsum/sum launches.  block_off[b][p] = start of block b's
+// reserved range in partition p; part_counts[p] = total rows.
+// Single workgroup; each thread owns partition columns (strided).
+extern "C" __global__ void k_part_offsets(const uint32_t* block_hist,
+                                          int64_t nblocks, int32_t nparts,
+                                          int64_t* block_off,
+                                          int64_t* part_counts) {
+  __shared__ int64_t totals[4096];
+  for (int p = threadIdx.x; p < nparts; p += blockDim.x) {
+    int64_t run = 0;
+    for (int64_t b = 0; b < nblocks; ++b) {
+      block_off[b * nparts + p] = run;  // within-partition offset
+      run += block_hist[b * nparts + p];
+    }
+    totals[p] = run;
+    part_counts[p] = run;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {  // exclusive scan of partition starts
+    int64_t run = 0;
+    for (int p = 0; p < nparts; ++p) {
+      int64_t c = totals[p];
+      totals[p] = run;
+      run += c;
+    }
+  }
+  __syncthreads();
+  for (int p = threadIdx.x; p < nparts; p += blockDim.x) {
+    int64_t base = totals[p];
+    for (int64_t b = 0; b < nblocks; ++b)
+      block_off[b * nparts + p] += base;
+  }
+}
+
 struct ScatterCols {
   ColDesc src[MAX_COLS];
   MutColDesc dst[MAX_COLS];
