@@ -237,6 +237,21 @@ void groupby_insert(torch::Tensor keys, std::vector<torch::Tensor> vals,
   HIP_CHECK(hipGetLastError());
 }
 
+torch::Tensor slot_pids(torch::Tensor keys, int64_t cap,
+                        int64_t nparts) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
+  keys = keys.contiguous();
+  int64_t n = keys.size(0);
+  auto pids = torch::empty({n}, keys.options().dtype(torch::kInt32));
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
+  hipLaunchKernelGGL(k_slot_pids, dim3(blocks), dim3(THREADS), 0,
+                     current_stream(), keys.data_ptr<int64_t>(), n, cap,
+                     (int32_t)nparts, 0x9acb0442u,
+                     pids.data_ptr<int32_t>());
+  HIP_CHECK(hipGetLastError());
+  return pids;
+}
+
 // Two-level LDS insert (single int64 SUM value): see groupby.hip.
 void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
                         torch::Tensor tkeys, torch::Tensor tab,
@@ -425,4 +440,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("agg_identity", &agg_identity, "aggregation identity fill");
   m.def("radix_argsort", &radix_argsort, "device radix argsort (K6)");
   m.def("radix_sort_keys", &radix_sort_keys, "device radix key sort");
+  m.def("slot_pids", &slot_pids, "table-slot-range partition ids");
 }

@@ -280,6 +280,23 @@ extern "C" __global__ void k_groupby_insert_sum_i64_lds(
 // 16-byte-aligned pair, so each row touches ONE cache line instead of
 // two — the insert is atomic-latency-bound, so halving touched lines
 // matters.  Table size 2*(cap+1); extra slot for the sentinel key.
+// Partition ids from table-slot high bits (for the partition-first
+// insert experiment: rows whose slots share a table range are inserted
+// together, giving the TCC atomics temporal locality).
+extern "C" __global__ void k_slot_pids(const int64_t* keys, int64_t n,
+                                       int64_t cap, int32_t nparts,
+                                       uint32_t seed, int32_t* pids) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t mask = (uint64_t)cap - 1;
+  uint64_t shift = 0;
+  while (((uint64_t)cap >> shift) > (uint64_t)nparts) ++shift;
+  for (; i < n; i += stride) {
+    uint64_t h = mm3_u64((uint64_t)keys[i], seed) & mask;
+    pids[i] = (int32_t)(h >> shift);
+  }
+}
+
 extern "C" __global__ void k_groupby_insert_packed_sum_i64(
     const int64_t* keys, const int64_t* vals, int64_t n, int64_t* table,
     int64_t cap, uint32_t seed, int32_t* sentinel_seen, int32_t* overflow,

@@ -35,8 +35,8 @@ def timeit(fn, iters):
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("which", choices=["groupby", "partition", "hash",
-                                      "compact"])
+    ap.add_argument("which", choices=["groupby", "groupby_pp",
+                                      "partition", "hash", "compact"])
     ap.add_argument("--rows", type=int, default=125_000_000)
     ap.add_argument("--nkeys", type=int, default=1_000_000)
     ap.add_argument("--nparts", type=int, default=8)
@@ -56,6 +56,33 @@ def main():
         def run():
             t = kernels.GroupTable([torch.int64], ["sum"], dev)
             t.insert(keys, [vals])
+            t.finish()
+        ms = timeit(run, args.iters)
+    elif args.which == "groupby_pp":
+        # partition-first: scatter rows by table-slot range, then insert
+        # bucket-by-bucket so atomics have TCC temporal locality
+        P = int(os.environ.get("PP_PARTS", "64"))
+
+        def run():
+            t = kernels.GroupTable([torch.int64], ["sum"], dev)
+            from bigslice_amd import config as cfg
+            cap = 1024
+            while cap < min(2 * keys.shape[0], cfg.GROUPBY_INITIAL_CAP):
+                cap <<= 1
+            t._alloc(cap)
+            pids = kernels._C.slot_pids(keys, cap, P)
+            (rk, rv), counts = kernels._C.scatter_by_partition(
+                [keys, vals], pids, P)
+            off = 0
+            t.rows = keys.shape[0]
+            t._mode = "global"
+            t.batches.append((keys, [vals]))
+            for c in counts.tolist():
+                if c:
+                    kernels._C.groupby_insert(
+                        rk[off:off + c], [rv[off:off + c]], t.codes,
+                        t.tkeys, t.tabs, t.flags, kernels.MAX_PROBES)
+                off += c
             t.finish()
         ms = timeit(run, args.iters)
     elif args.which == "compact":
