@@ -94,3 +94,64 @@ def test_dist_reduce_gloo():
 def test_dist_wordcount_gloo():
     results = _run_workers(_wordcount_worker, port=29627)
     assert results[0] == [("a", 3), ("b", 2), ("c", 1), ("d", 1)]
+
+
+def _matrix_worker(rank, world, port, q):
+    """Several operators through the SPMD executor in one process pair
+    (the reference's dual-executor matrix, slice_test.go:64-67)."""
+    _init(rank, world, port)
+    import bigslice_amd as bs
+
+    results = {}
+
+    # 1. Reshuffle: rows preserved
+    def b_reshuffle(m):
+        keys = torch.arange(200, dtype=torch.int64) % 11
+        vals = torch.arange(200, dtype=torch.int64)
+        return bs.Reshuffle(bs.Const(m, keys, vals))
+    fv1 = bs.func(b_reshuffle)
+
+    # 2. Cogroup
+    def b_cogroup(m):
+        a = bs.Const(m, torch.tensor([1, 2, 1], dtype=torch.int64),
+                     torch.tensor([10, 20, 30], dtype=torch.int64))
+        b = bs.Const(m, torch.tensor([2, 3], dtype=torch.int64),
+                     torch.tensor([5, 6], dtype=torch.int64))
+        return bs.Cogroup(a, b)
+    fv2 = bs.func(b_cogroup)
+
+    # 3. Fold
+    def b_fold(m):
+        keys = torch.tensor([1, 1, 2, 2, 2], dtype=torch.int64)
+        vals = torch.tensor([1, 2, 3, 4, 5], dtype=torch.int64)
+        return bs.Fold(bs.Const(m, keys, vals),
+                       lambda acc, v: (acc or 0) + v, out_schema=(int,))
+    fv3 = bs.func(b_fold)
+
+    # 4. Iterative: Result reuse
+    def b_map_prev(prev):
+        return bs.Map(prev, lambda k, v: (k, v * 2))
+    fv4 = bs.func(b_map_prev)
+
+    sess = bs.start(distributed=True, device="cpu")
+    r1 = sess.run(fv1, 3)
+    results["reshuffle"] = sorted(r1.scan())
+    r2 = sess.run(fv2, 2)
+    results["cogroup"] = sorted(
+        (k, sorted(a), sorted(b)) for k, a, b in r2.scan())
+    r3 = sess.run(fv3, 2)
+    results["fold"] = sorted(r3.scan())
+    r4 = sess.run(fv4, r3)
+    results["iterative"] = sorted(r4.scan())
+    q.put((rank, results))
+
+
+def test_dist_operator_matrix_gloo():
+    results = _run_workers(_matrix_worker, port=29641)
+    r0 = results[0]
+    keys = (torch.arange(200) % 11).tolist()
+    assert r0["reshuffle"] == sorted(zip(keys, range(200)))
+    assert r0["cogroup"] == [(1, [10, 30], []), (2, [20], [5]),
+                             (3, [], [6])]
+    assert r0["fold"] == [(1, 3), (2, 12)]
+    assert r0["iterative"] == [(1, 6), (2, 24)]
