@@ -183,34 +183,42 @@ class GroupTable:
             _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
                               self.tabs, self.flags, MAX_PROBES)
 
-    # Host-side two-level mode decision: the FIRST batch runs through
-    # the sampling LDS kernel (each block samples its own prefix and
-    # reports absorbed rows); the decision is read lazily at the SECOND
-    # insert — by then the sample kernel has long completed, so the
-    # readback costs no real sync.  Commit to LDS mode on high
-    # duplicate rates (hot keys), else the plain grid-stride kernel.
+    # Host-side two-level mode decision: sample a prefix of the first
+    # LARGE batch through the LDS kernel (which reports how many rows
+    # its LDS tier absorbed) and commit to LDS mode on high duplicate
+    # rates (hot keys), else the plain grid-stride kernel.  Small
+    # batches (e.g. pre-combined shuffle buckets on the consumer side)
+    # skip the sampling sync entirely and go straight to the global
+    # kernel — their cost is negligible either way.
+    _SAMPLE_ROWS = 1 << 22
+    _SMALL_BATCH = 1 << 21
+
     def _insert_adaptive(self, keys, vals):
         mode = getattr(self, "_mode", None)
-        pending = getattr(self, "_pending_sample", None)
         if mode is None:
             forced = os.environ.get("BIGSLICE_GB_MODE", "auto")
+            n = keys.shape[0]
             if forced in ("lds", "global"):
                 mode = self._mode = forced
-            elif pending is None:
-                n = keys.shape[0]
-                hits = torch.zeros(1, dtype=torch.int32,
-                                   device=self.device)
-                _C.groupby_insert_lds(keys, vals[0], self.tkeys,
-                                      self.tabs[0], self.flags,
-                                      MAX_PROBES, -1,
-                                      max(n // 16384, 128), hits)
-                self._pending_sample = (hits, n)
+            elif n < self._SMALL_BATCH:
+                _C.groupby_insert(keys, list(vals), self.codes,
+                                  self.tkeys, self.tabs, self.flags,
+                                  MAX_PROBES)
                 return
             else:
-                hits, sampled = pending
-                frac_ok = int(hits.item()) * 8 >= sampled
+                prefix = min(n, self._SAMPLE_ROWS)
+                hits = torch.zeros(1, dtype=torch.int32,
+                                   device=self.device)
+                _C.groupby_insert_lds(keys[:prefix], vals[0][:prefix],
+                                      self.tkeys, self.tabs[0],
+                                      self.flags, MAX_PROBES, -1,
+                                      max(prefix // 16384, 128), hits)
+                frac_ok = int(hits.item()) * 8 >= prefix  # one sync
                 mode = self._mode = "lds" if frac_ok else "global"
-                self._pending_sample = None
+                keys = keys[prefix:]
+                vals = [vals[0][prefix:]]
+                if keys.shape[0] == 0:
+                    return
         if mode == "lds":
             empty = torch.empty(0, dtype=torch.int32, device=self.device)
             blocks = int(os.environ.get("BIGSLICE_GB_LDS_BLOCKS", "4096"))
