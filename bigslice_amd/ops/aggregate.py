@@ -103,12 +103,22 @@ class TensorAggregator:
                     all(isinstance(a, str) for a in agg.aggs):
                 self._table = kernels.GroupTable(val_dts, agg.aggs, device)
 
+    # small device batches are concatenated before insert: one launch
+    # instead of many (consumer-side shuffle buckets are ~100k rows)
+    _SMALL_ROWS = 1 << 20
+
     def add(self, frame: Frame) -> None:
         if len(frame) == 0:
             return
         if frame.device != self.device:
             frame = frame.to(self.device)
         if self._table is not None:
+            if len(frame) < self._SMALL_ROWS:
+                self._pending.append(frame)
+                self._pending_rows += len(frame)
+                if self._pending_rows >= 4 * self._SMALL_ROWS:
+                    self._flush_table_pending()
+                return
             self._table.insert(frame.columns[0].contiguous(),
                                [c.contiguous() for c in frame.columns[1:]])
             return
@@ -116,6 +126,15 @@ class TensorAggregator:
         self._pending_rows += len(frame)
         if self._pending_rows >= config.COMBINER_TARGET_KEYS:
             self._flush()
+
+    def _flush_table_pending(self) -> None:
+        if not self._pending:
+            return
+        f = Frame.concat(self._pending)
+        self._pending.clear()
+        self._pending_rows = 0
+        self._table.insert(f.columns[0].contiguous(),
+                           [c.contiguous() for c in f.columns[1:]])
 
     def _flush(self) -> None:
         if not self._pending:
@@ -132,6 +151,7 @@ class TensorAggregator:
 
     def result_frames(self, chunk: int):
         if self._table is not None:
+            self._flush_table_pending()
             keys, vals = self._table.finish()
             if keys.shape[0]:
                 self.keys, self.vals = [keys], vals
@@ -148,6 +168,7 @@ class TensorAggregator:
 
     def num_keys(self) -> int:
         if self._table is not None:
+            self._flush_table_pending()
             return self._table.rows and self._table.finish()[0].shape[0]
         self._flush()
         return 0 if self.keys is None else self.keys[0].shape[0]
