@@ -1,6 +1,7 @@
 """Secondary BASELINE.json measurement configs (bench.py covers the
 north-star config 3):
 
+  1: Wordcount (Map->Reshuffle->Reduce), local executor, CPU-only
   2: Map+Filter scan over a 100M-row frame, 1 GPU (per-row kernel path)
   4: Cogroup two-slice join, 2 x keyed rows (partitioned sort-merge join)
   5: External sort with spill to host DRAM
@@ -22,6 +23,23 @@ import torch
 import bigslice_amd as bs
 
 _DATA = {}
+
+
+def build_wordcount(nshard, nlines):
+    words = ["alpha", "beta", "gamma", "delta", "epsilon", "zeta",
+             "eta", "theta"]
+
+    def lines():
+        import random
+        rng = random.Random(1)
+        for _ in range(nlines):
+            yield " ".join(rng.choice(words) for _ in range(8))
+    src = bs.ScanReader(nshard, lines)
+    toks = bs.Flatmap(src, lambda s: [(w,) for w in s.split()],
+                      out_schema=(str,), rowwise=True)
+    counts = bs.Map(toks, lambda w: (w, 1), out_schema=(str, int),
+                    rowwise=True)
+    return bs.Reduce(counts, "sum")
 
 
 def build_mapfilter(nshard):
@@ -65,6 +83,7 @@ def build_sort(nshard):
     return SortSlice(src.schema, nshard, deps=[bs.Dep(src)])
 
 
+FV_WC = bs.func(build_wordcount)
 FV_MF = bs.func(build_mapfilter)
 FV_CG = bs.func(build_cogroup)
 FV_SORT = bs.func(build_sort)
@@ -83,7 +102,8 @@ def emit(metric, rows, elapsed_s, steps, warmup, cfg, device):
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--config", type=int, required=True, choices=[2, 4, 5])
+    ap.add_argument("--config", type=int, required=True,
+                    choices=[1, 2, 4, 5])
     ap.add_argument("--rows", type=int, default=None)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
@@ -104,6 +124,30 @@ def main():
         return torch.randint(0, hi, (shape_rows,), dtype=torch.int64,
                              device=device, generator=gen)
 
+    if args.config == 1:
+        # CPU plumbing config: correctness + rows/sec through the whole
+        # engine on the host path (4 shards per BASELINE).
+        nshard = 4
+        sess = bs.start(parallelism=nshard, device="cpu")
+        nlines = args.rows or 20_000
+        rows = nlines * 8
+
+        def step():
+            res = sess.run(FV_WC, nshard, nlines)
+            total = sum(c for _, c in res.scan())
+            assert total == rows, (total, rows)
+            res.discard()
+        for _ in range(args.warmup):
+            step()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            step()
+        emit("rows/sec wordcount (CPU plumbing)", rows,
+             time.perf_counter() - t0, args.steps, args.warmup,
+             {"model": "Wordcount local CPU (BASELINE config 1)",
+              "rows_total": rows, "global_batch": rows,
+              "shards": nshard, "parallelism": "cpu"}, "cpu")
+        return
     if args.config == 2:
         rows = args.rows or 100_000_000
         per = rows // nshard

@@ -24,7 +24,7 @@ from .task import Task
 
 _funcs: List["FuncValue"] = []
 _funcs_lock = threading.Lock()
-_funcs_busy = False
+_funcs_busy = 0  # count of sessions currently running (func.go:26-28)
 
 
 class FuncValue:
@@ -57,7 +57,7 @@ def func(fn: Callable, exclusive: bool = False) -> FuncValue:
     registry digest is verified at distributed start."""
     global _funcs_busy
     with _funcs_lock:
-        if _funcs_busy:
+        if _funcs_busy > 0:
             raise RuntimeError(
                 "bigslice_amd.func called while a session is running; "
                 "Funcs must be registered at module init "
@@ -92,7 +92,7 @@ def registry_digest() -> str:
 def _mark_busy(busy: bool):
     global _funcs_busy
     with _funcs_lock:
-        _funcs_busy = busy
+        _funcs_busy += 1 if busy else -1
 
 
 class Invocation:

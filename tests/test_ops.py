@@ -248,3 +248,10 @@ def test_custom_ops_length_hash_colocation():
         for i in range(len(lengths_by_shard)):
             for j in range(i + 1, len(lengths_by_shard)):
                 assert not (lengths_by_shard[i] & lengths_by_shard[j])
+
+
+def test_reduce_keys_only_distinct():
+    # Reduce with zero value columns = distinct keys per the combiner
+    keys = torch.tensor([3, 1, 3, 2, 1, 1], dtype=torch.int64)
+    res = run_slice(lambda: bs.Reduce(bs.Const(3, keys), "sum"))
+    assert sorted_rows(res) == [1, 2, 3]
