@@ -213,3 +213,15 @@ def test_repartition_custom_partitioner_colocates():
         for j in range(i + 1, len(shard_mods)):
             assert not (shard_mods[i] & shard_mods[j])
     assert sorted(res.scan()) == list(range(90))
+
+
+def test_status_rollup():
+    from bigslice_amd.utils.status import format_status, rollup
+    fv = bs.func(lambda: bs.Reduce(
+        bs.Const(3, torch.arange(30, dtype=torch.int64) % 4,
+                 torch.ones(30, dtype=torch.int64)), "sum"))
+    sess = bs.start(parallelism=2, device="cpu")
+    res = sess.run(fv)
+    r = rollup(res.tasks)
+    assert all(v == {"OK": 3} for v in r.values())
+    assert "3/3" in format_status(res.tasks)
