@@ -107,3 +107,66 @@ def test_frame_reader_chunks():
     r = FrameReader(f, chunk=3)
     sizes = [len(b) for b in r]
     assert sizes == [3, 3, 3, 1]
+
+
+def test_fuzz_codec_sort_partition_roundtrip():
+    # randomized multi-dtype frames (reference fuzzFrame-style tests):
+    # codec roundtrip, prefix sort order, partition row preservation.
+    import io
+    import random
+
+    from bigslice_amd.runtime.partition import split_frame
+
+    rng = random.Random(11)
+    for trial in range(10):
+        n = rng.randint(1, 400)
+        cols = []
+        dts = [torch.int64, torch.int32, torch.float64, torch.float32,
+               torch.bool]
+        rng.shuffle(dts)
+        ncols = rng.randint(1, 4)
+        for dt in dts[:ncols]:
+            if dt == torch.bool:
+                cols.append(torch.randint(0, 2, (n,)).to(torch.bool))
+            elif dt.is_floating_point:
+                cols.append(torch.randn(n, dtype=dt))
+            else:
+                cols.append(torch.randint(-50, 50, (n,), dtype=dt))
+        if rng.random() < 0.5:
+            cols.append(["s%d" % rng.randint(0, 20) for _ in range(n)])
+        prefix = 1
+        f = Frame(cols, prefix)
+        # codec roundtrip
+        buf = io.BytesIO()
+        encode_frame(f, buf)
+        buf.seek(0)
+        g = decode_frame(buf)
+        assert g.schema == f.schema and len(g) == n
+        for a, b in zip(f.columns, g.columns):
+            if isinstance(a, torch.Tensor):
+                assert torch.equal(a, b)
+            else:
+                assert a == b
+        # sort by prefix is ordered and row-preserving
+        s = f.sort_by_prefix()
+        k = s.columns[0]
+        if isinstance(k, torch.Tensor) and not k.dtype.is_floating_point:
+            assert bool((k[1:] >= k[:-1]).all())
+        # partition preserves the multiset of first-column values
+        if isinstance(f.columns[0], torch.Tensor) and \
+                f.columns[0].dtype == torch.int64 and not f.has_objects:
+            parts = split_frame(f, 4, None)
+            got = sorted(v for p in parts if p is not None
+                         for v in p.columns[0].tolist())
+            assert got == sorted(f.columns[0].tolist())
+
+
+def test_unpicklable_object_column_errors():
+    # reference slice_test.go:983 (non-gob-encodable data): encoding a
+    # frame with an unpicklable value must raise, and the error must
+    # surface from a Cache write (task ERR, not a hang).
+    import io
+    f = Frame([[lambda: 1, lambda: 2]])
+    buf = io.BytesIO()
+    with pytest.raises(Exception):
+        encode_frame(f, buf)
