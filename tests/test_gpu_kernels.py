@@ -198,3 +198,38 @@ def test_multikey_reduce_gpu_matches_cpu():
     cpu = bs.start(parallelism=2, device="cpu").run(fv, 3)
     gpu = bs.start(parallelism=2, device="cuda:0").run(fv, 3)
     assert sorted(cpu.scan()) == sorted(gpu.scan())
+
+
+def test_chained_pipeline_gpu_matches_cpu():
+    # map -> filter -> reduce -> (reuse) -> map over a shuffle: one
+    # integration chain compared CPU vs GPU.
+    import bigslice_amd as bs
+
+    def build(nshard):
+        def gen(shard, ctx):
+            g = torch.Generator()
+            g.manual_seed(100 + shard)
+            keys = torch.randint(0, 3000, (300_000,), dtype=torch.int64,
+                                 generator=g)
+            vals = torch.randint(1, 50, (300_000,), dtype=torch.int64,
+                                 generator=g)
+            yield (keys, vals)
+        src = bs.ReaderFunc(nshard, gen, bs.schema_of(int, int))
+        mapped = bs.Map(src, lambda k, v: (k, v * 2))
+        filt = bs.Filter(mapped, lambda k, v: (k & 1) == 0)
+        red = bs.Reduce(filt, "sum")
+        return bs.Map(red, lambda k, s: (k, s + 1))
+
+    fv = bs.func(build)
+    cpu = bs.start(parallelism=2, device="cpu").run(fv, 4)
+    gpu = bs.start(parallelism=2, device="cuda:0").run(fv, 4)
+    assert sorted(cpu.scan()) == sorted(gpu.scan())
+
+
+def test_reduce_keys_only_gpu():
+    import bigslice_amd as bs
+    keys = torch.randint(0, 500, (1_000_000,), dtype=torch.int64)
+    fv = bs.func(lambda: bs.Reduce(bs.Const(3, keys), "sum"))
+    cpu = bs.start(parallelism=2, device="cpu").run(fv)
+    gpu = bs.start(parallelism=2, device="cuda:0").run(fv)
+    assert sorted(cpu.scan()) == sorted(gpu.scan())

@@ -225,3 +225,9 @@ def test_status_rollup():
     r = rollup(res.tasks)
     assert all(v == {"OK": 3} for v in r.values())
     assert "3/3" in format_status(res.tasks)
+
+
+def test_machine_stats():
+    from bigslice_amd.utils.machine import machine_stats
+    s = machine_stats()
+    assert s["rss_gb"] > 0
