@@ -156,11 +156,16 @@ class Filter(_PipelinedSlice):
 
 class Flatmap(_PipelinedSlice):
     """Vector-output UDF.  Vectorized form: fn(*cols) -> columns of any
-    (uniform) length.  Rowwise form: fn(*row) -> iterable of rows."""
+    (uniform) length.  Rowwise form: fn(*row) -> iterable of rows.
+    fn_factory, when given, builds a fresh fn per shard reader (for
+    UDFs carrying per-shard state, the reference's ReaderFunc-state
+    analog for flatmaps)."""
 
     def __init__(self, dep: Slice, fn: Callable, out_schema=None,
-                 rowwise: bool = False, prefix: int = None):
+                 rowwise: bool = False, prefix: int = None,
+                 fn_factory: Callable = None):
         self.fn = fn
+        self.fn_factory = fn_factory
         self.rowwise = rowwise
         if out_schema is not None:
             schema = out_schema if isinstance(out_schema, Schema) \
@@ -180,6 +185,8 @@ class Flatmap(_PipelinedSlice):
     def reader(self, shard, dep_readers, ctx: TaskContext) -> Reader:
         src = dep_readers[0]
         fn, rowwise, schema = self.fn, self.rowwise, self.schema
+        if self.fn_factory is not None:
+            fn = self.fn_factory()
 
         def gen():
             for f in src:

@@ -25,7 +25,17 @@ def build_wordcount(nshard, path):
     return bs.Reduce(counts, "sum")
 
 
+def build_fast_wordcount(nshard, path):
+    from bigslice_amd.recipes import fast_wordcount
+
+    def lines():
+        with open(path) as fp:
+            yield from fp
+    return fast_wordcount(nshard, lines)
+
+
 wordcount = bs.func(build_wordcount)
+wordcount_fast = bs.func(build_fast_wordcount)
 
 
 def main():
@@ -33,9 +43,12 @@ def main():
     ap.add_argument("file")
     ap.add_argument("--shards", type=int, default=8)
     ap.add_argument("--top", type=int, default=20)
+    ap.add_argument("--fast", action="store_true",
+                    help="dictionary-encoded device-reduce recipe")
     args, rest = ap.parse_known_args()
     sess, _ = bs.sliceconfig.parse(rest)
-    res = sess.run(wordcount, args.shards, args.file)
+    fv = wordcount_fast if args.fast else wordcount
+    res = sess.run(fv, args.shards, args.file)
     rows = sorted(res.scan(), key=lambda kv: (-kv[1], kv[0]))
     for w, c in rows[: args.top]:
         print(f"{c:8d}  {w}")

@@ -255,3 +255,14 @@ def test_reduce_keys_only_distinct():
     keys = torch.tensor([3, 1, 3, 2, 1, 1], dtype=torch.int64)
     res = run_slice(lambda: bs.Reduce(bs.Const(3, keys), "sum"))
     assert sorted_rows(res) == [1, 2, 3]
+
+
+def test_fast_wordcount_recipe():
+    from bigslice_amd.recipes import fast_wordcount
+    text = ["the quick brown fox", "jumps over the lazy dog",
+            "the fox"] * 5
+    fv = bs.func(lambda n: fast_wordcount(n, lambda: iter(text)))
+    res = bs.start(parallelism=2, device="cpu").run(fv, 4)
+    got = dict(res.scan())
+    assert got == {"the": 15, "quick": 5, "brown": 5, "fox": 10,
+                   "jumps": 5, "over": 5, "lazy": 5, "dog": 5}
