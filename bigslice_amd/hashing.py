@@ -65,29 +65,43 @@ def murmur3_u64(vals: np.ndarray, seed: int) -> np.ndarray:
     return _fmix(h, 8)
 
 
+_M32 = 0xFFFFFFFF
+
+
 def murmur3_bytes(data: bytes, seed: int) -> int:
-    """Scalar murmur3_x86_32 over arbitrary bytes (strings)."""
-    with np.errstate(over="ignore"):
-        h = np.uint32(seed)
-        n = len(data)
-        nblocks = n // 4
-        if nblocks:
-            blocks = np.frombuffer(data[: nblocks * 4], dtype="<u4")
-            for k in blocks:
-                h = _mix_block(h, np.uint32(k))
-        tail = data[nblocks * 4:]
-        k1 = np.uint32(0)
+    """Scalar murmur3_x86_32 over arbitrary bytes (strings).  Pure-int
+    arithmetic: ~50x faster per call than numpy scalar ops (this runs
+    per word in host string pipelines)."""
+    h = seed & _M32
+    n = len(data)
+    nblocks4 = n & ~3
+    for off in range(0, nblocks4, 4):
+        k = int.from_bytes(data[off:off + 4], "little")
+        k = (k * 0xCC9E2D51) & _M32
+        k = ((k << 15) | (k >> 17)) & _M32
+        k = (k * 0x1B873593) & _M32
+        h ^= k
+        h = ((h << 13) | (h >> 19)) & _M32
+        h = (h * 5 + 0xE6546B64) & _M32
+    tail = data[nblocks4:]
+    if tail:
+        k1 = 0
         if len(tail) >= 3:
-            k1 ^= np.uint32(tail[2]) << np.uint32(16)
+            k1 ^= tail[2] << 16
         if len(tail) >= 2:
-            k1 ^= np.uint32(tail[1]) << np.uint32(8)
-        if len(tail) >= 1:
-            k1 ^= np.uint32(tail[0])
-            k1 = k1 * _C1
-            k1 = _rotl32(k1, 15)
-            k1 = k1 * _C2
-            h = h ^ k1
-        return int(_fmix(h, n))
+            k1 ^= tail[1] << 8
+        k1 ^= tail[0]
+        k1 = (k1 * 0xCC9E2D51) & _M32
+        k1 = ((k1 << 15) | (k1 >> 17)) & _M32
+        k1 = (k1 * 0x1B873593) & _M32
+        h ^= k1
+    h ^= n
+    h ^= h >> 16
+    h = (h * 0x85EBCA6B) & _M32
+    h ^= h >> 13
+    h = (h * 0xC2B2AE35) & _M32
+    h ^= h >> 16
+    return h
 
 
 # User-defined per-type ops (reference frame.RegisterOps,
@@ -109,6 +123,17 @@ def custom_less_key(value):
     if ops is not None and ops[1] is not None:
         return ops[1](value)
     return value
+
+
+def _mm3_u32_int(x: int, seed: int) -> int:
+    """Pure-int reference hash32 (4 LE bytes)."""
+    return murmur3_bytes((x & _M32).to_bytes(4, "little"), seed)
+
+
+def _mm3_u64_int(x: int, seed: int) -> int:
+    """Pure-int reference hash64 (8 LE bytes)."""
+    return murmur3_bytes((x & 0xFFFFFFFFFFFFFFFF).to_bytes(8, "little"),
+                         seed)
 
 
 def _hash_host_column(col, seed: int) -> np.ndarray:
@@ -142,13 +167,13 @@ def _hash_host_column(col, seed: int) -> np.ndarray:
             elif isinstance(v, bytes):
                 out[i] = murmur3_bytes(v, seed)
             elif isinstance(v, bool):
-                out[i] = np.uint32(seed) + np.uint32(v)
+                out[i] = (seed + int(v)) & _M32
             elif isinstance(v, int):
-                out[i] = murmur3_u64(np.array([v], dtype=np.int64)
-                                     .view(np.uint64), seed)[0]
+                out[i] = _mm3_u64_int(v, seed)
             elif isinstance(v, float):
-                out[i] = murmur3_u64(np.array([v], dtype=np.float64)
-                                     .view(np.uint64), seed)[0]
+                import struct as _struct
+                bits = _struct.unpack("<Q", _struct.pack("<d", v))[0]
+                out[i] = _mm3_u64_int(bits, seed)
             elif isinstance(v, tuple):
                 # hash tuples by hashing the concatenated member hashes
                 h = np.uint32(seed)
