@@ -1,12 +1,19 @@
 """Recipes: composed patterns over the public API.
 
 fast_wordcount: the reference's headline wordcount
-(docs/index.md:93-155) restructured so the heavy aggregation runs on
-the GPU: the rowwise tokenizer (unavoidably host-side in Python)
+(docs/index.md:93-155) restructured so the aggregation runs on the
+GPU: the rowwise tokenizer (unavoidably host-side in Python)
 dictionary-encodes words to 64-bit murmur hashes AND emits each
 shard's (id -> word) mapping only once per new word, so the count
 Reduce runs device-native over int64 ids and only the small distinct
 mapping flows on the host path.
+
+MEASURED verdict (3M words, 20k vocabulary, 1x MI355X): the plain
+string-keyed wordcount wins (3.5 s vs 8.6 s) — Python tokenization
+dominates both, and a C-level dict count is cheaper than the extra
+encode pass.  Use this pattern when the per-key aggregation itself is
+heavy (many value columns, large key counts, downstream device
+compute), not for simple counting.
 
 Collision note: ids are murmur3-64 (two murmur3-32 lanes); distinct
 words colliding would merge counts with probability ~V^2/2^65 —

@@ -44,7 +44,8 @@ def main():
     ap.add_argument("--shards", type=int, default=8)
     ap.add_argument("--top", type=int, default=20)
     ap.add_argument("--fast", action="store_true",
-                    help="dictionary-encoded device-reduce recipe")
+                    help="dictionary-encoded device-reduce recipe "
+                         "(see recipes.fast_wordcount verdict)")
     args, rest = ap.parse_known_args()
     sess, _ = bs.sliceconfig.parse(rest)
     fv = wordcount_fast if args.fast else wordcount
