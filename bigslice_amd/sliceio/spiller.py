@@ -162,17 +162,56 @@ def load_batch(b, device: str) -> Frame:
 
 
 class SpillReader:
+    """Reads spilled batches back, prefetching the NEXT batch's H2D
+    copy on the copy stream while the caller consumes the current one
+    (keeps the host link busy through a k-way merge)."""
+
     def __init__(self, batches: List[object], device: str):
         self.batches = batches
         self.device = device
         self.i = 0
+        self._pre = None  # (frame_on_device, ready_event)
 
-    def read(self) -> Optional[Frame]:
-        if self.i >= len(self.batches):
-            return None
+    def _start_prefetch(self):
+        if self.i >= len(self.batches) or self.device == "cpu":
+            self._pre = None
+            return
         b = self.batches[self.i]
         self.i += 1
-        return load_batch(b, self.device)
+        if isinstance(b, _DiskBatch):
+            # disk loads are synchronous; no async prefetch
+            self._pre = (b.load(self.device), None)
+            return
+        if isinstance(b, _HostBatch):
+            f = b.ready(cpu_access=False)
+        else:
+            f = b
+        cs = _copy_stream(self.device)
+        with torch.cuda.stream(cs):
+            dev = f.to(self.device, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+        self._pre = (dev, ev)
+
+    def read(self) -> Optional[Frame]:
+        if self.device == "cpu":
+            if self.i >= len(self.batches):
+                return None
+            b = self.batches[self.i]
+            self.i += 1
+            return load_batch(b, self.device)
+        if self._pre is None:
+            if self.i >= len(self.batches):
+                return None
+            self._start_prefetch()
+            if self._pre is None:
+                return None
+        f, ev = self._pre
+        self._pre = None
+        self._start_prefetch()  # overlap next H2D with consumption
+        if ev is not None:
+            torch.cuda.current_stream().wait_event(ev)
+        return f
 
     def close(self) -> None:
         pass

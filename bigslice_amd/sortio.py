@@ -108,18 +108,12 @@ class SortReader(Reader):
 class _SpillRunReader(Reader):
     def __init__(self, spiller: Spiller, batch_idxs: List[int],
                  device: str):
-        self.spiller = spiller
-        self.idxs = batch_idxs
-        self.device = device
-        self.i = 0
+        from .sliceio.spiller import SpillReader
+        self._inner = SpillReader(
+            [spiller.batches[i] for i in batch_idxs], device)
 
     def read(self) -> Optional[Frame]:
-        if self.i >= len(self.idxs):
-            return None
-        b = self.spiller.batches[self.idxs[self.i]]
-        self.i += 1
-        from .sliceio.spiller import load_batch
-        return load_batch(b, self.device)
+        return self._inner.read()
 
 
 class _RunCursor:
