@@ -60,12 +60,15 @@ _copy_streams = {}
 _copy_lock = threading.Lock()
 
 
-def _copy_stream(device):
+def _copy_stream(device, direction: str = "d2h"):
+    """Per-(thread, direction) copy streams: concurrent tasks' spills
+    (D2H) and readbacks (H2D) must not serialize on one stream."""
+    key = (device, direction, threading.get_ident())
     with _copy_lock:
-        s = _copy_streams.get(device)
+        s = _copy_streams.get(key)
         if s is None:
             s = torch.cuda.Stream(device=device)
-            _copy_streams[device] = s
+            _copy_streams[key] = s
         return s
 
 
@@ -111,7 +114,7 @@ class Spiller:
         if not pin:
             return frame.to("cpu")
         device = frame.device
-        cs = _copy_stream(device)
+        cs = _copy_stream(device, "d2h")
         # the copy stream must see the producer's writes
         ready = torch.cuda.Event()
         ready.record()
@@ -182,12 +185,12 @@ class SpillReader:
             # disk loads are synchronous; no async prefetch
             self._pre = (b.load(self.device), None)
             return
-        if isinstance(b, _HostBatch):
-            f = b.ready(cpu_access=False)
-        else:
-            f = b
-        cs = _copy_stream(self.device)
+        cs = _copy_stream(self.device, "h2d")
         with torch.cuda.stream(cs):
+            # order the H2D read after the batch's D2H write (waits on
+            # cs, the stream doing the read)
+            f = b.ready(cpu_access=False) if isinstance(b, _HostBatch) \
+                else b
             dev = f.to(self.device, non_blocking=True)
             ev = torch.cuda.Event()
             ev.record()
