@@ -231,6 +231,8 @@ class GroupTable:
 
     def finish(self):
         """Returns (keys, [vals]) of the aggregated groups."""
+        from ..utils import stats
+        stats.DEFAULT.add("combiner/records", self.rows)
         if self.cap is None:
             empty = torch.empty(0, dtype=torch.int64, device=self.device)
             return empty, [torch.empty(0, dtype=dt, device=self.device)
@@ -274,6 +276,8 @@ class GroupTable:
                     vals = [torch.cat([v, t[self.cap:self.cap + 1]])
                             for v, t in zip(vals, self.tabs)]
             self.batches = []
+            from ..utils import stats
+            stats.DEFAULT.add("combiner/keys", int(keys.shape[0]))
             return keys, vals
 
 
