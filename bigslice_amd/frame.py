@@ -84,11 +84,19 @@ def _col_dtype(col: Column):
 
 
 class Frame:
-    """Equal-length typed columns with a key prefix."""
+    """Equal-length typed columns with a key prefix.
 
-    __slots__ = ("columns", "prefix")
+    combined_id: non-None marks a frame whose prefix keys are UNIQUE,
+    produced by one combiner instance (the id).  A keyed reduce that
+    receives frames of a single combiner instance may pass them through
+    without re-aggregating (the machine-combiners single-stream case).
+    Transforms that preserve row identity propagate it; merges drop it.
+    """
 
-    def __init__(self, columns: Sequence[Column], prefix: int = None):
+    __slots__ = ("columns", "prefix", "combined_id")
+
+    def __init__(self, columns: Sequence[Column], prefix: int = None,
+                 combined_id=None):
         self.columns: List[Column] = list(columns)
         n = None
         for c in self.columns:
@@ -100,6 +108,7 @@ class Frame:
         if prefix is None:
             prefix = 1 if self.columns else 0
         self.prefix = prefix
+        self.combined_id = combined_id
 
     # -- construction ----------------------------------------------------
 
@@ -172,7 +181,8 @@ class Frame:
     def slice(self, start: int, stop: int) -> "Frame":
         """Zero-copy row range (reference frame.Slice, frame.go:244)."""
         return Frame(
-            [c[start:stop] for c in self.columns], self.prefix)
+            [c[start:stop] for c in self.columns], self.prefix,
+            combined_id=self.combined_id)
 
     def select(self, indices: torch.Tensor) -> "Frame":
         """Gather rows by index tensor (device gather kernel)."""
@@ -185,7 +195,7 @@ class Frame:
             else:
                 idx = indices.cpu().tolist()
                 cols.append([c[i] for i in idx])
-        return Frame(cols, self.prefix)
+        return Frame(cols, self.prefix, combined_id=self.combined_id)
 
     def mask(self, keep: torch.Tensor) -> "Frame":
         """Filter rows by boolean mask (compaction; reference Filter's
@@ -199,7 +209,7 @@ class Frame:
             else:
                 km = keep.cpu().numpy()
                 cols.append([x for x, k in zip(c, km) if k])
-        return Frame(cols, self.prefix)
+        return Frame(cols, self.prefix, combined_id=self.combined_id)
 
     @staticmethod
     def concat(frames: Sequence["Frame"]) -> "Frame":
@@ -242,7 +252,7 @@ class Frame:
                     raise ValueError(
                         "object columns cannot move to device memory")
                 cols.append(c)
-        return Frame(cols, self.prefix)
+        return Frame(cols, self.prefix, combined_id=self.combined_id)
 
     def clone(self) -> "Frame":
         cols = []

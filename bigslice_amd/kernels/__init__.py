@@ -70,7 +70,10 @@ def partition_frame(frame, num_partitions: int, partitioner):
         key_cols = list(frame.columns[: frame.prefix])
         reordered_cols, counts = _C.hash_partition(
             list(frame.columns), key_cols, num_partitions, 0)
-    sorted_f = Frame(reordered_cols, frame.prefix)
+    # partitioning preserves row identity: a unique-keyed (combined)
+    # input stays unique per partition
+    sorted_f = Frame(reordered_cols, frame.prefix,
+                     combined_id=frame.combined_id)
     out = []
     off = 0
     for c in counts.tolist():

@@ -91,6 +91,14 @@ class TensorAggregator:
         self.vals: List[torch.Tensor] = []
         self._pending: List[Frame] = []
         self._pending_rows = 0
+        # Pre-combined passthrough: while every added frame carries the
+        # SAME combiner instance id, its keys are globally unique and
+        # re-aggregation is the identity (the machine-combiners
+        # single-stream case, e.g. one GPU's shared combine feeding its
+        # own consumer).  A frame from a different (or no) instance
+        # demotes everything to normal aggregation.
+        self._pt_frames: Optional[List[Frame]] = []
+        self._pt_id = None
         # Device fast path: stream batches straight into the HIP
         # GroupTable (no host sync until finish).
         self._table = None
@@ -112,6 +120,20 @@ class TensorAggregator:
             return
         if frame.device != self.device:
             frame = frame.to(self.device)
+        if self._pt_frames is not None:
+            cid = frame.combined_id
+            if cid is not None and (self._pt_id is None
+                                    or cid == self._pt_id):
+                self._pt_id = cid
+                self._pt_frames.append(frame)
+                return
+            # demote: replay buffered frames through normal aggregation
+            buffered, self._pt_frames = self._pt_frames, None
+            for f in buffered:
+                self._add_normal(f)
+        self._add_normal(frame)
+
+    def _add_normal(self, frame: Frame) -> None:
         if self._table is not None:
             if len(frame) < self._SMALL_ROWS:
                 self._pending.append(frame)
@@ -150,6 +172,11 @@ class TensorAggregator:
         self.keys, self.vals = _combine_once(keys, vals, self.agg)
 
     def result_frames(self, chunk: int):
+        if self._pt_frames is not None:
+            # single pre-combined stream: identity (keys already unique)
+            for f in self._pt_frames:
+                yield f
+            return
         if self._table is not None:
             self._flush_table_pending()
             keys, vals = self._table.finish()
@@ -159,14 +186,18 @@ class TensorAggregator:
             self._flush()
         if self.keys is None:
             return
+        import uuid
+        cid = uuid.uuid4().int & ((1 << 63) - 1)
         n = self.keys[0].shape[0]
         for off in range(0, n, chunk):
             stop = min(off + chunk, n)
             yield Frame([k[off:stop] for k in self.keys] +
                         [v[off:stop] for v in self.vals],
-                        self.schema.prefix)
+                        self.schema.prefix, combined_id=cid)
 
     def num_keys(self) -> int:
+        if self._pt_frames is not None:
+            return sum(len(f) for f in self._pt_frames)
         if self._table is not None:
             self._flush_table_pending()
             return self._table.rows and self._table.finish()[0].shape[0]

@@ -266,3 +266,25 @@ def test_fast_wordcount_recipe():
     got = dict(res.scan())
     assert got == {"the": 15, "quick": 5, "brown": 5, "fox": 10,
                    "jumps": 5, "over": 5, "lazy": 5, "dog": 5}
+
+
+def test_precombined_passthrough_correctness():
+    # The single-stream passthrough must yield identical results to
+    # full re-aggregation (machine-combiners on and off).
+    import os
+    keys = torch.randint(0, 97, (5000,), dtype=torch.int64)
+    vals = torch.randint(0, 10, (5000,), dtype=torch.int64)
+
+    fv = bs.func(lambda: bs.Reduce(bs.Const(6, keys, vals), "sum"))
+    on = bs.start(parallelism=3, device="cpu").run(fv)
+    from bigslice_amd.runtime.local import LocalExecutor
+    from bigslice_amd.runtime.session import Session
+    ex = LocalExecutor(parallelism=3, device="cpu")
+    ex.machine_combiners = False
+    off = Session(ex).run(fv)
+    assert sorted(on.scan()) == sorted(off.scan())
+    # oracle
+    want = {}
+    for k, v in zip(keys.tolist(), vals.tolist()):
+        want[k] = want.get(k, 0) + v
+    assert dict(on.scan()) == want
