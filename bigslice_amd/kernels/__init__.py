@@ -54,6 +54,10 @@ def hash_columns_device(cols: List[torch.Tensor], seed: int) -> torch.Tensor:
 
 def partition_supported(frame) -> bool:
     if _C is None:
+        if not ALLOW_FALLBACK:
+            # no silent eager fallback on a GPU box (the HIP path must
+            # be the one that runs)
+            _require("partition_frame")
         return False
     return all(isinstance(c, torch.Tensor) for c in frame.columns)
 
@@ -96,6 +100,8 @@ MAX_PROBES = 128
 def groupby_supported(keys: torch.Tensor, vals: List[torch.Tensor],
                       aggs: List[str]) -> bool:
     if _C is None:
+        if keys.is_cuda and not ALLOW_FALLBACK:
+            _require("groupby")
         return False
     if keys.dtype not in _GB_KEY_DTYPES:
         return False
@@ -302,6 +308,8 @@ def sort_pairs_supported(keys: torch.Tensor) -> bool:
 def radix_argsort(keys: torch.Tensor) -> torch.Tensor:
     """Device radix sort; returns the sorting permutation (int64)."""
     if _C is None:
+        if keys.is_cuda and not ALLOW_FALLBACK:
+            _require("radix_argsort")
         return torch.argsort(keys, stable=True)
     return _C.radix_argsort(keys)
 
@@ -309,5 +317,7 @@ def radix_argsort(keys: torch.Tensor) -> torch.Tensor:
 def radix_sort_keys(keys: torch.Tensor) -> torch.Tensor:
     """Device radix key-only sort (no permutation)."""
     if _C is None:
+        if keys.is_cuda and not ALLOW_FALLBACK:
+            _require("radix_sort_keys")
         return torch.sort(keys).values
     return _C.radix_sort_keys(keys)

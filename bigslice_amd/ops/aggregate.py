@@ -105,10 +105,14 @@ class TensorAggregator:
         if device.startswith("cuda") and self.nkey == 1:
             from .. import kernels
             val_dts = schema.dtypes[schema.prefix:]
-            if kernels.have_extension() and all(
-                    dt in kernels._GB_VAL_DTYPES for dt in val_dts) and \
-                    schema.dtypes[0] in kernels._GB_KEY_DTYPES and \
-                    all(isinstance(a, str) for a in agg.aggs):
+            shape_ok = (all(dt in kernels._GB_VAL_DTYPES
+                            for dt in val_dts)
+                        and schema.dtypes[0] in kernels._GB_KEY_DTYPES
+                        and all(isinstance(a, str) for a in agg.aggs))
+            if shape_ok and not kernels.have_extension() and \
+                    not kernels.ALLOW_FALLBACK:
+                kernels._require("groupby")  # no silent eager fallback
+            if shape_ok and kernels.have_extension():
                 self._table = kernels.GroupTable(val_dts, agg.aggs, device)
 
     # small device batches are concatenated before insert: one launch
