@@ -231,3 +231,42 @@ def test_machine_stats():
     from bigslice_amd.utils.machine import machine_stats
     s = machine_stats()
     assert s["rss_gb"] > 0
+
+
+def test_eventlog_file(tmp_path):
+    path = str(tmp_path / "events.jsonl")
+    fv = bs.func(lambda: bs.Const(2, torch.arange(4, dtype=torch.int64)))
+    sess = bs.start(parallelism=2, device="cpu", eventlog_path=path)
+    sess.run(fv)
+    sess.eventer.close()
+    lines = [json.loads(l) for l in open(path)]
+    assert lines[0]["event"] == "bigslice:sessionStart"
+    assert any(l["event"] == "bigslice:taskComplete" and
+               l["state"] == "OK" for l in lines)
+
+
+def test_slicetrace_analyzer(tmp_path):
+    trace = str(tmp_path / "t.json")
+    fv = bs.func(lambda: bs.Reduce(
+        bs.Const(2, torch.arange(10, dtype=torch.int64) % 3,
+                 torch.ones(10, dtype=torch.int64)), "sum"))
+    sess = bs.start(parallelism=2, device="cpu", trace_path=trace)
+    sess.run(fv)
+    sess.shutdown()
+    import subprocess
+    import sys as _sys
+    p = subprocess.run(
+        [_sys.executable, "-m", "bigslice_amd.tools.slicetrace", trace],
+        capture_output=True, text=True)
+    assert p.returncode == 0, p.stderr
+    assert "reduce" in p.stdout
+
+
+def test_stats_map():
+    from bigslice_amd.utils import stats
+    m = stats.Map()
+    m.add("x", 2)
+    m.add("x")
+    m.merge({"y": 5})
+    assert m.values() == {"x": 3, "y": 5}
+    assert "x=3" in str(m)
