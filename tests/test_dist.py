@@ -155,3 +155,47 @@ def test_dist_operator_matrix_gloo():
                              (3, [], [6])]
     assert r0["fold"] == [(1, 3), (2, 12)]
     assert r0["iterative"] == [(1, 6), (2, 24)]
+
+
+def _error_worker(rank, world, port, q):
+    """One rank's UDF raises: EVERY rank must raise (collective error
+    surfacing), and the session must stay usable for a subsequent run."""
+    _init(rank, world, port)
+    import bigslice_amd as bs
+
+    def build(nshard):
+        def gen(shard, ctx):
+            if shard == 1:
+                raise ValueError("boom on shard 1")
+            yield (torch.arange(5, dtype=torch.int64),)
+        return bs.ReaderFunc(nshard, gen, bs.schema_of(int))
+    fv_bad = bs.func(build)
+    fv_ok = bs.func(lambda: bs.Const(2, torch.arange(4,
+                                                     dtype=torch.int64)))
+    sess = bs.start(distributed=True, device="cpu")
+    raised = False
+    try:
+        sess.run(fv_bad, 4)
+    except Exception:
+        raised = True
+    res = sess.run(fv_ok)  # session still usable after the failure
+    rows = sorted(res.scan())
+    q.put((rank, raised, rows))
+
+
+def test_dist_error_propagates_to_all_ranks():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_error_worker, args=(r, 2, 29655, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, raised, rows = q.get()
+        results[rank] = (raised, rows)
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    assert results[0][0] and results[1][0]  # both ranks raised
+    assert results[0][1] == [0, 1, 2, 3]

@@ -270,3 +270,35 @@ def test_stats_map():
     m.merge({"y": 5})
     assert m.values() == {"x": 3, "y": 5}
     assert "x=3" in str(m)
+
+
+def test_concurrent_eval_and_discard_chaos():
+    # TestDiscardChaos analog: interleave runs of the same Func with
+    # discards of prior results; everything stays correct.
+    import threading
+
+    fv = bs.func(lambda: bs.Reduce(
+        bs.Const(4, torch.arange(400, dtype=torch.int64) % 9,
+                 torch.ones(400, dtype=torch.int64)), "sum"))
+    sess = bs.start(parallelism=4, device="cpu")
+    want = {}
+    for k in (torch.arange(400) % 9).tolist():
+        want[k] = want.get(k, 0) + 1
+    errs = []
+
+    def worker():
+        try:
+            for _ in range(5):
+                res = sess.run(fv)
+                got = dict(res.scan())
+                assert got == want, got
+                res.discard()
+        except BaseException as e:
+            errs.append(e)
+
+    ts = [threading.Thread(target=worker) for _ in range(3)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(120)
+    assert not errs, errs
