@@ -389,6 +389,39 @@ torch::Tensor radix_sort_keys(torch::Tensor keys) {
   return keys_out;
 }
 
+// Direct (key, value) sort for 2-column frames: no permutation, no
+// gather passes.  The value column is bit-cast to int64 (it does not
+// participate in ordering).
+std::vector<torch::Tensor> radix_sort_kv(torch::Tensor keys,
+                                         torch::Tensor vals) {
+  TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
+  TORCH_CHECK(vals.is_cuda() && vals.is_contiguous());
+  TORCH_CHECK(vals.element_size() == 8, "8-byte values only");
+  int64_t n = keys.size(0);
+  auto keys_out = torch::empty_like(keys);
+  auto vals_out = torch::empty_like(vals);
+  auto run = [&](auto fn) {
+    size_t temp_bytes = 0;
+    fn(keys.data_ptr(), keys_out.data_ptr(),
+       (const int64_t*)vals.data_ptr(), (int64_t*)vals_out.data_ptr(), n,
+       nullptr, temp_bytes, current_stream());
+    auto temp = torch::empty({(int64_t)temp_bytes},
+                             keys.options().dtype(torch::kUInt8));
+    fn(keys.data_ptr(), keys_out.data_ptr(),
+       (const int64_t*)vals.data_ptr(), (int64_t*)vals_out.data_ptr(), n,
+       temp.data_ptr(), temp_bytes, current_stream());
+  };
+  switch (keys.scalar_type()) {
+    case torch::kInt64: run(radix_sort_pairs_int64_t); break;
+    case torch::kInt32: run(radix_sort_pairs_int32_t); break;
+    case torch::kFloat32: run(radix_sort_pairs_float); break;
+    case torch::kFloat64: run(radix_sort_pairs_double); break;
+    default:
+      TORCH_CHECK(false, "radix_sort_kv: unsupported key dtype");
+  }
+  return {keys_out, vals_out};
+}
+
 torch::Tensor radix_argsort(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
@@ -440,5 +473,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("agg_identity", &agg_identity, "aggregation identity fill");
   m.def("radix_argsort", &radix_argsort, "device radix argsort (K6)");
   m.def("radix_sort_keys", &radix_sort_keys, "device radix key sort");
+  m.def("radix_sort_kv", &radix_sort_kv,
+        "direct (key, 8-byte value) radix sort");
   m.def("slot_pids", &slot_pids, "table-slot-range partition ids");
 }

@@ -314,6 +314,18 @@ def radix_argsort(keys: torch.Tensor) -> torch.Tensor:
     return _C.radix_argsort(keys)
 
 
+def radix_sort_kv(keys: torch.Tensor, val: torch.Tensor):
+    """Direct (key, value) radix sort for one 8-byte value column."""
+    if _C is None:
+        if keys.is_cuda and not ALLOW_FALLBACK:
+            _require("radix_sort_kv")
+        perm = torch.argsort(keys, stable=True)
+        return keys[perm], val[perm]
+    v64 = val.view(torch.int64) if val.dtype != torch.int64 else val
+    sk, sv = _C.radix_sort_kv(keys, v64)
+    return sk, (sv.view(val.dtype) if val.dtype != torch.int64 else sv)
+
+
 def radix_sort_keys(keys: torch.Tensor) -> torch.Tensor:
     """Device radix key-only sort (no permutation)."""
     if _C is None:

@@ -32,7 +32,15 @@ def sort_frame(frame: Frame) -> Frame:
     if (frame.prefix == 1 and isinstance(first, torch.Tensor)
             and first.is_cuda):
         from . import kernels
+        from .frame import Frame as _F
         if kernels.sort_pairs_supported(first):
+            if (frame.num_columns == 2
+                    and isinstance(frame.columns[1], torch.Tensor)
+                    and frame.columns[1].element_size() == 8):
+                # direct (key, value) sort: no permutation/gather passes
+                sk, sv = kernels.radix_sort_kv(
+                    first.contiguous(), frame.columns[1].contiguous())
+                return _F([sk, sv], 1, combined_id=frame.combined_id)
             perm = kernels.radix_argsort(first.contiguous())
             return frame.select(perm)
     return frame.sort_by_prefix()
