@@ -94,27 +94,38 @@ def _key_sort(k):
     return k
 
 
-def _device_cogroup_columns(dep_frames, union_keys):
-    """Sort one dep's rows by key; return per-value-column
-    SegmentedColumns aligned with union_keys (searchsorted segment
-    boundaries over the key-sorted values)."""
+def _segments_for(sorted_keys, value_cols, union_keys):
+    """Per-value-column SegmentedColumns aligned with union_keys
+    (searchsorted segment boundaries over key-sorted values)."""
     import torch
 
-    from ..frame import Frame, SegmentedColumn
-    from .. import kernels
-    from ..sortio import sort_frame
+    from ..frame import SegmentedColumn
 
-    cols = []
-    if dep_frames:
-        f = Frame.concat(dep_frames)
-        f = sort_frame(f)
-        keys = f.columns[0].contiguous()
-        starts = torch.searchsorted(keys, union_keys, right=False)
-        ends = torch.searchsorted(keys, union_keys, right=True)
-        for v in f.columns[1:]:
-            cols.append(SegmentedColumn(v.contiguous(), starts, ends))
-        return cols, keys
-    return cols, None
+    starts = torch.searchsorted(sorted_keys, union_keys, right=False)
+    ends = torch.searchsorted(sorted_keys, union_keys, right=True)
+    return [SegmentedColumn(v.contiguous(), starts, ends)
+            for v in value_cols]
+
+
+def _merge_sorted_unique(a, b):
+    """Sorted-unique union of two SORTED key arrays via searchsorted
+    scatter (2 passes; ~3x cheaper than re-sorting the concatenation)."""
+    import torch
+    n, m = a.shape[0], b.shape[0]
+    if n == 0:
+        merged = b
+    elif m == 0:
+        merged = a
+    else:
+        pos_a = torch.arange(n, device=a.device) +             torch.searchsorted(b, a, right=False)
+        pos_b = torch.arange(m, device=a.device) +             torch.searchsorted(a, b, right=True)
+        merged = torch.empty(n + m, dtype=a.dtype, device=a.device)
+        merged[pos_a] = a
+        merged[pos_b] = b
+    mask = torch.ones(merged.shape[0], dtype=torch.bool,
+                      device=merged.device)
+    mask[1:] = merged[1:] != merged[:-1]
+    return merged[mask]
 
 
 # Attach the device generator to Cogroup (kept separate for readability).
@@ -123,29 +134,42 @@ def _cogroup_device_gen(self, dep_readers, ctx):
 
     from ..frame import Frame, SegmentedColumn
 
-    dep_frames = [[f for f in r] for r in dep_readers]
-    device = ctx.device
-    all_keys = [f.columns[0] for frames in dep_frames for f in frames]
-    if not all_keys:
-        return
-    # sorted-unique union of all dep keys: one radix sort + boundary
-    # compaction (torch.unique re-sorts and computes inverses we don't
-    # need — measured ~4x slower here)
+    from ..sortio import sort_frame
     from .. import kernels
-    cat = torch.cat([k.contiguous() for k in all_keys])
-    if kernels.sort_pairs_supported(cat):
-        sk = kernels.radix_sort_keys(cat)
+
+    device = ctx.device
+    # 1. sort each dep side by key once (direct kv radix for 2-column)
+    sorted_deps = []
+    for r in dep_readers:
+        frames = [f for f in r]
+        sorted_deps.append(sort_frame(Frame.concat(frames))
+                           if frames else None)
+    key_arrays = [sd.columns[0].contiguous()
+                  for sd in sorted_deps if sd is not None]
+    if not key_arrays:
+        return
+    # 2. sorted-unique union: pairwise merge of the already-sorted key
+    # arrays (searchsorted scatter), radix only as a multi-dep fallback
+    if len(key_arrays) == 1:
+        u = key_arrays[0]
+        mask = torch.ones(u.shape[0], dtype=torch.bool, device=u.device)
+        mask[1:] = u[1:] != u[:-1]
+        union_keys = u[mask]
+    elif len(key_arrays) == 2:
+        union_keys = _merge_sorted_unique(key_arrays[0], key_arrays[1])
     else:
-        sk = torch.sort(cat).values
-    mask = torch.ones(sk.shape[0], dtype=torch.bool, device=sk.device)
-    mask[1:] = sk[1:] != sk[:-1]
-    union_keys = sk[mask]
+        cat = torch.cat(key_arrays)
+        sk = kernels.radix_sort_keys(cat)             if kernels.sort_pairs_supported(cat) else torch.sort(cat).values
+        mask = torch.ones(sk.shape[0], dtype=torch.bool,
+                          device=sk.device)
+        mask[1:] = sk[1:] != sk[:-1]
+        union_keys = sk[mask]
+    # 3. per-dep segments over the sorted values
     out_cols = [union_keys]
-    for di, frames in enumerate(dep_frames):
-        nv = self._val_counts[di]
-        segs, _ = _device_cogroup_columns(frames, union_keys)
-        if segs:
-            out_cols.extend(segs)
+    for di, sd in enumerate(sorted_deps):
+        if sd is not None:
+            out_cols.extend(_segments_for(sd.columns[0].contiguous(),
+                                          sd.columns[1:], union_keys))
         else:
             empty_vals = [torch.empty(0, dtype=dt, device=device)
                           for dt in self.deps[di].slice.schema.dtypes[1:]]
