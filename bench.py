@@ -60,19 +60,23 @@ def main():
     ap.add_argument("--nkeys", type=int, default=DEFAULT_NKEYS)
     ap.add_argument("--shards", type=int, default=DEFAULT_SHARDS)
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--trace", type=str, default=None,
+                    help="write a Chrome trace of the run to this path")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     distributed = world > 1
     if distributed:
-        sess = bs.start(distributed=True, device=args.device)
+        sess = bs.start(distributed=True, device=args.device,
+                        trace_path=args.trace)
         device = sess.executor.device
         comm = sess.executor.comm
     else:
         device = args.device or (
             "cuda:0" if torch.cuda.is_available() else "cpu")
-        sess = bs.start(parallelism=args.shards, device=device)
+        sess = bs.start(parallelism=args.shards, device=device,
+                        trace_path=args.trace)
         comm = None
 
     on_gpu = device.startswith("cuda")
@@ -121,6 +125,8 @@ def main():
     ms_per_step = elapsed * 1000.0 / args.steps
     rows_per_sec = (rows_per_shard * nshard) / (ms_per_step / 1000.0)
 
+    if args.trace:
+        sess.shutdown()
     if rank == 0:
         out = {
             "metric": "rows/sec (whole node) Reshuffle+Reduce group-by",
