@@ -267,10 +267,12 @@ void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
   int64_t cap = tkeys.size(0) - 1;
   TORCH_CHECK(cap > 0 && (cap & (cap - 1)) == 0);
   const uint32_t seed = 0x9acb0442u;
-  if (target_blocks <= 0) target_blocks = 4096;
-  int64_t rpb = (n + target_blocks - 1) / target_blocks;
-  if (rpb < 1024) rpb = 1024;
-  int64_t nblocks = (n + rpb - 1) / rpb;
+  // grid-stride geometry (matches the one-level kernel); target_blocks
+  // caps the grid so the LDS flush volume stays bounded
+  if (target_blocks <= 0) target_blocks = 32768;
+  int64_t nblocks = std::min<int64_t>((n + THREADS - 1) / THREADS,
+                                      target_blocks);
+  int64_t rpb = 0;  // unused by the grid-stride kernel
   hipLaunchKernelGGL(k_groupby_insert_sum_i64_lds, dim3((uint32_t)nblocks),
                      dim3(THREADS), 0, current_stream(),
                      keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(), n,

@@ -221,7 +221,7 @@ class GroupTable:
                 _C.groupby_insert_lds(keys[:prefix], vals[0][:prefix],
                                       self.tkeys, self.tabs[0],
                                       self.flags, MAX_PROBES, -1,
-                                      max(prefix // 16384, 128), hits)
+                                      32768, hits)
                 frac_ok = int(hits.item()) * 8 >= prefix  # one sync
                 mode = self._mode = "lds" if frac_ok else "global"
                 keys = keys[prefix:]
@@ -230,7 +230,8 @@ class GroupTable:
                     return
         if mode == "lds":
             empty = torch.empty(0, dtype=torch.int32, device=self.device)
-            blocks = int(os.environ.get("BIGSLICE_GB_LDS_BLOCKS", "4096"))
+            blocks = int(os.environ.get("BIGSLICE_GB_LDS_BLOCKS",
+                                        "32768"))
             _C.groupby_insert_lds(keys, vals[0], self.tkeys,
                                   self.tabs[0], self.flags, MAX_PROBES,
                                   1, blocks, empty)
