@@ -218,11 +218,15 @@ class GroupTable:
                 prefix = min(n, self._SAMPLE_ROWS)
                 hits = torch.zeros(1, dtype=torch.int32,
                                    device=self.device)
+                # few enough blocks that every thread runs the full
+                # 16-iteration sample phase
+                blocks = max(prefix // (256 * 16), 64)
                 _C.groupby_insert_lds(keys[:prefix], vals[0][:prefix],
                                       self.tkeys, self.tabs[0],
                                       self.flags, MAX_PROBES, -1,
-                                      32768, hits)
-                frac_ok = int(hits.item()) * 8 >= prefix  # one sync
+                                      blocks, hits)
+                sampled = min(prefix, blocks * 256 * 16)
+                frac_ok = int(hits.item()) * 8 >= sampled  # one sync
                 mode = self._mode = "lds" if frac_ok else "global"
                 keys = keys[prefix:]
                 vals = [vals[0][prefix:]]
