@@ -234,8 +234,10 @@ class GroupTable:
                     return
         if mode == "lds":
             empty = torch.empty(0, dtype=torch.int32, device=self.device)
+            # hot-key regime: fewer blocks bound the per-launch LDS
+            # flush volume (blocks x slots contended adds)
             blocks = int(os.environ.get("BIGSLICE_GB_LDS_BLOCKS",
-                                        "32768"))
+                                        "4096"))
             _C.groupby_insert_lds(keys, vals[0], self.tkeys,
                                   self.tabs[0], self.flags, MAX_PROBES,
                                   1, blocks, empty)
