@@ -232,6 +232,25 @@ class Session:
             slice_ = funcv.invoke(args)
             compiler = Compiler(inv_index, self.env)
             tasks = compiler.compile(slice_)
+            if funcv.exclusive:
+                # Exclusive Funcs (func.go Exclusive): their tasks do
+                # not share the executor with others (locally: take all
+                # proc permits; reference: a dedicated cluster).
+                from ..ops.slice_base import Pragma
+                seen = set()
+
+                def mark(t):
+                    if id(t) in seen:
+                        return
+                    seen.add(id(t))
+                    p = t.pragma or Pragma()
+                    t.pragma = Pragma(procs=p.procs, exclusive=True,
+                                      materialize=p.materialize)
+                    for dep in t.deps:
+                        for h in dep.head_tasks:
+                            mark(h)
+                for t in tasks:
+                    mark(t)
             self.env.freeze()
             ev = getattr(self.executor, "evaluate", None)
             if ev is not None:

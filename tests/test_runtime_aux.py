@@ -302,3 +302,50 @@ def test_concurrent_eval_and_discard_chaos():
     for t in ts:
         t.join(120)
     assert not errs, errs
+
+
+def test_exclusive_func_serializes_tasks():
+    import threading
+    import time as _time
+    active, peak = [], [0]
+    lock = threading.Lock()
+
+    def track(x):
+        with lock:
+            active.append(1)
+            peak[0] = max(peak[0], len(active))
+        _time.sleep(0.02)
+        with lock:
+            active.pop()
+        return (x,)
+
+    fv = bs.func(lambda: bs.Map(
+        bs.Const(4, torch.arange(8, dtype=torch.int64)), track,
+        out_schema=(int,)), exclusive=True)
+    sess = bs.start(parallelism=4, device="cpu")
+    res = sess.run(fv)
+    assert len(list(res.scan())) == 8
+    assert peak[0] == 1
+
+
+def test_reader_helpers():
+    from bigslice_amd.sliceio import (ErrReader, FuncReader, Scanner,
+                                      read_all_or_empty, EmptyReader)
+    from bigslice_amd.schema import Schema
+    with pytest.raises(ValueError):
+        ErrReader(ValueError("x")).read()
+    calls = []
+
+    def fn():
+        if calls:
+            return None
+        calls.append(1)
+        from bigslice_amd.frame import Frame
+        return Frame([torch.arange(3, dtype=torch.int64)])
+    fr = FuncReader(fn)
+    frames = list(fr)
+    assert len(frames) == 1
+    empty = read_all_or_empty(EmptyReader(), Schema([torch.int64]))
+    assert len(empty) == 0
+    s = Scanner(EmptyReader())
+    assert list(s.frames()) == []
