@@ -27,9 +27,6 @@ struct ValCols {
   int n;
 };
 
-template <typename T>
-__device__ __forceinline__ void atomic_cas_combine(T* addr, T v, int agg);
-
 template <typename T, typename U>
 __device__ __forceinline__ void cas_loop(T* addr, T v, int agg) {
   U* a = (U*)addr;
