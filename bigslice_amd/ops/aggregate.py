@@ -58,6 +58,12 @@ class Aggregation:
     def all_builtin(self) -> bool:
         return all(isinstance(a, str) for a in self.aggs)
 
+    def key(self):
+        """Structural identity for compiler memoization: equal specs
+        share producer tasks, different specs must not."""
+        return tuple(a if isinstance(a, str) else id(a)
+                     for a in self.aggs)
+
     def __repr__(self):
         return f"Aggregation({self.aggs})"
 

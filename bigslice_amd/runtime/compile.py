@@ -100,7 +100,14 @@ class Compiler:
         if isinstance(target, Result):
             return self._reuse_result(target, num_partitions, partitioner,
                                       combiner, shuffle_out)
-        key = (id(slice_), num_partitions, shuffle_out)
+        # Memoization key: a producer compiled for a shuffle carries the
+        # consumer's partitioner AND combiner, so consumers with
+        # different combine specs (e.g. Reduce sum vs max over one
+        # slice) must compile separate producer tasks.
+        comb_key = combiner.key() if combiner is not None else None
+        key = (id(slice_), num_partitions, shuffle_out,
+               id(partitioner) if partitioner is not None else None,
+               comb_key)
         if key in self.memo:
             return self.memo[key]
 

@@ -41,8 +41,22 @@ def test_shuffle_breaks_pipeline():
           inv1_const_map@2:1""")
 
 
-def test_branch_memoized():
-    # two consumers of one producer compile the producer once
+def test_branch_memoized_same_combiner():
+    # two consumers with IDENTICAL combine specs share producer tasks
+    base = bs.Map(bs.Const(2, torch.arange(4, dtype=torch.int64),
+                           torch.ones(4, dtype=torch.int64)),
+                  lambda k, v: (k, v))
+    r1 = bs.Reduce(base, "sum")
+    r2 = bs.Reduce(base, "sum")
+    c = Compiler(1)
+    t1 = c.compile(r1)
+    t2 = c.compile(r2)
+    assert t1[0].deps[0].head_tasks[0] is t2[0].deps[0].head_tasks[0]
+
+
+def test_branch_not_memoized_across_combiners():
+    # different combine specs must NOT share producers (pre-combine
+    # would apply the wrong aggregation)
     base = bs.Map(bs.Const(2, torch.arange(4, dtype=torch.int64),
                            torch.ones(4, dtype=torch.int64)),
                   lambda k, v: (k, v))
@@ -51,8 +65,7 @@ def test_branch_memoized():
     c = Compiler(1)
     t1 = c.compile(r1)
     t2 = c.compile(r2)
-    # both reduces share the same producer task objects
-    assert t1[0].deps[0].head_tasks[0] is t2[0].deps[0].head_tasks[0]
+    assert t1[0].deps[0].head_tasks[0] is not t2[0].deps[0].head_tasks[0]
 
 
 def test_materialize_pragma_breaks_pipeline():

@@ -288,3 +288,17 @@ def test_precombined_passthrough_correctness():
     for k, v in zip(keys.tolist(), vals.tolist()):
         want[k] = want.get(k, 0) + v
     assert dict(on.scan()) == want
+
+
+def test_mixed_combiners_over_shared_producer():
+    # regression: Reduce("sum") and Reduce("max") branching from ONE
+    # slice must not share pre-combined producer output.
+    keys = torch.tensor([1, 1, 2], dtype=torch.int64)
+    vals = torch.tensor([5, 7, 3], dtype=torch.int64)
+
+    def build():
+        base = bs.Map(bs.Const(2, keys, vals), lambda k, v: (k, v))
+        return bs.Cogroup(bs.Reduce(base, "sum"), bs.Reduce(base, "max"))
+
+    res = bs.slicetest.run(build)
+    assert sorted(res.scan()) == [(1, [12], [7]), (2, [3], [3])]
