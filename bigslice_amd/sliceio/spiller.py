@@ -94,6 +94,7 @@ class Spiller:
     def spill(self, frame: Frame) -> int:
         """Spill one batch; returns bytes spilled.  Device batches copy
         out asynchronously on the copy stream."""
+        self._release_completed()
         nbytes = frame.nbytes()
         self.rows += len(frame)
         if self.host_used + nbytes <= self.host_budget:
@@ -134,6 +135,15 @@ class Spiller:
             done = torch.cuda.Event()
             done.record()
         return _HostBatch(Frame(cols, frame.prefix), done, src_refs)
+
+    def _release_completed(self) -> None:
+        """Drop device-source references of finished D2H copies so
+        spilling actually relieves HBM pressure during the build phase
+        (not only at readback)."""
+        for b in self.batches:
+            if isinstance(b, _HostBatch) and b.src_refs is not None:
+                if b.event is None or b.event.query():
+                    b.src_refs = None
 
     def num_batches(self) -> int:
         return len(self.batches)
