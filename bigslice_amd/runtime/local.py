@@ -223,6 +223,9 @@ class LocalExecutor(Executor):
     def discard(self, task: Task) -> None:
         self.store.discard_task(task.name)
 
+    def shutdown(self) -> None:
+        self.pool.shutdown(wait=False)
+
     def merged_scope(self, tasks):
         """Merge task scopes across the reachable graph
         (Result.Scope semantics, exec/session.go:418-426)."""

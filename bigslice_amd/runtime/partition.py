@@ -39,7 +39,9 @@ def split_frame(frame: Frame, num_partitions: int,
         return [None] * num_partitions
     if frame.device != "cpu" and not frame.has_objects:
         from .. import kernels
-        if kernels.partition_supported(frame):
+        # the LDS histogram kernel supports up to 4096 partitions; wider
+        # fans fall back to the sort-based split below
+        if num_partitions <= 4096 and kernels.partition_supported(frame):
             return kernels.partition_frame(frame, num_partitions,
                                            partitioner)
     p = partition_ids(frame, num_partitions, partitioner)
