@@ -24,6 +24,8 @@ class TarReader(Slice):
 
     def reader(self, shard: int, dep_readers, ctx: TaskContext) -> Reader:
         nshard = self.num_shards
+        from .. import config
+        chunk = min(ctx.chunk, config.HOST_CHUNK_ROWS)
 
         def gen():
             names: List[str] = []
@@ -34,7 +36,7 @@ class TarReader(Slice):
                         continue
                     names.append(entry.name)
                     datas.append(tf.extractfile(entry).read())
-                    if len(names) >= ctx.chunk:
+                    if len(names) >= chunk:
                         yield Frame([list(names), list(datas)], prefix=1)
                         names.clear()
                         datas.clear()

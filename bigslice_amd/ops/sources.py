@@ -112,6 +112,10 @@ class ScanReader(Slice):
 
     def reader(self, shard: int, dep_readers, ctx: TaskContext) -> Reader:
         nshard = self.num_shards
+        from .. import config
+        # object columns are host-resident regardless of ctx device:
+        # batch at host granularity
+        chunk = min(ctx.chunk, config.HOST_CHUNK_ROWS)
 
         def gen():
             buf: List[str] = []
@@ -119,7 +123,7 @@ class ScanReader(Slice):
                 if i % nshard != shard:
                     continue
                 buf.append(line.rstrip("\n"))
-                if len(buf) >= ctx.chunk:
+                if len(buf) >= chunk:
                     yield Frame([list(buf)])
                     buf.clear()
             if buf:
