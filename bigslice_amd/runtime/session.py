@@ -230,7 +230,14 @@ class Session:
                 self._inv_counter += 1
                 inv_index = self._inv_counter
             slice_ = funcv.invoke(args)
-            compiler = Compiler(inv_index, self.env)
+            # fresh CompileEnv per run: cache decisions are re-checked
+            # each invocation (exec/compile.go recomputes presence) and
+            # concurrent runs must not share frozen state.  NOTE for
+            # distributed mode: every rank computes decisions against
+            # the same filesystem; a rank-divergent cache view would
+            # need the reference's env-broadcast (tracked in NOTES.md).
+            env = CompileEnv()
+            compiler = Compiler(inv_index, env)
             tasks = compiler.compile(slice_)
             if funcv.exclusive:
                 # Exclusive Funcs (func.go Exclusive): their tasks do
@@ -251,13 +258,12 @@ class Session:
                             mark(h)
                 for t in tasks:
                     mark(t)
-            self.env.freeze()
+            env.freeze()
             ev = getattr(self.executor, "evaluate", None)
             if ev is not None:
                 ev(tasks)
             else:
                 evaluate(self.executor, tasks)
-            self.env = CompileEnv()  # fresh cache decisions per run
             return Result(self, slice_, tasks)
         finally:
             _mark_busy(False)

@@ -349,3 +349,24 @@ def test_reader_helpers():
     assert len(empty) == 0
     s = Scanner(EmptyReader())
     assert list(s.frames()) == []
+
+
+def test_materialize_executes_as_checkpoint():
+    computed = []
+
+    def gen(shard, ctx):
+        computed.append(shard)
+        yield (torch.arange(3, dtype=torch.int64) + shard * 10,)
+
+    def build():
+        src = bs.ReaderFunc(2, gen, bs.schema_of(int))
+        bs.materialize(src)
+        a = bs.Map(src, lambda x: (x + 1,))
+        b = bs.Map(src, lambda x: (x * 2,))
+        return bs.Cogroup(bs.Map(a, lambda x: (x, x)),
+                          bs.Map(b, lambda x: (x, x)))
+
+    res = bs.slicetest.run(build)
+    assert len(list(res.scan())) > 0
+    # materialized source computed once per shard despite two consumers
+    assert sorted(computed) == [0, 1]
