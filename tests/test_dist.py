@@ -60,7 +60,16 @@ def _wordcount_worker(rank, world, port, q):
     q.put((rank, sorted(res.scan())))
 
 
-def _run_workers(fn, world=2, port=29611):
+def _free_port():
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _run_workers(fn, world=2, port=None):
+    if port is None:
+        port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=fn, args=(r, world, port, q))
@@ -78,7 +87,7 @@ def _run_workers(fn, world=2, port=29611):
 
 
 def test_dist_reduce_gloo():
-    results = _run_workers(_reduce_worker, port=29611)
+    results = _run_workers(_reduce_worker)
     # each of 4 shards contributes 100 rows over 7 keys; values shard+1
     # total per key: sum over shards of (count of key in shard) * (shard+1)
     keys = (torch.arange(100, dtype=torch.int64) % 7)
@@ -92,7 +101,7 @@ def test_dist_reduce_gloo():
 
 
 def test_dist_wordcount_gloo():
-    results = _run_workers(_wordcount_worker, port=29627)
+    results = _run_workers(_wordcount_worker)
     assert results[0] == [("a", 3), ("b", 2), ("c", 1), ("d", 1)]
 
 
@@ -147,7 +156,7 @@ def _matrix_worker(rank, world, port, q):
 
 
 def test_dist_operator_matrix_gloo():
-    results = _run_workers(_matrix_worker, port=29641)
+    results = _run_workers(_matrix_worker)
     r0 = results[0]
     keys = (torch.arange(200) % 11).tolist()
     assert r0["reshuffle"] == sorted(zip(keys, range(200)))
@@ -186,7 +195,8 @@ def _error_worker(rank, world, port, q):
 def test_dist_error_propagates_to_all_ranks():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_error_worker, args=(r, 2, 29655, q))
+    port = _free_port()
+    procs = [ctx.Process(target=_error_worker, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
         p.start()

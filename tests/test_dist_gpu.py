@@ -82,7 +82,11 @@ def _gpu_dist_worker(rank, world, port, q):
 def test_spmd_two_procs_one_gpu_gloo():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_gpu_dist_worker, args=(r, 2, 29735, q))
+    import socket
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        port = sk.getsockname()[1]
+    procs = [ctx.Process(target=_gpu_dist_worker, args=(r, 2, port, q))
              for r in range(2)]
     for p in procs:
         p.start()
