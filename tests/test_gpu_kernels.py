@@ -233,3 +233,31 @@ def test_reduce_keys_only_gpu():
     cpu = bs.start(parallelism=2, device="cpu").run(fv)
     gpu = bs.start(parallelism=2, device="cuda:0").run(fv)
     assert sorted(cpu.scan()) == sorted(gpu.scan())
+
+
+def test_packed_groupby_variant(kernels, monkeypatch):
+    # experimental packed-slot layout stays correct
+    monkeypatch.setenv("BIGSLICE_GB_PACKED", "1")
+    keys = torch.randint(0, 5000, (1_000_000,), dtype=torch.int64,
+                         device="cuda:0")
+    vals = torch.randint(0, 100, (1_000_000,), dtype=torch.int64,
+                         device="cuda:0")
+    uk, outs = kernels.groupby(keys, [vals], ["sum"])
+    ref_uk, inv = torch.unique(keys, return_inverse=True)
+    ref = torch.zeros(ref_uk.shape[0], dtype=torch.int64,
+                      device="cuda:0").scatter_reduce_(
+        0, inv, vals, reduce="sum", include_self=False)
+    order = torch.argsort(uk)
+    assert torch.equal(uk[order], ref_uk)
+    assert torch.equal(outs[0][order], ref)
+
+
+def test_forced_lds_and_global_modes(kernels, monkeypatch):
+    keys = torch.randint(0, 200, (2_000_000,), dtype=torch.int64,
+                         device="cuda:0")
+    vals = torch.ones_like(keys)
+    for mode in ("lds", "global"):
+        monkeypatch.setenv("BIGSLICE_GB_MODE", mode)
+        uk, outs = kernels.groupby(keys, [vals], ["sum"])
+        assert int(outs[0].sum()) == 2_000_000
+        assert uk.shape[0] <= 200
