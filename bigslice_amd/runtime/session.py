@@ -286,7 +286,8 @@ class Session:
 
 def start(parallelism: int = None, device: str = None,
           executor: Executor = None, distributed: bool = None,
-          trace_path: str = None, eventlog_path: str = None) -> Session:
+          trace_path: str = None, eventlog_path: str = None,
+          checkpoint_dir: str = None) -> Session:
     """Create a session (exec.Start analog).
 
     distributed=True (or WORLD_SIZE>1 in the environment) starts the SPMD
@@ -303,9 +304,19 @@ def start(parallelism: int = None, device: str = None,
             from .dist import DistExecutor
             comm = init_comm(device=device)
             comm.check_registry(registry_digest())
-            executor = DistExecutor(comm)
+            store = None
+            if checkpoint_dir is not None:
+                from .store import FileStore
+                store = FileStore(os.path.join(
+                    checkpoint_dir, f"rank{comm.rank:03d}"))
+            executor = DistExecutor(comm, store=store)
         else:
             from .local import LocalExecutor
-            executor = LocalExecutor(parallelism=parallelism, device=device)
+            store = None
+            if checkpoint_dir is not None:
+                from .store import FileStore
+                store = FileStore(checkpoint_dir)
+            executor = LocalExecutor(parallelism=parallelism,
+                                     device=device, store=store)
     return Session(executor, parallelism, trace_path=trace_path,
                    eventlog_path=eventlog_path)
