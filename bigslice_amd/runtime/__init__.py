@@ -1,6 +1,7 @@
 """Execution runtime exports."""
 
 from .compile import CompileEnv, Compiler, pipeline_slices  # noqa
+from .dist import DistExecutor  # noqa
 from .eval import Executor, TooManyTriesError, evaluate  # noqa
 from .local import LocalExecutor, TaskLost  # noqa
 from .session import (FuncValue, Invocation, Result, Session, func,  # noqa
