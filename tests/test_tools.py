@@ -146,3 +146,18 @@ def test_typecheck_tool(tmp_path):
     assert p.returncode == 1
     assert "takes 2 argument(s)" in p.stdout
     assert p.stdout.count("\n") == 1  # only the bad call flagged
+
+
+def test_kmeans_example_converges():
+    sys.path.insert(0, "examples")
+    import importlib
+    km = importlib.import_module("kmeans")
+    torch.manual_seed(0)
+    g = torch.Generator().manual_seed(1)
+    for s in range(4):
+        km._POINTS[s] = (torch.rand(2000, generator=g),
+                         torch.rand(2000, generator=g))
+    sess = bs.start(parallelism=4, device="cpu")
+    cx, cy = km.run_kmeans(sess, 4, 4, 8, "cpu")
+    assert len(cx) == 4
+    assert all(0.0 <= c <= 1.0 for c in cx + cy)
