@@ -67,3 +67,11 @@ class Ticker:
         self._thread.join(timeout=1)
         print(format_status(self.roots), file=self.out)
         return False
+
+
+def top_n(counts, n: int):
+    """Largest-n (key, count) pairs (reference exec/topn.go:36-57's
+    diagnostics heap): used by the debug pages to show the heaviest
+    task groups / counters."""
+    import heapq
+    return heapq.nlargest(n, counts.items(), key=lambda kv: (kv[1], kv[0]))

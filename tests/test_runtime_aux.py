@@ -370,3 +370,10 @@ def test_materialize_executes_as_checkpoint():
     assert len(list(res.scan())) > 0
     # materialized source computed once per shard despite two consumers
     assert sorted(computed) == [0, 1]
+
+
+def test_top_n():
+    from bigslice_amd.utils.status import top_n
+    counts = {"a": 5, "b": 9, "c": 1, "d": 9}
+    assert top_n(counts, 2) == [("d", 9), ("b", 9)]
+    assert top_n(counts, 10)[-1] == ("c", 1)
