@@ -98,7 +98,8 @@ std::tuple<std::vector<torch::Tensor>, torch::Tensor> partition_common(
   auto opts64 = cols[0].options().dtype(torch::kInt64);
   auto block_off = torch::empty({nblocks * nparts}, opts64);
   auto part_counts = torch::empty({nparts}, opts64);
-  hipLaunchKernelGGL(k_part_offsets, dim3(1), dim3(THREADS), 0,
+  // 1024 threads = 16 waves; the kernel assigns one wave per partition
+  hipLaunchKernelGGL(k_part_offsets, dim3(1), dim3(1024), 0,
                      current_stream(), block_hist.data_ptr<uint32_t>(),
                      nblocks, (int32_t)nparts,
                      block_off.data_ptr<int64_t>(),
@@ -367,17 +368,37 @@ std::vector<torch::Tensor> groupby_compact(torch::Tensor tkeys,
 
 // ---------------------------------------------------------------- sort
 
+// Reduced sort width: integer keys that are all non-negative order
+// correctly on their value bits alone (the sign-flip bit rocPRIM's key
+// codec adds is constant across them), so one cheap aminmax buys fewer
+// radix passes — the common case of dense ids/categories (20-bit keys:
+// 3 passes instead of 8).  Floats and negative keys sort full-width.
+static int sort_end_bit(const torch::Tensor& keys) {
+  const int width = (int)keys.element_size() * 8;
+  if (keys.numel() < (1 << 20)) return width;
+  auto st = keys.scalar_type();
+  if (st != torch::kInt64 && st != torch::kInt32) return width;
+  auto mm = torch::aminmax(keys);
+  const int64_t mn = std::get<0>(mm).item<int64_t>();
+  if (mn < 0) return width;
+  const int64_t mx = std::get<1>(mm).item<int64_t>();
+  int bits = 1;
+  while (bits < width && (mx >> bits) != 0) ++bits;
+  return bits;
+}
+
 torch::Tensor radix_sort_keys(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
+  const int end_bit = sort_end_bit(keys);
   auto keys_out = torch::empty_like(keys);
   auto run = [&](auto fn) {
     size_t temp_bytes = 0;
-    fn(keys.data_ptr(), keys_out.data_ptr(), n, nullptr, temp_bytes,
-       current_stream());
+    fn(keys.data_ptr(), keys_out.data_ptr(), n, end_bit, nullptr,
+       temp_bytes, current_stream());
     auto temp = torch::empty({(int64_t)temp_bytes},
                              keys.options().dtype(torch::kUInt8));
-    fn(keys.data_ptr(), keys_out.data_ptr(), n, temp.data_ptr(),
+    fn(keys.data_ptr(), keys_out.data_ptr(), n, end_bit, temp.data_ptr(),
        temp_bytes, current_stream());
   };
   switch (keys.scalar_type()) {
@@ -400,18 +421,19 @@ std::vector<torch::Tensor> radix_sort_kv(torch::Tensor keys,
   TORCH_CHECK(vals.is_cuda() && vals.is_contiguous());
   TORCH_CHECK(vals.element_size() == 8, "8-byte values only");
   int64_t n = keys.size(0);
+  const int end_bit = sort_end_bit(keys);
   auto keys_out = torch::empty_like(keys);
   auto vals_out = torch::empty_like(vals);
   auto run = [&](auto fn) {
     size_t temp_bytes = 0;
     fn(keys.data_ptr(), keys_out.data_ptr(),
        (const int64_t*)vals.data_ptr(), (int64_t*)vals_out.data_ptr(), n,
-       nullptr, temp_bytes, current_stream());
+       end_bit, nullptr, temp_bytes, current_stream());
     auto temp = torch::empty({(int64_t)temp_bytes},
                              keys.options().dtype(torch::kUInt8));
     fn(keys.data_ptr(), keys_out.data_ptr(),
        (const int64_t*)vals.data_ptr(), (int64_t*)vals_out.data_ptr(), n,
-       temp.data_ptr(), temp_bytes, current_stream());
+       end_bit, temp.data_ptr(), temp_bytes, current_stream());
   };
   switch (keys.scalar_type()) {
     case torch::kInt64: run(radix_sort_pairs_int64_t); break;
@@ -427,6 +449,7 @@ std::vector<torch::Tensor> radix_sort_kv(torch::Tensor keys,
 torch::Tensor radix_argsort(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
+  const int end_bit = sort_end_bit(keys);
   auto perm_in = torch::arange(n, keys.options().dtype(torch::kInt64));
   auto perm_out = torch::empty_like(perm_in);
   auto keys_out = torch::empty_like(keys);
@@ -434,12 +457,12 @@ torch::Tensor radix_argsort(torch::Tensor keys) {
     size_t temp_bytes = 0;
     fn(keys.data_ptr(), keys_out.data_ptr(),
        perm_in.data_ptr<int64_t>(), perm_out.data_ptr<int64_t>(), n,
-       nullptr, temp_bytes, current_stream());
+       end_bit, nullptr, temp_bytes, current_stream());
     auto temp = torch::empty({(int64_t)temp_bytes},
                              keys.options().dtype(torch::kUInt8));
     fn(keys.data_ptr(), keys_out.data_ptr(),
        perm_in.data_ptr<int64_t>(), perm_out.data_ptr<int64_t>(), n,
-       temp.data_ptr(), temp_bytes, current_stream());
+       end_bit, temp.data_ptr(), temp_bytes, current_stream());
   };
   switch (keys.scalar_type()) {
     case torch::kInt64: run(radix_sort_pairs_int64_t); break;

@@ -261,3 +261,20 @@ def test_forced_lds_and_global_modes(kernels, monkeypatch):
         uk, outs = kernels.groupby(keys, [vals], ["sum"])
         assert int(outs[0].sum()) == 2_000_000
         assert uk.shape[0] <= 200
+
+
+def test_radix_sort_reduced_bits(kernels):
+    # >1M rows with small non-negative keys triggers the reduced
+    # end_bit path; negative keys must fall back to full width.
+    n = 2_000_000
+    for lo, hi in ((0, 1000), (0, 2**40), (-50, 50)):
+        k = torch.randint(lo, hi, (n,), dtype=torch.int64, device="cuda:0")
+        v = torch.arange(n, dtype=torch.int64, device="cuda:0")
+        ks, vs = kernels.radix_sort_kv(k, v)
+        ref_k, ref_perm = torch.sort(k, stable=True)
+        assert torch.equal(ks, ref_k), (lo, hi)
+        # values must travel with their keys (pairing, not order)
+        assert torch.equal(k[vs], ks), (lo, hi)
+        perm = kernels.radix_argsort(k)
+        assert torch.equal(k[perm], ref_k), (lo, hi)
+        assert torch.equal(kernels.radix_sort_keys(k), ref_k), (lo, hi)
