@@ -278,3 +278,61 @@ def test_radix_sort_reduced_bits(kernels):
         perm = kernels.radix_argsort(k)
         assert torch.equal(k[perm], ref_k), (lo, hi)
         assert torch.equal(kernels.radix_sort_keys(k), ref_k), (lo, hi)
+
+
+def test_grouptable_sort_combine_paths(kernels):
+    # Force both combine paths and check they agree with torch.
+    import os
+    n, nkeys = 3_000_000, 500_000
+    k = torch.randint(0, nkeys, (n,), dtype=torch.int64, device="cuda:0")
+    v = torch.randint(-100, 100, (n,), dtype=torch.int64, device="cuda:0")
+    ref_uk, inv = torch.unique(k, return_inverse=True)
+    ref_sum = torch.zeros_like(ref_uk).index_add_(0, inv, v)
+    for mode in ("sort", "hash", "auto"):
+        os.environ["BIGSLICE_GB_COMBINE"] = mode
+        try:
+            t = kernels.GroupTable([torch.int64], ["sum"],
+                                   torch.device("cuda:0"))
+            for off in range(0, n, 1_000_000):
+                t.insert(k[off:off + 1_000_000],
+                         [v[off:off + 1_000_000]])
+            uk, (us,) = t.finish()
+        finally:
+            del os.environ["BIGSLICE_GB_COMBINE"]
+        order = torch.argsort(uk)
+        assert torch.equal(uk[order], ref_uk), mode
+        assert torch.equal(us[order], ref_sum), mode
+
+
+def test_grouptable_mixed_provisional_sort(kernels):
+    # First large batch is hash-inserted provisionally; wide key space
+    # flips the decision to sort for the rest; finish merges both.
+    n, nkeys = 9_000_000, 8_000_000
+    k = torch.randint(0, nkeys, (n,), dtype=torch.int64, device="cuda:0")
+    v = torch.ones(n, dtype=torch.int64, device="cuda:0")
+    t = kernels.GroupTable([torch.int64], ["sum"], torch.device("cuda:0"))
+    t.insert(k[:3_000_000], [v[:3_000_000]])     # provisional (pending)
+    t.insert(k[3_000_000:6_000_000], [v[3_000_000:6_000_000]])
+    t.insert(k[6_000_000:], [v[6_000_000:]])
+    uk, (us,) = t.finish()
+    ref_uk, counts = torch.unique(k, return_counts=True)
+    order = torch.argsort(uk)
+    assert torch.equal(uk[order], ref_uk)
+    assert torch.equal(us[order], counts)
+
+
+def test_grouptable_hotkeys_streams_lds(kernels):
+    # Hot keys: decision lands on the LDS tier; results still exact.
+    n = 6_000_000
+    k = torch.randint(0, 500, (n,), dtype=torch.int64, device="cuda:0")
+    v = torch.randint(-10, 10, (n,), dtype=torch.int64, device="cuda:0")
+    t = kernels.GroupTable([torch.int64], ["sum"], torch.device("cuda:0"))
+    for off in range(0, n, 3_000_000):
+        t.insert(k[off:off + 3_000_000], [v[off:off + 3_000_000]])
+    assert t._mode in ("pending", "lds")
+    uk, (us,) = t.finish()
+    ref_uk, inv = torch.unique(k, return_inverse=True)
+    ref = torch.zeros_like(ref_uk).index_add_(0, inv, v)
+    order = torch.argsort(uk)
+    assert torch.equal(uk[order], ref_uk)
+    assert torch.equal(us[order], ref)
