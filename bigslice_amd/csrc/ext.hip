@@ -446,6 +446,35 @@ std::vector<torch::Tensor> radix_sort_kv(torch::Tensor keys,
   return {keys_out, vals_out};
 }
 
+// K16: (unique_keys, sums, count) over key-sorted int64 pairs.  count
+// comes back as a 1-element device tensor; the caller slices after one
+// sync.
+std::vector<torch::Tensor> segment_sum_sorted_i64(torch::Tensor keys,
+                                                  torch::Tensor vals) {
+  TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
+  TORCH_CHECK(vals.is_cuda() && vals.is_contiguous());
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64 &&
+              vals.scalar_type() == torch::kInt64,
+              "segment_sum_sorted_i64: int64 only");
+  int64_t n = keys.size(0);
+  auto uniq = torch::empty_like(keys);
+  auto sums = torch::empty_like(vals);
+  auto count = torch::zeros({1}, keys.options());
+  if (n == 0) return {uniq, sums, count};
+  size_t temp_bytes = 0;
+  reduce_by_key_sum_i64(keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(),
+                        n, uniq.data_ptr<int64_t>(),
+                        sums.data_ptr<int64_t>(), count.data_ptr<int64_t>(),
+                        nullptr, temp_bytes, current_stream());
+  auto temp = torch::empty({(int64_t)temp_bytes},
+                           keys.options().dtype(torch::kUInt8));
+  reduce_by_key_sum_i64(keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(),
+                        n, uniq.data_ptr<int64_t>(),
+                        sums.data_ptr<int64_t>(), count.data_ptr<int64_t>(),
+                        temp.data_ptr(), temp_bytes, current_stream());
+  return {uniq, sums, count};
+}
+
 torch::Tensor radix_argsort(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
@@ -498,6 +527,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("agg_identity", &agg_identity, "aggregation identity fill");
   m.def("radix_argsort", &radix_argsort, "device radix argsort (K6)");
   m.def("radix_sort_keys", &radix_sort_keys, "device radix key sort");
+  m.def("segment_sum_sorted", &segment_sum_sorted_i64,
+        "reduce-by-key sum over sorted int64 pairs (K16)");
   m.def("radix_sort_kv", &radix_sort_kv,
         "direct (key, 8-byte value) radix sort");
   m.def("slot_pids", &slot_pids, "table-slot-range partition ids");

@@ -61,3 +61,21 @@ INSTANTIATE_SORT_KEYS(int64_t)
 INSTANTIATE_SORT_KEYS(int32_t)
 INSTANTIATE_SORT_KEYS(float)
 INSTANTIATE_SORT_KEYS(double)
+
+// K16: segment reduce over key-sorted (or run-grouped) pairs via
+// rocPRIM's tuned reduce-by-key (decoupled-lookback single pass).
+// Replaces the mask/nonzero/cumsum torch chain in the sort-combine.
+#include <rocprim/device/device_reduce_by_key.hpp>
+
+void reduce_by_key_sum_i64(const int64_t* keys, const int64_t* vals,
+                           int64_t n, int64_t* uniq_out, int64_t* sums_out,
+                           int64_t* count_out, void* temp,
+                           size_t& temp_bytes, hipStream_t s) {
+  hipError_t err = rocprim::reduce_by_key(
+      temp, temp_bytes, keys, vals, (size_t)n, uniq_out, sums_out,
+      count_out, rocprim::plus<int64_t>(), rocprim::equal_to<int64_t>(),
+      s);
+  if (err != hipSuccess)
+    throw std::runtime_error(std::string("rocprim::reduce_by_key: ") +
+                             hipGetErrorString(err));
+}

@@ -345,7 +345,6 @@ class GroupTable:
         batches, self._deferred = self._deferred, []
         keys = (batches[0][0] if len(batches) == 1
                 else torch.cat([k for k, _ in batches]))
-        n = keys.shape[0]
         if mode is None:
             self._start_sample(keys)
             mode = self._mode = self._read_sample()
@@ -368,20 +367,16 @@ class GroupTable:
                 col = (batches[0][1][c] if len(batches) == 1
                        else torch.cat([v[c] for _, v in batches]))
                 vcols.append(col[perm])
-        mask = torch.empty(n, dtype=torch.bool, device=ks.device)
-        mask[0] = True
-        torch.ne(ks[1:], ks[:-1], out=mask[1:])
-        starts = mask.nonzero(as_tuple=True)[0]
-        uk = ks[starts]
-        ends = torch.cat([starts[1:], starts.new_tensor([n])]) - 1
+        # K16: one decoupled-lookback reduce-by-key pass per value
+        # column (exact for ints — wraps like the atomics would)
+        uk = None
         outs = []
-        for v in vcols:  # exact for ints (wraps like the atomics)
-            cs = torch.cumsum(v, 0)
-            tot = cs[ends]
-            o = torch.empty_like(tot)
-            o[0] = tot[0]
-            torch.sub(tot[1:], tot[:-1], out=o[1:])
-            outs.append(o)
+        for v in vcols:
+            uq, sums, cnt = _C.segment_sum_sorted(ks, v.contiguous())
+            if uk is None:
+                m = int(cnt.item())
+                uk = uq[:m]
+            outs.append(sums[:m])
         if self.cap is not None:
             # merge with the provisionally hash-inserted first batch
             self._insert_now(uk, outs, "global")
