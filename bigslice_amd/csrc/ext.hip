@@ -286,6 +286,32 @@ void groupby_insert_lds(torch::Tensor keys, torch::Tensor vals,
   HIP_CHECK(hipGetLastError());
 }
 
+// Multi-row software-pipelined one-level insert (single int64 SUM
+// value): see groupby.hip.
+void groupby_insert_mlp(torch::Tensor keys, torch::Tensor vals,
+                        torch::Tensor tkeys, torch::Tensor tab,
+                        torch::Tensor flags, int64_t max_probes) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
+  TORCH_CHECK(vals.scalar_type() == torch::kInt64);
+  keys = keys.contiguous();
+  vals = vals.contiguous();
+  int64_t n = keys.size(0);
+  if (n == 0) return;
+  int64_t cap = tkeys.size(0) - 1;
+  TORCH_CHECK(cap > 0 && (cap & (cap - 1)) == 0);
+  const uint32_t seed = 0x9acb0442u;
+  int64_t nblocks = std::min<int64_t>(
+      (n + THREADS * 4 - 1) / (THREADS * 4), 32768);
+  hipLaunchKernelGGL(k_groupby_insert_sum_i64_mlp, dim3((uint32_t)nblocks),
+                     dim3(THREADS), 0, current_stream(),
+                     keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(), n,
+                     tkeys.data_ptr<int64_t>(),
+                     (long long*)tab.data_ptr<int64_t>(), cap, seed,
+                     flags.data_ptr<int32_t>(),
+                     flags.data_ptr<int32_t>() + 1, max_probes);
+  HIP_CHECK(hipGetLastError());
+}
+
 // Packed-slot fast path (single int64 SUM value): see groupby.hip.
 void groupby_insert_packed(torch::Tensor keys, torch::Tensor vals,
                            torch::Tensor table, torch::Tensor flags,
@@ -516,6 +542,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hash-aggregate insert pass (K9)");
   m.def("groupby_insert_lds", &groupby_insert_lds,
         "two-level LDS+global insert (int64 sum)");
+  m.def("groupby_insert_mlp", &groupby_insert_mlp,
+        "multi-row software-pipelined insert (int64 sum)");
   m.def("groupby_insert_packed", &groupby_insert_packed,
         "packed-slot insert (int64 sum fast path)");
   m.def("groupby_compact_packed", &groupby_compact_packed,

@@ -426,3 +426,81 @@ extern "C" __global__ void k_groupby_insert(
       accum(vals.tab[c], slot, vals.src[c], i, vals.agg[c]);
   }
 }
+
+// Resolve one row whose FIRST probe was already loaded (cur0): the MLP
+// insert batches first-probe loads of several rows before resolving,
+// so the random-access latency of the table read is paid R-ways in
+// parallel instead of serially per row.
+__device__ __forceinline__ void gb_resolve_add(
+    int64_t k, long long v, long long cur0, uint64_t h, int64_t* tkeys,
+    long long* tab, uint64_t mask, int64_t cap, uint32_t /*seed*/,
+    int32_t* sentinel_seen, int32_t* overflow, int64_t max_probes) {
+  int64_t slot;
+  if (k == GB_SENTINEL) {
+    atomicOr(sentinel_seen, 1);
+    slot = cap;
+  } else {
+    long long cur = cur0;
+    int64_t probes = 0;
+    for (;;) {
+      if (cur == k) break;
+      if (cur == GB_SENTINEL) {
+        long long prev = atomicCAS((unsigned long long*)&tkeys[h],
+                                   (unsigned long long)GB_SENTINEL,
+                                   (unsigned long long)k);
+        if (prev == GB_SENTINEL || prev == k) break;
+      }
+      h = (h + 1) & mask;
+      if (++probes >= max_probes) {
+        atomicOr(overflow, 1);
+        return;
+      }
+      cur = ((volatile long long*)tkeys)[h];
+    }
+    slot = (int64_t)h;
+  }
+  atomicAdd((unsigned long long*)&tab[slot], (unsigned long long)v);
+}
+
+#define GB_MLP_R 4
+
+// Multi-row software-pipelined one-level insert (single int64 SUM
+// shape): R grid-stride rows per iteration, all R first-probe loads in
+// flight together (NOTES.md headroom item 1).
+extern "C" __global__ void k_groupby_insert_sum_i64_mlp(
+    const int64_t* keys, const int64_t* vals, int64_t n, int64_t* tkeys,
+    long long* tab, int64_t cap, uint32_t seed, int32_t* sentinel_seen,
+    int32_t* overflow, int64_t max_probes) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t mask = (uint64_t)cap - 1;
+  for (int64_t base = tid; base < n; base += stride * GB_MLP_R) {
+    int64_t k[GB_MLP_R];
+    long long v[GB_MLP_R];
+    uint64_t h[GB_MLP_R];
+    long long cur[GB_MLP_R];
+    bool act[GB_MLP_R];
+#pragma unroll
+    for (int r = 0; r < GB_MLP_R; ++r) {
+      int64_t i = base + (int64_t)r * stride;  // coalesced per r
+      act[r] = i < n;
+      if (act[r]) {
+        k[r] = keys[i];
+        v[r] = vals[i];
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < GB_MLP_R; ++r)
+      if (act[r] && k[r] != GB_SENTINEL)
+        h[r] = mm3_u64((uint64_t)k[r], seed) & mask;
+#pragma unroll
+    for (int r = 0; r < GB_MLP_R; ++r)  // R loads in flight
+      if (act[r] && k[r] != GB_SENTINEL)
+        cur[r] = ((volatile long long*)tkeys)[h[r]];
+#pragma unroll
+    for (int r = 0; r < GB_MLP_R; ++r)
+      if (act[r])
+        gb_resolve_add(k[r], v[r], cur[r], h[r], tkeys, tab, mask, cap,
+                       seed, sentinel_seen, overflow, max_probes);
+  }
+}

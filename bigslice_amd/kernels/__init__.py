@@ -217,6 +217,17 @@ class GroupTable:
             return self._sample_forced
         distinct = 1 + int(self._sample_ne.item())
         s = self._sample_n
+        # Invert d = K(1-e^(-s/K)) to estimate the key-space size and
+        # presize the table: the 10M-key hash insert measured 6.2 ms
+        # properly sized vs 22 ms through the overflow-regrow grind.
+        if distinct < s:
+            import math
+            K = float(distinct)
+            for _ in range(20):
+                K = distinct / (1.0 - math.exp(-s / K))
+            if self.cap_hint is None:
+                self.cap_hint = _next_pow2(
+                    min(max(int(4 * K), 1024), 1 << 30))
         if distinct * 32 <= s:
             return "lds"
         if distinct * 100 <= s * 88:
@@ -277,6 +288,10 @@ class GroupTable:
                                   1, blocks, dummy)
         elif self._lds and mode is None:
             self._insert_adaptive(keys, vals)
+        elif (self._sum_i64 and mode == "global"
+              and os.environ.get("BIGSLICE_GB_GLOBAL_MLP", "0") == "1"):
+            _C.groupby_insert_mlp(keys, vals[0], self.tkeys,
+                                  self.tabs[0], self.flags, MAX_PROBES)
         else:
             _C.groupby_insert(keys, list(vals), self.codes, self.tkeys,
                               self.tabs, self.flags, MAX_PROBES)
