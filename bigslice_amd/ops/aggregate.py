@@ -282,6 +282,25 @@ class DictAggregator:
         vals = lists[nkey:]
         state = self.state
         aggs = self.agg.aggs
+        if len(vals) == 1 and aggs == ["sum"]:
+            vcol = vals[0]
+            if isinstance(vcol, list) and vcol.count(1) == len(vcol):
+                # counting shape (all-ones values): C-speed Counter
+                from collections import Counter
+                for k, n in Counter(keys).items():
+                    cur = state.get(k)
+                    if cur is None:
+                        state[k] = [n]
+                    else:
+                        cur[0] += n
+            else:
+                for k, v in zip(keys, vcol):
+                    cur = state.get(k)
+                    if cur is None:
+                        state[k] = [v]
+                    else:
+                        cur[0] += v
+            return
         for i, k in enumerate(keys):
             row = tuple(v[i] for v in vals)
             cur = state.get(k)
