@@ -336,3 +336,27 @@ def test_grouptable_hotkeys_streams_lds(kernels):
     order = torch.argsort(uk)
     assert torch.equal(uk[order], ref_uk)
     assert torch.equal(us[order], ref)
+
+
+def test_grouptable_sort_combine_multicol(kernels):
+    import os
+    n = 3_000_000
+    k = torch.randint(0, 2_500_000, (n,), dtype=torch.int64,
+                      device="cuda:0")
+    v1 = torch.randint(-5, 5, (n,), dtype=torch.int64, device="cuda:0")
+    v2 = torch.ones(n, dtype=torch.int64, device="cuda:0")
+    os.environ["BIGSLICE_GB_COMBINE"] = "sort"
+    try:
+        t = kernels.GroupTable([torch.int64, torch.int64],
+                               ["sum", "sum"], torch.device("cuda:0"))
+        t.insert(k, [v1, v2])
+        uk, (s1, s2) = t.finish()
+    finally:
+        del os.environ["BIGSLICE_GB_COMBINE"]
+    ref_uk, inv = torch.unique(k, return_inverse=True)
+    r1 = torch.zeros_like(ref_uk).index_add_(0, inv, v1)
+    r2 = torch.zeros_like(ref_uk).index_add_(0, inv, v2)
+    order = torch.argsort(uk)
+    assert torch.equal(uk[order], ref_uk)
+    assert torch.equal(s1[order], r1)
+    assert torch.equal(s2[order], r2)
