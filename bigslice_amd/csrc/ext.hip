@@ -472,33 +472,50 @@ std::vector<torch::Tensor> radix_sort_kv(torch::Tensor keys,
   return {keys_out, vals_out};
 }
 
-// K16: (unique_keys, sums, count) over key-sorted int64 pairs.  count
-// comes back as a 1-element device tensor; the caller slices after one
-// sync.
-std::vector<torch::Tensor> segment_sum_sorted_i64(torch::Tensor keys,
-                                                  torch::Tensor vals) {
+// K16: (unique_keys, aggregates, count) over key-sorted pairs via
+// rocPRIM reduce-by-key: deterministic segmented reduction (fixed tree
+// order — float sums are run-to-run reproducible, unlike atomics).
+// count comes back as a 1-element device tensor; the caller slices
+// after one sync.  code: 0=sum 1=min 2=max.
+std::vector<torch::Tensor> segment_reduce_sorted(torch::Tensor keys,
+                                                 torch::Tensor vals,
+                                                 int64_t code) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   TORCH_CHECK(vals.is_cuda() && vals.is_contiguous());
-  TORCH_CHECK(keys.scalar_type() == torch::kInt64 &&
-              vals.scalar_type() == torch::kInt64,
-              "segment_sum_sorted_i64: int64 only");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64,
+              "segment_reduce_sorted: int64 keys only");
   int64_t n = keys.size(0);
   auto uniq = torch::empty_like(keys);
-  auto sums = torch::empty_like(vals);
+  auto aggs = torch::empty_like(vals);
   auto count = torch::zeros({1}, keys.options());
-  if (n == 0) return {uniq, sums, count};
-  size_t temp_bytes = 0;
-  reduce_by_key_sum_i64(keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(),
-                        n, uniq.data_ptr<int64_t>(),
-                        sums.data_ptr<int64_t>(), count.data_ptr<int64_t>(),
-                        nullptr, temp_bytes, current_stream());
-  auto temp = torch::empty({(int64_t)temp_bytes},
-                           keys.options().dtype(torch::kUInt8));
-  reduce_by_key_sum_i64(keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(),
-                        n, uniq.data_ptr<int64_t>(),
-                        sums.data_ptr<int64_t>(), count.data_ptr<int64_t>(),
-                        temp.data_ptr(), temp_bytes, current_stream());
-  return {uniq, sums, count};
+  if (n == 0) return {uniq, aggs, count};
+  auto run = [&](auto fn) {
+    size_t temp_bytes = 0;
+    fn(keys.data_ptr<int64_t>(), vals.data_ptr(), n,
+       uniq.data_ptr<int64_t>(), aggs.data_ptr(),
+       count.data_ptr<int64_t>(), (int)code, nullptr, temp_bytes,
+       current_stream());
+    auto temp = torch::empty({(int64_t)temp_bytes},
+                             keys.options().dtype(torch::kUInt8));
+    fn(keys.data_ptr<int64_t>(), vals.data_ptr(), n,
+       uniq.data_ptr<int64_t>(), aggs.data_ptr(),
+       count.data_ptr<int64_t>(), (int)code, temp.data_ptr(), temp_bytes,
+       current_stream());
+  };
+  switch (vals.scalar_type()) {
+    case torch::kInt64: run(reduce_by_key_int64_t); break;
+    case torch::kInt32: run(reduce_by_key_int32_t); break;
+    case torch::kFloat32: run(reduce_by_key_float); break;
+    case torch::kFloat64: run(reduce_by_key_double); break;
+    default:
+      TORCH_CHECK(false, "segment_reduce_sorted: unsupported val dtype");
+  }
+  return {uniq, aggs, count};
+}
+
+std::vector<torch::Tensor> segment_sum_sorted_i64(torch::Tensor keys,
+                                                  torch::Tensor vals) {
+  return segment_reduce_sorted(keys, vals, 0);
 }
 
 torch::Tensor radix_argsort(torch::Tensor keys) {
@@ -557,6 +574,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("radix_sort_keys", &radix_sort_keys, "device radix key sort");
   m.def("segment_sum_sorted", &segment_sum_sorted_i64,
         "reduce-by-key sum over sorted int64 pairs (K16)");
+  m.def("segment_reduce_sorted", &segment_reduce_sorted,
+        "typed reduce-by-key over sorted pairs (K16; deterministic)");
   m.def("radix_sort_kv", &radix_sort_kv,
         "direct (key, 8-byte value) radix sort");
   m.def("slot_pids", &slot_pids, "table-slot-range partition ids");

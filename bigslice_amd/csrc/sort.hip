@@ -67,15 +67,47 @@ INSTANTIATE_SORT_KEYS(double)
 // Replaces the mask/nonzero/cumsum torch chain in the sort-combine.
 #include <rocprim/device/device_reduce_by_key.hpp>
 
-void reduce_by_key_sum_i64(const int64_t* keys, const int64_t* vals,
-                           int64_t n, int64_t* uniq_out, int64_t* sums_out,
-                           int64_t* count_out, void* temp,
-                           size_t& temp_bytes, hipStream_t s) {
+template <typename V, typename Op>
+static void reduce_by_key_impl(const int64_t* keys, const V* vals,
+                               int64_t n, int64_t* uniq_out, V* aggs_out,
+                               int64_t* count_out, Op op, void* temp,
+                               size_t& temp_bytes, hipStream_t s) {
   hipError_t err = rocprim::reduce_by_key(
-      temp, temp_bytes, keys, vals, (size_t)n, uniq_out, sums_out,
-      count_out, rocprim::plus<int64_t>(), rocprim::equal_to<int64_t>(),
-      s);
+      temp, temp_bytes, keys, vals, (size_t)n, uniq_out, aggs_out,
+      count_out, op, rocprim::equal_to<int64_t>(), s);
   if (err != hipSuccess)
     throw std::runtime_error(std::string("rocprim::reduce_by_key: ") +
                              hipGetErrorString(err));
 }
+
+// agg codes match kernels._AGG_CODES: 0=sum 1=min 2=max 3=prod
+#define INSTANTIATE_RBK(V)                                                \
+  void reduce_by_key_##V(const int64_t* keys, const void* vals,           \
+                         int64_t n, int64_t* uniq_out, void* aggs_out,    \
+                         int64_t* count_out, int code, void* temp,        \
+                         size_t& temp_bytes, hipStream_t s) {             \
+    switch (code) {                                                       \
+      case 0:                                                             \
+        reduce_by_key_impl(keys, (const V*)vals, n, uniq_out,             \
+                           (V*)aggs_out, count_out, rocprim::plus<V>(),   \
+                           temp, temp_bytes, s);                          \
+        break;                                                            \
+      case 1:                                                             \
+        reduce_by_key_impl(keys, (const V*)vals, n, uniq_out,             \
+                           (V*)aggs_out, count_out,                       \
+                           rocprim::minimum<V>(), temp, temp_bytes, s);   \
+        break;                                                            \
+      case 2:                                                             \
+        reduce_by_key_impl(keys, (const V*)vals, n, uniq_out,             \
+                           (V*)aggs_out, count_out,                       \
+                           rocprim::maximum<V>(), temp, temp_bytes, s);   \
+        break;                                                            \
+      default:                                                            \
+        throw std::runtime_error("reduce_by_key: bad agg code");          \
+    }                                                                     \
+  }
+
+INSTANTIATE_RBK(int64_t)
+INSTANTIATE_RBK(int32_t)
+INSTANTIATE_RBK(float)
+INSTANTIATE_RBK(double)

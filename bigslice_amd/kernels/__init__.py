@@ -175,14 +175,18 @@ class GroupTable:
 
     @property
     def _sort_combinable(self) -> bool:
-        """Shapes the sort+segment-reduce combine handles exactly:
-        all-int64 SUM columns (cumsum-difference is exact for ints).
-        Hash vs sort is decided at finish() from sampled cardinality;
+        """Shapes the sort+reduce-by-key combine handles: sum/min/max
+        over numeric columns.  The segmented reduction is DETERMINISTIC
+        (fixed tree order), so float sums through this path are
+        run-to-run reproducible — unlike the hash table's atomic adds.
+        Hash vs sort is decided from the sampled cardinality;
         BIGSLICE_GB_COMBINE forces {hash,sort}."""
         return (_C is not None
                 and str(self.device).startswith("cuda")
-                and all(a == "sum" for a in self.aggs)
-                and all(dt == torch.int64 for dt in self.val_dtypes)
+                and all(a in ("sum", "min", "max") for a in self.aggs)
+                and all(dt in (torch.int64, torch.int32, torch.float32,
+                               torch.float64)
+                        for dt in self.val_dtypes)
                 and os.environ.get("BIGSLICE_GB_COMBINE", "auto")
                 != "hash")
 
@@ -368,7 +372,7 @@ class GroupTable:
                     self._insert_now(k, v, mode)
                 return None
         ncols = len(self.val_dtypes)
-        if ncols == 1:
+        if ncols == 1 and self.val_dtypes[0].itemsize == 8:
             vals = (batches[0][1][0] if len(batches) == 1
                     else torch.cat([v[0] for _, v in batches]))
             ks, vs = _C.radix_sort_kv(keys.contiguous(),
@@ -383,15 +387,16 @@ class GroupTable:
                        else torch.cat([v[c] for _, v in batches]))
                 vcols.append(col[perm])
         # K16: one decoupled-lookback reduce-by-key pass per value
-        # column (exact for ints — wraps like the atomics would)
+        # column — deterministic, exact for ints (wraps like atomics)
         uk = None
         outs = []
-        for v in vcols:
-            uq, sums, cnt = _C.segment_sum_sorted(ks, v.contiguous())
+        for c, v in enumerate(vcols):
+            uq, aggs, cnt = _C.segment_reduce_sorted(
+                ks, v.contiguous(), self.codes[c])
             if uk is None:
                 m = int(cnt.item())
                 uk = uq[:m]
-            outs.append(sums[:m])
+            outs.append(aggs[:m])
         if self.cap is not None:
             # merge with the provisionally hash-inserted first batch
             self._insert_now(uk, outs, "global")

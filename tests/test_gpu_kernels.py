@@ -360,3 +360,50 @@ def test_grouptable_sort_combine_multicol(kernels):
     assert torch.equal(uk[order], ref_uk)
     assert torch.equal(s1[order], r1)
     assert torch.equal(s2[order], r2)
+
+
+def test_grouptable_sort_combine_float_deterministic(kernels):
+    # float sums via the sort path: deterministic across runs and
+    # accurate against an fp64 reference; min/max exact.
+    import os
+    n = 2_500_000
+    k = torch.randint(0, 2_000_000, (n,), dtype=torch.int64,
+                      device="cuda:0")
+    v = torch.randn(n, dtype=torch.float32, device="cuda:0")
+    os.environ["BIGSLICE_GB_COMBINE"] = "sort"
+    try:
+        outs = []
+        for _ in range(2):
+            t = kernels.GroupTable([torch.float32], ["sum"],
+                                   torch.device("cuda:0"))
+            t.insert(k, [v])
+            uk, (s,) = t.finish()
+            o = torch.argsort(uk)
+            outs.append((uk[o], s[o]))
+        assert torch.equal(outs[0][1], outs[1][1])  # bitwise repeatable
+        ref_uk, inv = torch.unique(k, return_inverse=True)
+        ref = torch.zeros(ref_uk.shape[0], dtype=torch.float64,
+                          device="cuda:0").index_add_(0, inv,
+                                                      v.to(torch.float64))
+        assert torch.equal(outs[0][0], ref_uk)
+        assert torch.allclose(outs[0][1].to(torch.float64), ref,
+                              atol=1e-3, rtol=1e-5)
+        t = kernels.GroupTable([torch.float32, torch.int64],
+                               ["min", "max"], torch.device("cuda:0"))
+        vi = torch.randint(-10**9, 10**9, (n,), dtype=torch.int64,
+                           device="cuda:0")
+        t.insert(k, [v, vi])
+        uk2, (mn, mx) = t.finish()
+        o2 = torch.argsort(uk2)
+        ref_mn = torch.full((ref_uk.shape[0],), float("inf"),
+                            device="cuda:0").scatter_reduce_(
+            0, inv, v, "amin").to(torch.float32)
+        ref_mx = torch.full((ref_uk.shape[0],), -(2**62),
+                            dtype=torch.int64,
+                            device="cuda:0").scatter_reduce_(
+            0, inv, vi, "amax")
+        assert torch.equal(uk2[o2], ref_uk)
+        assert torch.equal(mn[o2], ref_mn)
+        assert torch.equal(mx[o2], ref_mx)
+    finally:
+        del os.environ["BIGSLICE_GB_COMBINE"]
