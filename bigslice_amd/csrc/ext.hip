@@ -17,6 +17,7 @@
 #include "hash_partition.hip"
 #include "groupby.hip"
 #include "sort.hip"
+#include "vecagg.hip"
 
 namespace {
 
@@ -518,6 +519,26 @@ std::vector<torch::Tensor> segment_sum_sorted_i64(torch::Tensor keys,
   return segment_reduce_sorted(keys, vals, 0);
 }
 
+// Vector-aggregate experiment (see vecagg.hip): colsum of [N,16] f32.
+torch::Tensor colsum16(torch::Tensor x, bool mfma) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kFloat32 && x.dim() == 2 &&
+              x.size(1) == 16);
+  int64_t n = x.size(0);
+  auto out = torch::zeros({16}, x.options());
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 8192);
+  if (mfma)
+    hipLaunchKernelGGL(k_colsum_mfma, dim3(blocks), dim3(THREADS), 0,
+                       current_stream(), x.data_ptr<float>(), n,
+                       out.data_ptr<float>());
+  else
+    hipLaunchKernelGGL(k_colsum_valu, dim3(blocks), dim3(THREADS), 0,
+                       current_stream(), x.data_ptr<float>(), n,
+                       out.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 torch::Tensor radix_argsort(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
@@ -576,6 +597,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "reduce-by-key sum over sorted int64 pairs (K16)");
   m.def("segment_reduce_sorted", &segment_reduce_sorted,
         "typed reduce-by-key over sorted pairs (K16; deterministic)");
+  m.def("colsum16", &colsum16,
+        "vector-aggregate experiment: [N,16] f32 colsum, VALU vs MFMA");
   m.def("radix_sort_kv", &radix_sort_kv,
         "direct (key, 8-byte value) radix sort");
   m.def("slot_pids", &slot_pids, "table-slot-range partition ids");

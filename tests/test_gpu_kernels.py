@@ -407,3 +407,12 @@ def test_grouptable_sort_combine_float_deterministic(kernels):
         assert torch.equal(mx[o2], ref_mx)
     finally:
         del os.environ["BIGSLICE_GB_COMBINE"]
+
+
+def test_colsum16_valu_and_mfma(kernels):
+    from bigslice_amd.kernels import _C
+    x = torch.randn(1_000_003, 16, dtype=torch.float32, device="cuda:0")
+    ref = x.to(torch.float64).sum(0)
+    for mfma in (False, True):
+        out = _C.colsum16(x, mfma).to(torch.float64)
+        assert torch.allclose(out, ref, atol=1e-1, rtol=1e-5), mfma
