@@ -18,6 +18,7 @@
 #include "groupby.hip"
 #include "sort.hip"
 #include "vecagg.hip"
+#include "strings.hip"
 
 namespace {
 
@@ -519,6 +520,43 @@ std::vector<torch::Tensor> segment_sum_sorted_i64(torch::Tensor keys,
   return segment_reduce_sorted(keys, vals, 0);
 }
 
+// K17: murmur3 over variable-length byte rows (device strings).
+torch::Tensor hash_bytes(torch::Tensor bytes, torch::Tensor offsets,
+                         int64_t seed) {
+  TORCH_CHECK(bytes.is_cuda() && bytes.is_contiguous() &&
+              bytes.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(offsets.is_cuda() && offsets.is_contiguous() &&
+              offsets.scalar_type() == torch::kInt64);
+  int64_t n = offsets.size(0) - 1;
+  auto out = torch::empty({n}, bytes.options().dtype(torch::kUInt32));
+  if (n == 0) return out;
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
+  hipLaunchKernelGGL(k_hash_bytes, dim3(blocks), dim3(THREADS), 0,
+                     current_stream(), bytes.data_ptr<uint8_t>(),
+                     offsets.data_ptr<int64_t>(), n, (uint32_t)seed,
+                     out.data_ptr<uint32_t>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
+torch::Tensor hash_bytes64(torch::Tensor bytes, torch::Tensor offsets,
+                           int64_t seed_hi, int64_t seed_lo) {
+  TORCH_CHECK(bytes.is_cuda() && bytes.is_contiguous() &&
+              bytes.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(offsets.is_cuda() && offsets.is_contiguous() &&
+              offsets.scalar_type() == torch::kInt64);
+  int64_t n = offsets.size(0) - 1;
+  auto out = torch::empty({n}, bytes.options().dtype(torch::kInt64));
+  if (n == 0) return out;
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
+  hipLaunchKernelGGL(k_hash_bytes64, dim3(blocks), dim3(THREADS), 0,
+                     current_stream(), bytes.data_ptr<uint8_t>(),
+                     offsets.data_ptr<int64_t>(), n, (uint32_t)seed_hi,
+                     (uint32_t)seed_lo, out.data_ptr<int64_t>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 // Vector-aggregate experiment (see vecagg.hip): colsum of [N,16] f32.
 torch::Tensor colsum16(torch::Tensor x, bool mfma) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
@@ -597,6 +635,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "reduce-by-key sum over sorted int64 pairs (K16)");
   m.def("segment_reduce_sorted", &segment_reduce_sorted,
         "typed reduce-by-key over sorted pairs (K16; deterministic)");
+  m.def("hash_bytes", &hash_bytes,
+        "murmur3-32 over varlen byte rows (K17, device strings)");
+  m.def("hash_bytes64", &hash_bytes64,
+        "64-bit two-seed murmur3 dictionary ids (K17)");
   m.def("colsum16", &colsum16,
         "vector-aggregate experiment: [N,16] f32 colsum, VALU vs MFMA");
   m.def("radix_sort_kv", &radix_sort_kv,

@@ -83,3 +83,82 @@ def fast_wordcount(nshard: int, open_fn: Callable[[], Iterable[str]]
         return (words[0], sum(cs))
 
     return Map(joined, resolve, out_schema=(str, int), rowwise=True)
+
+
+# -- gpu_wordcount: device dictionary ids via K17 ------------------------
+
+_GPU_WC_RUNS = {}
+_GPU_WC_SEQ = [0]
+
+
+def _build_gpu_wordcount(nshard, token):
+    import torch
+
+    from . import strings
+    from .ops import ReaderFunc
+    from .ops.elementwise import schema_of
+
+    lines_list, device, state = _GPU_WC_RUNS[token]
+
+    def gen(shard, ctx):
+        words_all = []
+        state[shard] = words_all
+        base = shard << 40
+
+        def emit(buf):
+            ids = strings.string_ids(buf, device)
+            start = base + len(words_all)
+            words_all.extend(buf)
+            n = len(buf)
+            ones = torch.ones(n, dtype=torch.int64, device=ids.device)
+            idx = torch.arange(start, start + n, dtype=torch.int64,
+                               device=ids.device)
+            return (ids, ones, idx)
+
+        buf = []
+        for li in range(shard, len(lines_list), nshard):
+            buf.extend(lines_list[li].split())
+            if len(buf) >= 500_000:
+                yield emit(buf)
+                buf = []
+        if buf:
+            yield emit(buf)
+
+    src = ReaderFunc(nshard, gen, schema_of(int, int, int, prefix=1))
+    from .ops import Reduce
+    return Reduce(src, ("sum", "min"))
+
+
+def _register_gpu_wc():
+    from .runtime.session import func
+    return func(_build_gpu_wordcount)
+
+
+_gpu_wc_func = _register_gpu_wc()
+
+
+def gpu_wordcount(sess, nshard: int, lines, device) -> dict:
+    """word -> count with hashing AND counting on device: the host
+    only tokenizes and packs bytes; K17 (csrc/strings.hip) computes
+    64-bit dictionary ids on the GPU and the count runs as a
+    device-native int64 Reduce(("sum","min")) — "min" keeps a
+    representative row index per id so readback recovers the word
+    without any host-side id map on the hot path.
+
+    Single-process sessions (the retained per-shard token lists are
+    shared-memory state); see fast_wordcount's side-channel pattern
+    for the distributed variant.  Collision note as fast_wordcount."""
+    _GPU_WC_SEQ[0] += 1
+    token = _GPU_WC_SEQ[0]
+    state = {}
+    _GPU_WC_RUNS[token] = (list(lines), device, state)
+    try:
+        res = sess.run(_gpu_wc_func, nshard, token)
+        out = {}
+        mask = (1 << 40) - 1
+        for _id, cnt, gidx in res.scan():
+            out[state[gidx >> 40][gidx & mask]] = cnt
+        res.discard()
+    finally:
+        _GPU_WC_RUNS.pop(token, None)
+    return out

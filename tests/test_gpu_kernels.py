@@ -416,3 +416,32 @@ def test_colsum16_valu_and_mfma(kernels):
     for mfma in (False, True):
         out = _C.colsum16(x, mfma).to(torch.float64)
         assert torch.allclose(out, ref, atol=1e-1, rtol=1e-5), mfma
+
+
+def test_hash_bytes_matches_host(kernels):
+    import random
+    from bigslice_amd import strings
+    from bigslice_amd.hashing import murmur3_bytes
+    rng = random.Random(5)
+    words = ["", "a", "ab", "abc", "abcd", "ünïcödé-wörd", "x" * 33]
+    words += ["".join(chr(rng.randrange(33, 1000)) for _ in
+                      range(rng.randrange(0, 20))) for _ in range(200)]
+    for seed in (0, 7, 0x9ACB0442):
+        dev = strings.hash_strings(words, "cuda:0", seed).cpu()
+        ref = torch.tensor([murmur3_bytes(w.encode("utf-8"), seed)
+                            for w in words], dtype=torch.int64)
+        assert torch.equal(dev, ref), seed
+    ids_dev = strings.string_ids(words, "cuda:0").cpu()
+    ids_host = strings.string_ids(words, "cpu")
+    assert torch.equal(ids_dev, ids_host)
+
+
+def test_gpu_wordcount_recipe_device(kernels):
+    from collections import Counter
+    import bigslice_amd as bs
+    from bigslice_amd import recipes
+    lines = [f"alpha beta g{i % 50} delta e{i % 11}" for i in range(4000)]
+    ref = Counter(w for ln in lines for w in ln.split())
+    sess = bs.start(parallelism=4, device="cuda:0")
+    got = recipes.gpu_wordcount(sess, 4, lines, "cuda:0")
+    assert got == dict(ref)

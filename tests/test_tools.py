@@ -161,3 +161,13 @@ def test_kmeans_example_converges():
     cx, cy = km.run_kmeans(sess, 4, 4, 8, "cpu")
     assert len(cx) == 4
     assert all(0.0 <= c <= 1.0 for c in cx + cy)
+
+
+def test_gpu_wordcount_recipe_cpu():
+    from collections import Counter
+    from bigslice_amd import recipes
+    lines = [f"a b c d{i % 7} e{i % 3}" for i in range(500)]
+    ref = Counter(w for ln in lines for w in ln.split())
+    sess = bs.start(parallelism=4, device="cpu")
+    got = recipes.gpu_wordcount(sess, 4, lines, "cpu")
+    assert got == dict(ref)
