@@ -24,7 +24,8 @@ def pack_strings(strs: Sequence[str]) -> Tuple[np.ndarray, np.ndarray]:
     offsets = np.zeros(len(enc) + 1, dtype=np.int64)
     np.cumsum(np.fromiter((len(b) for b in enc), np.int64, len(enc)),
               out=offsets[1:])
-    data = np.frombuffer(b"".join(enc), dtype=np.uint8)
+    # copy: frombuffer views are read-only, which torch rejects
+    data = np.frombuffer(b"".join(enc), dtype=np.uint8).copy()
     return data, offsets
 
 
