@@ -1,6 +1,5 @@
 """`python -m bigslice_amd`: environment and build info."""
 
-import sys
 
 import torch
 

@@ -15,9 +15,8 @@ hipMemcpyAsync under torch.
 
 from __future__ import annotations
 
-from typing import Iterable, List, Optional, Sequence, Union
+from typing import List, Sequence, Union
 
-import numpy as np
 import torch
 
 from . import hashing

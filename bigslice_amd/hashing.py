@@ -18,7 +18,6 @@ from typing import Sequence
 import numpy as np
 import torch
 
-from .schema import OBJECT
 
 _C1 = np.uint32(0xCC9E2D51)
 _C2 = np.uint32(0x1B873593)

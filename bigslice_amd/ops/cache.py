@@ -12,7 +12,7 @@ so every worker process compiles the identical graph.
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+from typing import List
 
 from ..sliceio import IterReader, Reader, codec
 from .slice_base import Dep, Name, Slice, TaskContext

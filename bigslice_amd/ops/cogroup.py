@@ -13,11 +13,11 @@ Two paths:
 
 from __future__ import annotations
 
-from typing import List, Sequence
+from typing import List
 
 from ..frame import Frame
 from ..schema import OBJECT, Schema, common_key_schema
-from ..sliceio import IterReader, Reader, read_all
+from ..sliceio import IterReader, Reader
 from .slice_base import Dep, Name, Slice, TaskContext
 
 

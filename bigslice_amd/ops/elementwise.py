@@ -15,12 +15,12 @@ device-native.  ``rowwise=True`` selects a host fallback for object columns
 
 from __future__ import annotations
 
-from typing import Callable, List, Optional, Sequence
+from typing import Callable, List, Optional
 
 import torch
 
 from ..frame import Frame
-from ..schema import OBJECT, Schema, infer_dtype, is_object
+from ..schema import OBJECT, Schema, is_object
 from ..sliceio import IterReader, Reader
 from .slice_base import Dep, Name, Slice, TaskContext
 

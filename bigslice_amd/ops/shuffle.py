@@ -12,7 +12,7 @@ from __future__ import annotations
 
 from typing import Callable, Optional
 
-from ..sliceio import MultiReader, Reader
+from ..sliceio import Reader
 from .slice_base import Dep, Name, Slice, TaskContext
 
 

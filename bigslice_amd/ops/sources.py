@@ -7,7 +7,7 @@ per-shard state), scan.go:22-69 (ScanReader: line-sharded text source).
 
 from __future__ import annotations
 
-from typing import Callable, Iterable, List, Optional, Sequence
+from typing import Callable, Iterable, List
 
 import torch
 

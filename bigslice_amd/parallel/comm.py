@@ -19,8 +19,7 @@ over all_gather_object.
 from __future__ import annotations
 
 import os
-import pickle
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

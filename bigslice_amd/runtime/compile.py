@@ -208,7 +208,6 @@ class Compiler:
         tasks: List[Task] = []
         for shard in range(nshard):
             def do(dep_readers, ctx, _shard=shard):
-                from ..sliceio import MultiReader
                 return dep_readers[0]
             t = Task(
                 name=f"{base}:{shard}",

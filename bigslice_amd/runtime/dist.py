@@ -18,7 +18,6 @@ from __future__ import annotations
 
 from typing import Dict, List, Optional, Sequence, Tuple
 
-import torch
 
 from ..ops.slice_base import TaskContext
 from ..parallel.comm import Comm

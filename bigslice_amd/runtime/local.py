@@ -10,7 +10,6 @@ from __future__ import annotations
 
 import threading
 import traceback
-from typing import List, Optional
 
 import torch
 

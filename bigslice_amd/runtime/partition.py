@@ -11,7 +11,7 @@ this dispatches the HIP partition kernel when available; the torch fallback
 
 from __future__ import annotations
 
-from typing import Callable, Dict, List, Optional
+from typing import Callable, List, Optional
 
 import torch
 

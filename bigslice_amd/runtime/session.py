@@ -11,10 +11,9 @@ from __future__ import annotations
 
 import hashlib
 import threading
-from typing import Callable, List, Optional, Sequence
+from typing import Callable, List, Sequence
 
 from ..ops.slice_base import Name, Slice, TaskContext
-from ..schema import Schema
 from ..sliceio import MultiReader, Reader, Scanner
 from .compile import CompileEnv, Compiler
 from .eval import Executor, evaluate
@@ -147,7 +146,7 @@ class Result(Slice):
                 evaluate(session.executor, [t])
                 return session.executor.reader(t, 0)
 
-        from ..sliceio import FuncReader, Reader
+        from ..sliceio import Reader
 
         class _LazyTaskReader(Reader):
             def __init__(self, t):

@@ -12,7 +12,7 @@ from __future__ import annotations
 import os
 import struct
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from ..frame import Frame
 from ..sliceio import Reader, codec

@@ -11,7 +11,6 @@ from __future__ import annotations
 
 import argparse
 import os
-import sys
 from typing import Dict, Optional, Tuple
 
 from .runtime.session import Session, start

@@ -5,7 +5,7 @@ from __future__ import annotations
 
 from typing import List
 
-from .runtime.session import FuncValue, Session, func, start
+from .runtime.session import FuncValue, func, start
 
 
 def run(builder_or_func, *args, device: str = "cpu",

@@ -10,7 +10,7 @@ recipes.gpu_wordcount)."""
 
 from __future__ import annotations
 
-from typing import List, Sequence, Tuple
+from typing import Sequence, Tuple
 
 import numpy as np
 import torch

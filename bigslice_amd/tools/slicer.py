@@ -11,7 +11,6 @@ Each subcommand asserts its invariant and exits non-zero on violation.
 from __future__ import annotations
 
 import argparse
-import sys
 import time
 
 import torch

@@ -7,7 +7,6 @@ wordcount): ScanReader -> Flatmap(split) -> Map((w,1)) -> Reduce(+).
 from __future__ import annotations
 
 import argparse
-import sys
 
 import bigslice_amd as bs
 

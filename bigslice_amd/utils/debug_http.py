@@ -8,7 +8,7 @@ from __future__ import annotations
 import json
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
-from typing import List, Optional
+from typing import List
 
 
 def serve_session(session, port: int) -> "ThreadingHTTPServer":

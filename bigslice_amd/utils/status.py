@@ -6,11 +6,10 @@ from __future__ import annotations
 
 import sys
 import threading
-import time
 from collections import Counter
-from typing import Dict, List, Sequence
+from typing import Dict, Sequence
 
-from ..runtime.task import Task, TaskState
+from ..runtime.task import Task
 
 
 def rollup(roots: Sequence[Task]) -> Dict[str, Dict[str, int]]:

@@ -12,7 +12,7 @@ from __future__ import annotations
 import json
 import threading
 import time
-from typing import List, Optional
+from typing import List
 
 
 class Tracer:
