@@ -30,3 +30,4 @@ from .utils import metrics  # noqa
 from .runtime import (Result, Session, func, registry_digest, start)  # noqa
 from . import sliceio  # noqa
 from . import config  # noqa
+from . import strings  # noqa  (K17 device string hashing / dict ids)
