@@ -21,7 +21,8 @@ from .frame import Frame  # noqa
 from .ops import (Aggregation, Cache, Cogroup, Const, Dep, Filter,  # noqa
                   Flatmap, Fold, Head, Map, Pragma, Prefixed, ReadCache,
                   ReaderFunc, Reduce, Repartition, Reshard, Reshuffle, Scan,
-                  ScanReader, Slice, WriterFunc, exclusive, materialize,
+                  ScanReader, Slice, TarReader, WriterFunc, exclusive,
+                  materialize,
                   procs, schema_of, unwrap)
 from . import slicetest  # noqa
 from . import sliceconfig  # noqa

@@ -10,3 +10,4 @@ from .reduce import Fold, Reduce  # noqa
 from .cogroup import Cogroup  # noqa
 from .cache import Cache, ReadCache  # noqa
 from .aggregate import Aggregation  # noqa
+from .archive import TarReader  # noqa
