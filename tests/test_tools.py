@@ -2,6 +2,7 @@
 
 import io
 import json
+import os
 import subprocess
 import sys
 import tarfile
@@ -171,3 +172,28 @@ def test_gpu_wordcount_recipe_cpu():
     sess = bs.start(parallelism=4, device="cpu")
     got = recipes.gpu_wordcount(sess, 4, lines, "cpu")
     assert got == dict(ref)
+
+
+def test_bench_contract_cpu():
+    # The round-end driver depends on bench.py's JSON line; guard it.
+    import json
+    import subprocess
+    import sys as _sys
+    out = subprocess.run(
+        [_sys.executable, "bench.py", "--rows-per-gpu", "80000",
+         "--nkeys", "1000", "--steps", "2", "--warmup", "1",
+         "--device", "cpu"],
+        capture_output=True, text=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(
+            __file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling",
+              "vs_baseline", "dtype", "data", "config"):
+        assert k in d, k
+    assert d["n_gpus"] == 1 and d["scaling"] == "weak"
+    assert d["higher_is_better"] is True and d["data"] == "synthetic"
+    assert d["config"]["rows_per_gpu"] == 80000
+    assert d["value"] > 0
