@@ -445,3 +445,33 @@ def test_gpu_wordcount_recipe_device(kernels):
     sess = bs.start(parallelism=4, device="cuda:0")
     got = recipes.gpu_wordcount(sess, 4, lines, "cuda:0")
     assert got == dict(ref)
+
+
+def test_chaos_task_loss_on_device(kernels):
+    # fault tolerance with device-resident frames: random transient
+    # task losses + stored-output loss; results stay exact.
+    import random
+    import bigslice_amd as bs
+    from bigslice_amd.runtime.local import LocalExecutor, TaskLost
+    from bigslice_amd.runtime.session import Session
+    rng = random.Random(11)
+
+    def build():
+        keys = torch.randint(0, 997, (200_000,), dtype=torch.int64,
+                             device="cuda:0")
+        vals = torch.ones(200_000, dtype=torch.int64, device="cuda:0")
+        return bs.Reduce(bs.Const(6, keys, vals), "sum")
+
+    ex = LocalExecutor(parallelism=4, device="cuda:0")
+
+    def chaos(task):
+        if task.consecutive_lost < 2 and rng.random() < 0.35:
+            ex.store.discard_task(task.name)
+            raise TaskLost(task.name)
+    ex.fault_hook = chaos
+    sess = Session(ex)
+    fv = bs.func(build)
+    res = sess.run(fv)
+    got = dict(res.scan())
+    assert sum(got.values()) == 200_000
+    assert len(got) == 997
