@@ -36,7 +36,8 @@ def timeit(fn, iters):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("which", choices=["groupby", "groupby_pp",
-                                      "partition", "hash", "compact"])
+                                      "partition", "hash", "compact",
+                                      "sortcombine", "hashbytes"])
     ap.add_argument("--rows", type=int, default=125_000_000)
     ap.add_argument("--nkeys", type=int, default=1_000_000)
     ap.add_argument("--nparts", type=int, default=8)
@@ -108,6 +109,31 @@ def main():
         ms = timeit(run, args.iters)
     out["ms"] = ms
     out["grows_per_sec"] = args.rows / ms / 1e6
+    if args.which == "sortcombine":
+        from bigslice_amd.kernels import _C
+
+        def run():
+            ks, vs = _C.radix_sort_kv(keys, vals)
+            uq, sm, cnt = _C.segment_reduce_sorted(ks, vs, 0)
+            int(cnt.item())
+        out["ms"] = timeit(run, args.iters)
+        out["grows_per_s"] = args.rows / out["ms"] / 1e6
+    if args.which == "hashbytes":
+        import random
+
+        from bigslice_amd import strings
+        rng = random.Random(2)
+        words = ["w%05d" % rng.randrange(args.nkeys)
+                 for _ in range(min(args.rows, 10_000_000))]
+        data, offs = strings.pack_strings(words)
+        b, o = strings.to_device(data, offs, "cuda:0")
+        from bigslice_amd.kernels import _C
+
+        def run():
+            _C.hash_bytes64(b, o, 0x9ACB0442, 0x85EBCA6B)
+        out["ms"] = timeit(run, args.iters)
+        out["rows"] = len(words)
+        out["mwords_per_s"] = len(words) / out["ms"] / 1e3
     print(json.dumps(out))
 
 
