@@ -103,12 +103,15 @@ def main():
         def run():
             kernels.partition_frame(f, args.nparts, None)
         ms = timeit(run, args.iters)
-    else:
+    elif args.which == "hash":
         def run():
             kernels.hash_columns_device([keys], 0)
         ms = timeit(run, args.iters)
-    out["ms"] = ms
-    out["grows_per_sec"] = args.rows / ms / 1e6
+    else:
+        ms = None
+    if ms is not None:
+        out["ms"] = ms
+        out["grows_per_sec"] = args.rows / ms / 1e6
     if args.which == "sortcombine":
         from bigslice_amd.kernels import _C
 
