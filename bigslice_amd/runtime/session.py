@@ -286,8 +286,11 @@ class Session:
 def start(parallelism: int = None, device: str = None,
           executor: Executor = None, distributed: bool = None,
           trace_path: str = None, eventlog_path: str = None,
-          checkpoint_dir: str = None) -> Session:
-    """Create a session (exec.Start analog).
+          checkpoint_dir: str = None,
+          machine_combiners: bool = None) -> Session:
+    """Create a session (exec.Start analog; option set parity with
+    exec/session.go:98-176 — parallelism, executor choice, trace path,
+    eventer, machine combiners, checkpointing).
 
     distributed=True (or WORLD_SIZE>1 in the environment) starts the SPMD
     executor: one process per GPU over RCCL; every rank must call the
@@ -317,5 +320,8 @@ def start(parallelism: int = None, device: str = None,
                 store = FileStore(checkpoint_dir)
             executor = LocalExecutor(parallelism=parallelism,
                                      device=device, store=store)
+    if machine_combiners is not None and hasattr(executor,
+                                                 "machine_combiners"):
+        executor.machine_combiners = machine_combiners
     return Session(executor, parallelism, trace_path=trace_path,
                    eventlog_path=eventlog_path)
