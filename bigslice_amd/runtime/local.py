@@ -55,6 +55,28 @@ class LocalExecutor(Executor):
         self.machine_combiners = config.MACHINE_COMBINERS
         self._shared_combiners = {}  # phase id -> SharedPhaseCombiner
         self._shared_lock = threading.Lock()
+        # Task-completion ordering.  Default: hard per-task stream
+        # sync.  BIGSLICE_TASK_SYNC=0 switches to event-based ordering
+        # (consumers wait_event, driver synchronizes events for host
+        # reads) — measured neutral-to-slightly-negative on the
+        # BASELINE configs (the per-task syncs overlap across the 8
+        # worker threads, and letting phases overlap hurts the
+        # atomic-bound combine kernels), kept as an opt-in experiment.
+        import os as _os
+        self._hard_sync = _os.environ.get("BIGSLICE_TASK_SYNC",
+                                          "1") == "1"
+        self._events = {}  # task name -> torch.cuda.Event
+
+    def _record_done(self, task: Task, also_first: bool = False):
+        """Record the task's completion event on its stream (consumers
+        order against it; the driver synchronizes it for host reads)."""
+        if self._hard_sync or not self.device.startswith("cuda"):
+            return
+        ev = torch.cuda.Event()
+        ev.record()
+        self._events[task.name] = ev
+        if also_first:
+            self._events[task.group[0].name] = ev
 
     def _stream(self):
         """Per-worker-thread HIP stream: concurrent shard tasks overlap
@@ -94,9 +116,9 @@ class LocalExecutor(Executor):
                 if stream is not None:
                     with torch.cuda.stream(stream):
                         self._run_inner(task)
-                    # stored frames must be materialized before consumers
-                    # (possibly on other streams) read them
-                    stream.synchronize()
+                    if self._hard_sync:
+                        # stored frames materialized before any reader
+                        stream.synchronize()
                 else:
                     self._run_inner(task)
                 if span:
@@ -121,9 +143,14 @@ class LocalExecutor(Executor):
         ctx = TaskContext(device=self.device)
         dep_readers = []
         missing = []
+        use_cuda = self.device.startswith("cuda")
         for dep in task.deps:
             readers = []
             for h in dep.head_tasks:
+                if use_cuda:
+                    ev = self._events.get(h.name)
+                    if ev is not None:
+                        torch.cuda.current_stream().wait_event(ev)
                 try:
                     readers.append(self.store.open(
                         h.name, dep.partition, device=self.device))
@@ -148,6 +175,7 @@ class LocalExecutor(Executor):
             for _ in out:
                 pass
             self.store.put(task.name, 0, [], 0)
+            self._record_done(task)
             return
         shared = self._shared_combiner_for(task)
         if shared is not None:
@@ -163,6 +191,10 @@ class LocalExecutor(Executor):
                     self.store.put(first.name, pi, frames, rows)
                 with self._shared_lock:
                     self._shared_combiners.pop(id(task.group[0]), None)
+                # the finisher's kernels produced group[0]'s stored
+                # output; consumers of first.name must order on THIS
+                # task's event
+                self._record_done(task, also_first=True)
             if task is not task.group[0]:
                 for pi in range(task.num_partitions):
                     self.store.put(task.name, pi, [], 0)
@@ -172,6 +204,8 @@ class LocalExecutor(Executor):
                 # entries are written by the last finisher above; make
                 # sure empty markers exist if it was also the last.
                 pass
+            if task.name not in self._events:
+                self._record_done(task)
             return
         w = PartitionWriter(task.num_partitions, task.partitioner,
                             task.combiner, task.schema, self.device,
@@ -182,6 +216,7 @@ class LocalExecutor(Executor):
         for pi, frames in enumerate(buckets):
             rows = sum(len(f) for f in frames)
             self.store.put(task.name, pi, frames, rows)
+        self._record_done(task)
 
     def _shared_combiner_for(self, task: Task):
         """The phase's shared combiner, when machine-combiners mode
@@ -217,9 +252,13 @@ class LocalExecutor(Executor):
             return sc
 
     def reader(self, task: Task, partition: int) -> Reader:
+        ev = self._events.get(task.name)
+        if ev is not None:
+            ev.synchronize()  # host read: the event must be complete
         return self.store.open(task.name, partition, device="cpu")
 
     def discard(self, task: Task) -> None:
+        self._events.pop(task.name, None)
         self.store.discard_task(task.name)
 
     def shutdown(self) -> None:

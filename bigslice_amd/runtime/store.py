@@ -14,6 +14,12 @@ import struct
 import threading
 from typing import Dict, List, Tuple
 
+import torch
+
+# event-ordered executor mode (BIGSLICE_TASK_SYNC=0): cross-stream
+# readers must record_stream so the allocator defers block reuse
+_EVENT_MODE = os.environ.get("BIGSLICE_TASK_SYNC", "1") != "1"
+
 from ..frame import Frame
 from ..sliceio import Reader, codec
 
@@ -69,6 +75,17 @@ class MemoryStore(Store):
                     return None
                 f = frames[self.i]
                 self.i += 1
+                if f.device != "cpu" and _EVENT_MODE and \
+                        torch.cuda.is_available():
+                    # cross-stream consumers: keep the caching
+                    # allocator from reusing these blocks until the
+                    # reading stream passes this point (the producer
+                    # recorded its event; ordering is via wait_event
+                    # in the executor)
+                    cur = torch.cuda.current_stream()
+                    for c in f.columns:
+                        if isinstance(c, torch.Tensor) and c.is_cuda:
+                            c.record_stream(cur)
                 if device != "cpu" and f.device != device and \
                         not f.has_objects:
                     f = f.to(device, non_blocking=True)
