@@ -34,6 +34,16 @@ COMBINER_TARGET_KEYS = _env_int("BIGSLICE_COMBINER_TARGET_KEYS", 1 << 26)
 # Spill target bytes for external sort runs (reference cogroup.go:126-127
 # uses 32 MiB; device-scaled default 4 GiB keeps runs HBM-resident).
 SORT_SPILL_TARGET_BYTES = _env_int("BIGSLICE_SORT_SPILL_BYTES", 4 << 30)
+# Sorted runs spill in slices of this size so a k-way merge's per-run
+# readahead windows stay small (80 runs x 2 GiB windows would not fit
+# HBM beside a large resident input).
+SORT_SPILL_CHUNK_BYTES = _env_int("BIGSLICE_SORT_SPILL_CHUNK_BYTES",
+                                  256 << 20)
+# Spill backpressure: cap un-drained device bytes held by in-flight
+# D2H copies (the sort produces runs faster than the host link drains
+# them; unbounded backlog OOMs large jobs).
+SPILL_BACKPRESSURE_BYTES = _env_int("BIGSLICE_SPILL_BACKPRESSURE_BYTES",
+                                    32 << 30)
 
 # Initial hash-aggregate table capacity (slots); the table grows x2 on
 # probe-chain overflow (reference combiner grow policy, exec/combiner.go:47).

@@ -253,6 +253,22 @@ class Frame:
                 cols.append(c)
         return Frame(cols, self.prefix, combined_id=self.combined_id)
 
+    def to_pinned_host(self) -> "Frame":
+        """Copy device columns into pinned host DRAM (fast H2D
+        readback; the store's high-water tiering uses this)."""
+        cols: List[Column] = []
+        for c in self.columns:
+            if isinstance(c, torch.Tensor) and c.is_cuda:
+                dst = torch.empty(c.shape, dtype=c.dtype, device="cpu",
+                                  pin_memory=True)
+                dst.copy_(c)
+                cols.append(dst)
+            elif isinstance(c, SegmentedColumn):
+                cols.append(c.to("cpu"))
+            else:
+                cols.append(c)
+        return Frame(cols, self.prefix, combined_id=self.combined_id)
+
     def clone(self) -> "Frame":
         cols = []
         for c in self.columns:
@@ -326,3 +342,24 @@ class Frame:
     def __repr__(self):
         return (f"Frame({len(self)} rows, {self.schema}, "
                 f"device={self.device})")
+
+
+# -- HBM high-water check (store/writer tiering) ---------------------------
+
+_HW_TOTAL = None
+
+
+def over_high_water() -> bool:
+    """True when HBM allocation passes BIGSLICE_STORE_HIGH_WATER
+    (fraction of device total, default 0.85): accumulating writers and
+    the memory store tier device frames to pinned host DRAM past this
+    point so jobs larger than HBM keep running instead of OOMing."""
+    global _HW_TOTAL
+    if _HW_TOTAL is None:
+        _HW_TOTAL = (torch.cuda.get_device_properties(0).total_memory
+                     if torch.cuda.is_available() else 0)
+    if _HW_TOTAL == 0:
+        return False
+    import os
+    frac = float(os.environ.get("BIGSLICE_STORE_HIGH_WATER", "0.85"))
+    return torch.cuda.memory_allocated() > frac * _HW_TOTAL

@@ -156,11 +156,18 @@ class PartitionWriter:
             self.agg.add(frame)
             return
         parts = split_frame(frame, self.num_partitions, self.partitioner)
+        from ..frame import over_high_water
+        tier = frame.device != "cpu" and over_high_water()
         for pi, pf in enumerate(parts):
             if pf is None or len(pf) == 0:
                 continue
             if self.aggs is not None:
                 self.aggs[pi].add(pf)
+            elif tier:
+                # accumulating past the HBM high-water mark: hold the
+                # bucket in pinned host DRAM (e.g. a sort whose output
+                # cannot coexist with its input in HBM)
+                self.buckets[pi].append(pf.to_pinned_host())
             else:
                 self.buckets[pi].append(pf)
 

@@ -45,13 +45,25 @@ class Store:
 
 
 class MemoryStore(Store):
-    """In-memory partition buffers (frames stay on their device)."""
+    """In-memory partition buffers (frames stay on their device).
+
+    High-water tiering: when HBM allocation passes
+    BIGSLICE_STORE_HIGH_WATER (fraction of total, default 0.85) at put
+    time, the stored frames copy to PINNED host DRAM instead — jobs
+    whose outputs no longer fit beside their inputs (e.g. a 10B-row
+    single-GPU sort: 160 GB in + 160 GB out > 288 GB) keep running at
+    host-link speed instead of raising OOM.  open() moves tiered
+    frames back on demand."""
 
     def __init__(self):
         self._data: Dict[Tuple[str, int], Tuple[List[Frame], int]] = {}
         self._lock = threading.Lock()
 
     def put(self, task_name, partition, frames, rows):
+        from ..frame import over_high_water
+        if frames and over_high_water():
+            frames = [f.to_pinned_host() if f.device != "cpu" else f
+                      for f in frames]
         with self._lock:
             self._data[(task_name, partition)] = (frames, rows)
 

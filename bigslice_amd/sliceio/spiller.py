@@ -36,12 +36,21 @@ class _DiskBatch:
 class _HostBatch:
     """A host-resident batch whose D2H copy may still be in flight."""
 
-    __slots__ = ("frame", "event", "src_refs")
+    __slots__ = ("frame", "event", "src_refs", "src_bytes", "owner")
 
-    def __init__(self, frame: Frame, event, src_refs):
+    def __init__(self, frame: Frame, event, src_refs, src_bytes=0,
+                 owner=None):
         self.frame = frame
         self.event = event  # torch.cuda.Event or None
         self.src_refs = src_refs  # keep device tensors alive until done
+        self.src_bytes = src_bytes
+        self.owner = owner  # Spiller outstanding-bytes accounting
+
+    def _release(self):
+        if self.src_refs is not None:
+            if self.owner is not None:
+                self.owner._outstanding -= self.src_bytes
+            self.src_refs = None
 
     def ready(self, cpu_access: bool = False):
         if self.event is not None:
@@ -52,7 +61,7 @@ class _HostBatch:
                 # device consumption: stream ordering suffices
                 torch.cuda.current_stream().wait_event(self.event)
             if self.event.query():
-                self.src_refs = None
+                self._release()
         return self.frame
 
 
@@ -84,6 +93,7 @@ class Spiller:
         self._dir = dir
         self._tmpdir = None
         self._pin = pin
+        self._outstanding = 0  # device bytes held by in-flight D2H
 
     def _ensure_dir(self) -> str:
         if self._dir is None:
@@ -93,8 +103,15 @@ class Spiller:
 
     def spill(self, frame: Frame) -> int:
         """Spill one batch; returns bytes spilled.  Device batches copy
-        out asynchronously on the copy stream."""
+        out asynchronously on the copy stream; past
+        SPILL_BACKPRESSURE_BYTES of un-drained device memory the caller
+        blocks on the oldest copy (producers outrun the host link on
+        large jobs and the backlog would OOM HBM)."""
         self._release_completed()
+        from .. import config
+        while self._outstanding > config.SPILL_BACKPRESSURE_BYTES:
+            self._wait_oldest()
+            self._release_completed()
         nbytes = frame.nbytes()
         self.rows += len(frame)
         if self.host_used + nbytes <= self.host_budget:
@@ -122,6 +139,7 @@ class Spiller:
         cs.wait_event(ready)
         cols = []
         src_refs = []
+        src_bytes = 0
         with torch.cuda.stream(cs):
             for c in frame.columns:
                 if isinstance(c, torch.Tensor) and c.is_cuda:
@@ -130,11 +148,14 @@ class Spiller:
                     dst.copy_(c, non_blocking=True)
                     cols.append(dst)
                     src_refs.append(c)
+                    src_bytes += c.numel() * c.element_size()
                 else:
                     cols.append(c)
             done = torch.cuda.Event()
             done.record()
-        return _HostBatch(Frame(cols, frame.prefix), done, src_refs)
+        self._outstanding += src_bytes
+        return _HostBatch(Frame(cols, frame.prefix), done, src_refs,
+                          src_bytes, self)
 
     def _release_completed(self) -> None:
         """Drop device-source references of finished D2H copies so
@@ -143,7 +164,14 @@ class Spiller:
         for b in self.batches:
             if isinstance(b, _HostBatch) and b.src_refs is not None:
                 if b.event is None or b.event.query():
-                    b.src_refs = None
+                    b._release()
+
+    def _wait_oldest(self) -> None:
+        for b in self.batches:
+            if isinstance(b, _HostBatch) and b.src_refs is not None \
+                    and b.event is not None:
+                b.event.synchronize()
+                return
 
     def num_batches(self) -> int:
         return len(self.batches)

@@ -68,6 +68,18 @@ class SortReader(Reader):
         runs: List[List[Frame]] = []
         first_run: List[Optional[Frame]] = [None]
 
+        def spill_run(run: Frame):
+            # spill in slices: merge readahead then holds one small
+            # window per run, not the whole run (config
+            # SORT_SPILL_CHUNK_BYTES)
+            start = spiller.num_batches()
+            rows = len(run)
+            per_row = max(1, run.nbytes() // max(rows, 1))
+            step = max(1, int(config.SORT_SPILL_CHUNK_BYTES // per_row))
+            for off in range(0, rows, step):
+                spiller.spill(run.slice(off, min(off + step, rows)))
+            runs.append(list(range(start, spiller.num_batches())))
+
         def flush_run():
             nonlocal pending, pending_bytes
             if not pending:
@@ -82,13 +94,9 @@ class SortReader(Reader):
                 first_run[0] = run
                 return
             if first_run[0] is not None:
-                start = spiller.num_batches()
-                spiller.spill(first_run[0])
-                runs.append(list(range(start, spiller.num_batches())))
+                spill_run(first_run[0])
                 first_run[0] = None
-            start = spiller.num_batches()
-            spiller.spill(run)
-            runs.append(list(range(start, spiller.num_batches())))
+            spill_run(run)
 
         for f in self.source:
             pending.append(f)
