@@ -48,8 +48,9 @@ class _HostBatch:
 
     def _release(self):
         if self.src_refs is not None:
-            if self.owner is not None:
-                self.owner._outstanding -= self.src_bytes
+            if self.src_bytes:
+                with _outstanding_lock:
+                    _outstanding[0] -= self.src_bytes
             self.src_refs = None
 
     def ready(self, cpu_access: bool = False):
@@ -67,6 +68,12 @@ class _HostBatch:
 
 _copy_streams = {}
 _copy_lock = threading.Lock()
+
+# GLOBAL un-drained D2H backlog (device bytes alive only to feed
+# in-flight spill copies): concurrent tasks' spillers share the HBM,
+# so the backpressure cap must be process-wide.
+_outstanding = [0]
+_outstanding_lock = threading.Lock()
 
 
 def _copy_stream(device, direction: str = "d2h"):
@@ -93,7 +100,7 @@ class Spiller:
         self._dir = dir
         self._tmpdir = None
         self._pin = pin
-        self._outstanding = 0  # device bytes held by in-flight D2H
+
 
     def _ensure_dir(self) -> str:
         if self._dir is None:
@@ -109,7 +116,7 @@ class Spiller:
         large jobs and the backlog would OOM HBM)."""
         self._release_completed()
         from .. import config
-        while self._outstanding > config.SPILL_BACKPRESSURE_BYTES:
+        while _outstanding[0] > config.SPILL_BACKPRESSURE_BYTES:
             self._wait_oldest()
             self._release_completed()
         nbytes = frame.nbytes()
@@ -153,7 +160,8 @@ class Spiller:
                     cols.append(c)
             done = torch.cuda.Event()
             done.record()
-        self._outstanding += src_bytes
+        with _outstanding_lock:
+            _outstanding[0] += src_bytes
         return _HostBatch(Frame(cols, frame.prefix), done, src_refs,
                           src_bytes, self)
 
