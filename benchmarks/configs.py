@@ -84,17 +84,16 @@ def build_sort(nshard):
         if entry[0] == "lazy":
             # very large inputs: generate chunks inside the timed
             # region (EXTRA work, kept honest) so 160+ GB of synthetic
-            # input is never resident beside the sorted output
+            # input is never resident beside the sorted output.  Plain
+            # randint (no explicit Generator: per-thread Generator
+            # construction is unreliable under worker streams).
             _, per, hi, seed, device = entry
-            g = torch.Generator(device=device)
-            g.manual_seed(seed)
             for off in range(0, per, ctx.chunk):
                 n = min(ctx.chunk, per - off)
                 yield (torch.randint(0, hi, (n,), dtype=torch.int64,
-                                     device=device, generator=g),
+                                     device=device),
                        torch.randint(0, 1 << 30, (n,),
-                                     dtype=torch.int64, device=device,
-                                     generator=g))
+                                     dtype=torch.int64, device=device))
             return
         keys, vals = entry
         for off in range(0, keys.shape[0], ctx.chunk):
@@ -206,7 +205,9 @@ def main():
         rows = args.rows or 1_000_000_000
         per = rows // nshard
         for s in range(nshard):
-            if per * 16 * nshard > 100 << 30:
+            lazy_over = int(os.environ.get(
+                "BIGSLICE_BENCH_LAZY_OVER", str(100 << 30)))
+            if per * 16 * nshard > lazy_over:
                 _DATA[("sort", s)] = ("lazy", per, 1 << 62, 500 + s,
                                       device)
             else:
