@@ -356,8 +356,14 @@ def over_high_water() -> bool:
     point so jobs larger than HBM keep running instead of OOMing."""
     global _HW_TOTAL
     if _HW_TOTAL is None:
-        _HW_TOTAL = (torch.cuda.get_device_properties(0).total_memory
-                     if torch.cuda.is_available() else 0)
+        # device_count/properties can fail when first called from a
+        # worker thread on ROCm; LocalExecutor prewarms this from the
+        # main thread, and failures leave the cache unset for retry
+        try:
+            _HW_TOTAL = (torch.cuda.get_device_properties(0).total_memory
+                         if torch.cuda.is_available() else 0)
+        except Exception:
+            return False
     if _HW_TOTAL == 0:
         return False
     import os

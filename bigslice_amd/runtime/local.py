@@ -66,6 +66,10 @@ class LocalExecutor(Executor):
         self._hard_sync = _os.environ.get("BIGSLICE_TASK_SYNC",
                                           "1") == "1"
         self._events = {}  # task name -> torch.cuda.Event
+        # prewarm the HBM high-water cache from this (main) thread:
+        # CUDA device queries can fail when first issued from workers
+        from ..frame import over_high_water
+        over_high_water()
 
     def _record_done(self, task: Task, also_first: bool = False):
         """Record the task's completion event on its stream (consumers
