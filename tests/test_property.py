@@ -106,3 +106,47 @@ def test_partition_property(keys, nparts):
             continue
         for k in p.columns[0].tolist():
             assert seen.setdefault(k, pi) == pi
+
+
+@given(rows=st.lists(st.tuples(st.integers(0, 30),
+                               st.integers(-100, 100)),
+                     min_size=0, max_size=150),
+       ops=st.lists(st.sampled_from(["mapadd", "filter", "reshuffle",
+                                     "reshard"]), max_size=4),
+       nshard=st.integers(1, 4))
+@settings(max_examples=40, deadline=None)
+def test_random_pipeline_matches_oracle(rows, ops, nshard):
+    """Random op compositions agree with a plain-Python oracle."""
+    import bigslice_amd as bs
+
+    keys = torch.tensor([k for k, _ in rows], dtype=torch.int64)
+    vals = torch.tensor([v for _, v in rows], dtype=torch.int64)
+
+    def build():
+        s = bs.Const(nshard, keys, vals, prefix=1)
+        for i, op in enumerate(ops):
+            if op == "mapadd":
+                s = bs.Map(s, lambda k, v: (k, v + 3))
+            elif op == "filter":
+                s = bs.Filter(s, lambda k, v: (k & 3) != 1)
+            elif op == "reshuffle":
+                s = bs.Reshuffle(s)
+            else:
+                s = bs.Reshard(s, 2)
+        return bs.Reduce(s, "sum")
+
+    # oracle
+    oracle = {}
+    for k, v in rows:
+        for op in ops:
+            if op == "mapadd":
+                v = v + 3
+            elif op == "filter":
+                if (k & 3) == 1:
+                    break
+        else:
+            oracle[k] = oracle.get(k, 0) + v
+
+    sess = bs.start(parallelism=3, device="cpu")
+    got = dict(sess.run(bs.func(build)).scan())
+    assert got == oracle, (ops, nshard)
