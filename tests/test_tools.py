@@ -197,3 +197,14 @@ def test_bench_contract_cpu():
     assert d["higher_is_better"] is True and d["data"] == "synthetic"
     assert d["config"]["rows_per_gpu"] == 80000
     assert d["value"] > 0
+
+
+def test_urls_example():
+    sys.path.insert(0, "examples")
+    import importlib
+    urls = importlib.import_module("urls")
+    text = urls.synthesize_csv(3000)
+    sess = bs.start(parallelism=4)
+    counts = dict(sess.run(urls.domain_counts, 4, text).scan())
+    assert sum(counts.values()) == 3000
+    assert set(counts) <= set(urls._DOMAINS)
