@@ -44,26 +44,70 @@ class Store:
         raise NotImplementedError
 
 
+class _DiskEntry:
+    """A stored partition that overflowed the host tier to disk."""
+
+    __slots__ = ("path",)
+
+    def __init__(self, path: str):
+        self.path = path
+
+
 class MemoryStore(Store):
     """In-memory partition buffers (frames stay on their device).
 
-    High-water tiering: when HBM allocation passes
-    BIGSLICE_STORE_HIGH_WATER (fraction of total, default 0.85) at put
-    time, the stored frames copy to PINNED host DRAM instead — jobs
-    whose outputs no longer fit beside their inputs (e.g. a 10B-row
-    single-GPU sort: 160 GB in + 160 GB out > 288 GB) keep running at
-    host-link speed instead of raising OOM.  open() moves tiered
-    frames back on demand."""
+    Tiering: when HBM allocation passes BIGSLICE_STORE_HIGH_WATER
+    (fraction of total, default 0.85) at put time, stored frames copy
+    to PINNED host DRAM instead — jobs whose outputs no longer fit
+    beside their inputs (e.g. a 10B-row single-GPU sort: 160 GB in +
+    160 GB out > 288 GB) keep running at host-link speed instead of
+    raising OOM.  Past BIGSLICE_STORE_HOST_BUDGET_BYTES of host-tiered
+    frames (default 128 GB) the entry overflows to disk (CRC32 wire
+    codec).  open() moves tiered frames back on demand."""
 
     def __init__(self):
-        self._data: Dict[Tuple[str, int], Tuple[List[Frame], int]] = {}
+        self._data: Dict[Tuple[str, int], object] = {}
         self._lock = threading.Lock()
+        self._host_bytes = 0
+        self._host_keys: Dict[Tuple[str, int], int] = {}
+        self._tmpdir = None
+
+    def _tier_mode(self, frames) -> str:
+        from ..frame import over_high_water
+        if not frames or not over_high_water():
+            return "none"
+        budget = int(os.environ.get("BIGSLICE_STORE_HOST_BUDGET_BYTES",
+                                    str(128 << 30)))
+        return "host" if self._host_bytes < budget else "disk"
+
+    def _disk_path(self, task_name, partition) -> str:
+        import tempfile
+        with self._lock:
+            if self._tmpdir is None:
+                self._tmpdir = tempfile.TemporaryDirectory(
+                    prefix="bigslice-store-")
+        safe = task_name.replace("/", "_")
+        return os.path.join(self._tmpdir.name,
+                            f"{safe}-p{partition:03d}")
 
     def put(self, task_name, partition, frames, rows):
-        from ..frame import over_high_water
-        if frames and over_high_water():
+        mode = self._tier_mode(frames)
+        if mode == "host":
             frames = [f.to_pinned_host() if f.device != "cpu" else f
                       for f in frames]
+            nb = sum(f.nbytes() for f in frames)
+            with self._lock:
+                self._host_bytes += nb
+                self._host_keys[(task_name, partition)] = nb
+        elif mode == "disk":
+            path = self._disk_path(task_name, partition)
+            with open(path, "wb") as fp:
+                for f in frames:
+                    codec.encode_frame(f.to("cpu"), fp)
+            with self._lock:
+                self._data[(task_name, partition)] = (
+                    _DiskEntry(path), rows)
+            return
         with self._lock:
             self._data[(task_name, partition)] = (frames, rows)
 
@@ -77,6 +121,22 @@ class MemoryStore(Store):
         if entry is None:
             raise KeyError(f"no output for {task_name} p{partition}")
         frames, _ = entry
+        if isinstance(frames, _DiskEntry):
+            path = frames.path
+
+            class _D(Reader):
+                def __init__(self):
+                    self._fp = open(path, "rb")
+                    self._size = os.path.getsize(path)
+
+                def read(self):
+                    if self._fp.tell() >= self._size:
+                        return None
+                    return codec.decode_frame(self._fp, device)
+
+                def close(self):
+                    self._fp.close()
+            return _D()
 
         class _R(Reader):
             def __init__(self):
@@ -110,12 +170,20 @@ class MemoryStore(Store):
         if entry is None:
             return (0, 0)
         frames, rows = entry
+        if isinstance(frames, _DiskEntry):
+            return (os.path.getsize(frames.path), rows)
         return (sum(f.nbytes() for f in frames), rows)
 
     def discard_task(self, task_name):
         with self._lock:
             for k in [k for k in self._data if k[0] == task_name]:
-                del self._data[k]
+                frames, _ = self._data.pop(k)
+                self._host_bytes -= self._host_keys.pop(k, 0)
+                if isinstance(frames, _DiskEntry):
+                    try:
+                        os.unlink(frames.path)
+                    except OSError:
+                        pass
 
 
 class FileStore(Store):

@@ -377,3 +377,24 @@ def test_top_n():
     counts = {"a": 5, "b": 9, "c": 1, "d": 9}
     assert top_n(counts, 2) == [("d", 9), ("b", 9)]
     assert top_n(counts, 10)[-1] == ("c", 1)
+
+
+def test_store_disk_tier_roundtrip(monkeypatch):
+    # Force the disk tier and check exact roundtrip + cleanup.
+    from bigslice_amd.frame import Frame
+    from bigslice_amd.runtime.store import MemoryStore, _DiskEntry
+    st = MemoryStore()
+    monkeypatch.setattr(st, "_tier_mode", lambda frames: "disk")
+    k = torch.arange(5000, dtype=torch.int64)
+    v = torch.randint(-9, 9, (5000,), dtype=torch.int64)
+    st.put("t", 2, [Frame([k, v], prefix=1)], 5000)
+    entry = st._data[("t", 2)][0]
+    assert isinstance(entry, _DiskEntry) and os.path.exists(entry.path)
+    assert st.stat("t", 2)[1] == 5000
+    back = list(st.open("t", 2))
+    assert torch.equal(back[0].columns[0], k)
+    assert torch.equal(back[0].columns[1], v)
+    path = entry.path
+    st.discard_task("t")
+    assert not os.path.exists(path)
+    assert not st.has("t", 2)
