@@ -255,12 +255,18 @@ class Frame:
 
     def to_pinned_host(self) -> "Frame":
         """Copy device columns into pinned host DRAM (fast H2D
-        readback; the store's high-water tiering uses this)."""
+        readback; the store's high-water tiering uses this).  Falls
+        back to pageable memory when the pinned allocator is
+        exhausted."""
         cols: List[Column] = []
         for c in self.columns:
             if isinstance(c, torch.Tensor) and c.is_cuda:
-                dst = torch.empty(c.shape, dtype=c.dtype, device="cpu",
-                                  pin_memory=True)
+                try:
+                    dst = torch.empty(c.shape, dtype=c.dtype,
+                                      device="cpu", pin_memory=True)
+                except RuntimeError:
+                    dst = torch.empty(c.shape, dtype=c.dtype,
+                                      device="cpu")
                 dst.copy_(c)
                 cols.append(dst)
             elif isinstance(c, SegmentedColumn):
