@@ -150,8 +150,12 @@ class Spiller:
         with torch.cuda.stream(cs):
             for c in frame.columns:
                 if isinstance(c, torch.Tensor) and c.is_cuda:
-                    dst = torch.empty(c.shape, dtype=c.dtype, device="cpu",
-                                      pin_memory=True)
+                    try:
+                        dst = torch.empty(c.shape, dtype=c.dtype,
+                                          device="cpu", pin_memory=True)
+                    except RuntimeError:  # pinned allocator exhausted
+                        dst = torch.empty(c.shape, dtype=c.dtype,
+                                          device="cpu")
                     dst.copy_(c, non_blocking=True)
                     cols.append(dst)
                     src_refs.append(c)
