@@ -50,15 +50,24 @@ class Executor:
 def _reachable(roots: Sequence[Task]) -> List[Task]:
     out: List[Task] = []
     seen: Set[int] = set()
-    stack = list(roots)
+    seen_lists: Set[int] = set()  # consumers share per-phase head lists
+    stack: List[Task] = []
+    for t in roots:
+        if id(t) not in seen:
+            seen.add(id(t))
+            stack.append(t)
     while stack:
         t = stack.pop()
-        if id(t) in seen:
-            continue
-        seen.add(id(t))
         out.append(t)
         for dep in t.deps:
-            stack.extend(dep.head_tasks)
+            heads = dep.head_tasks
+            if id(heads) in seen_lists:
+                continue
+            seen_lists.add(id(heads))
+            for h in heads:
+                if id(h) not in seen:
+                    seen.add(id(h))
+                    stack.append(h)
     return out
 
 
@@ -73,17 +82,33 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
     phase_tasks: Dict[int, List[Task]] = {}
     for t in tasks:
         phase_tasks.setdefault(gid(t), []).append(t)
-    # consumers of each phase (tasks with a dep whose heads are in it)
+    # consumers of each phase (tasks with a dep whose heads are in it).
+    # Consumers share per-phase head LISTS, so the group scan runs once
+    # per unique list (O(tasks + heads)), not per consumer x heads.
     phase_consumers: Dict[int, List[Task]] = {}
     dep_groups: Dict[int, List[int]] = {}  # task -> dep phase ids
+    list_groups: Dict[int, List[int]] = {}  # id(head list) -> group ids
+
+    def groups_of(heads) -> List[int]:
+        got = list_groups.get(id(heads))
+        if got is None:
+            got = []
+            sg = set()
+            # heads normally share one group; scan defensively for
+            # hand-built graphs (tests) with ungrouped heads
+            for h in heads:
+                g = gid(h)
+                if g not in sg:
+                    sg.add(g)
+                    got.append(g)
+            list_groups[id(heads)] = got
+        return got
+
     for t in tasks:
         groups = []
         seen_g = set()
         for dep in t.deps:
-            # heads normally share one group; scan defensively for
-            # hand-built graphs (tests) with ungrouped heads
-            for h in dep.head_tasks:
-                g = gid(h)
+            for g in groups_of(dep.head_tasks):
                 if g not in seen_g:
                     seen_g.add(g)
                     groups.append(g)

@@ -34,7 +34,12 @@ class TaskDep:
 
     def __init__(self, head_tasks: Sequence["Task"], partition: int,
                  expand: bool = False, combiner=None):
-        self.head_tasks = list(head_tasks)
+        # alias list inputs: all consumers of a phase then share ONE
+        # head list, letting the evaluator dedupe phase scans by list
+        # identity (O(tasks+heads) bookkeeping instead of
+        # O(consumers x heads); callers pass completed phase lists)
+        self.head_tasks = (head_tasks if isinstance(head_tasks, list)
+                           else list(head_tasks))
         self.partition = partition
         self.expand = expand
         self.combiner = combiner
