@@ -163,6 +163,9 @@ class Result(Slice):
     def scanner(self) -> Scanner:
         return Scanner(self.open())
 
+    def __iter__(self):
+        return self.scan()
+
     def scan(self):
         """Iterate result rows."""
         return self.scanner().rows()
@@ -269,6 +272,13 @@ class Session:
 
     def must(self, funcv: FuncValue, *args) -> Result:
         return self.run(funcv, *args)
+
+    def __enter__(self) -> "Session":
+        return self
+
+    def __exit__(self, *exc):
+        self.shutdown()
+        return False
 
     def shutdown(self):
         if self.tracer is not None and self.trace_path:

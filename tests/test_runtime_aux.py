@@ -398,3 +398,12 @@ def test_store_disk_tier_roundtrip(monkeypatch):
     st.discard_task("t")
     assert not os.path.exists(path)
     assert not st.has("t", 2)
+
+
+def test_session_context_manager_and_result_iter():
+    def build():
+        return bs.Const(2, torch.tensor([1, 2, 1], dtype=torch.int64),
+                        torch.tensor([5, 6, 7], dtype=torch.int64))
+    with bs.start(parallelism=2, device="cpu") as sess:
+        rows = sorted(sess.run(bs.func(build)))
+    assert rows == [(1, 5), (1, 7), (2, 6)]
