@@ -208,3 +208,21 @@ def test_urls_example():
     counts = dict(sess.run(urls.domain_counts, 4, text).scan())
     assert sum(counts.values()) == 3000
     assert set(counts) <= set(urls._DOMAINS)
+
+
+def test_benchdiff(tmp_path):
+    from bigslice_amd.tools import benchdiff
+    rec = {"metric": "m", "value": 100.0, "unit": "u",
+           "higher_is_better": True,
+           "config": {"model": "M", "rows_total": 10}}
+    base = tmp_path / "base"
+    base.mkdir()
+    (base / "base.json").write_text(json.dumps(rec) + "\n")
+    good = dict(rec, value=98.0)
+    bad = dict(rec, value=50.0)
+    (tmp_path / "new.json").write_text(json.dumps(good) + "\n")
+    assert benchdiff.main([str(tmp_path / "new.json"),
+                           "--baseline-dir", str(base)]) == 0
+    (tmp_path / "new2.json").write_text(json.dumps(bad) + "\n")
+    assert benchdiff.main([str(tmp_path / "new2.json"),
+                           "--baseline-dir", str(base)]) == 1
