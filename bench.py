@@ -2,10 +2,12 @@
 
 BASELINE.json north-star config 3: group-by-sum over 1B int64 rows in 8
 shards (1B at 8 GPUs; per-GPU rows fixed -> weak scaling).  Each step runs
-the full engine pipeline on pre-generated synthetic device data: murmur3
-hash + fused partition scatter (K4), producer-side hash-aggregate
-pre-combine (K9), RCCL all-to-allv over xGMI at N>1, and the consumer-side
-final hash-aggregate.
+the full engine pipeline on pre-generated synthetic device data: the
+adaptive producer-side combine (LDS-tier hash insert / streaming global
+insert / radix-sort+reduce-by-key, chosen from a sampled key
+cardinality: K9/K10/K16), the fused hash+scatter partitioner (K4),
+RCCL all-to-allv over xGMI at N>1, and the consumer-side final
+aggregate (pre-combined passthrough at N=1).
 
 Usage:
   python bench.py --gpus 1 --steps 5 --warmup 2
