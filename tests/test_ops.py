@@ -302,3 +302,13 @@ def test_mixed_combiners_over_shared_producer():
 
     res = bs.slicetest.run(build)
     assert sorted(res.scan()) == [(1, [12], [7]), (2, [3], [3])]
+
+
+def test_reshard_up_and_down():
+    k = torch.arange(100, dtype=torch.int64) % 7
+    v = torch.ones(100, dtype=torch.int64)
+    for nshard in (1, 3, 16):
+        res = run_slice(lambda: bs.Reduce(
+            bs.Reshard(bs.Const(2, k, v, prefix=1), nshard), "sum"))
+        rows = sorted_rows(res)
+        assert len(rows) == 7 and sum(c for _, c in rows) == 100
