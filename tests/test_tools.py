@@ -226,3 +226,17 @@ def test_benchdiff(tmp_path):
     (tmp_path / "new2.json").write_text(json.dumps(bad) + "\n")
     assert benchdiff.main([str(tmp_path / "new2.json"),
                            "--baseline-dir", str(base)]) == 1
+
+
+def test_wordcount_example_cli(tmp_path):
+    import subprocess
+    import sys as _sys
+    path = tmp_path / "in.txt"
+    path.write_text("a b a\nc a\n")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [_sys.executable, os.path.join(repo, "examples", "wordcount.py"),
+         str(path)], capture_output=True, text=True, timeout=300,
+        cwd=repo)
+    assert out.returncode == 0, out.stderr[-1000:]
+    assert out.stdout.splitlines()[0].split() == ["3", "a"]
