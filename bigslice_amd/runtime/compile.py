@@ -199,7 +199,13 @@ class Compiler:
             for t in prev))
         if not needs_shuffle:
             return prev
-        key = (id(result), num_partitions)
+        # the pass-through tasks carry the consumer's partitioner AND
+        # combiner: consumers with different combine specs must not
+        # share them (same rule as the main memo key)
+        comb_key = combiner.key() if combiner is not None else None
+        key = (id(result), num_partitions,
+               id(partitioner) if partitioner is not None else None,
+               comb_key)
         if key in self.memo:
             return self.memo[key]
         nshard = len(prev)

@@ -117,3 +117,21 @@ def test_cache_decision_cuts_deps(tmp_path):
     assert len(computed) == n_first
     # and the cached tasks have no deps
     assert all(not t.deps for t in r2.tasks)
+
+
+def test_result_reuse_distinct_combiners():
+    # Two consumers with different combiners over ONE prior Result must
+    # not share pass-through shuffle tasks (max previously returned
+    # sum's values).
+    def build_src():
+        k = torch.tensor([1, 1, 2, 2], dtype=torch.int64)
+        v = torch.tensor([5, 9, 3, 4], dtype=torch.int64)
+        return bs.Const(2, k, v, prefix=1)
+
+    def build_two(res):
+        return bs.Cogroup(bs.Reduce(res, "sum"), bs.Reduce(res, "max"))
+
+    sess = bs.start(parallelism=2, device="cpu")
+    res = sess.run(bs.func(build_src))
+    out = sorted(sess.run(bs.func(build_two), res).scan())
+    assert out == [(1, [14], [9]), (2, [7], [4])], out
