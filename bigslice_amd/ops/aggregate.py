@@ -205,15 +205,6 @@ class TensorAggregator:
                         [v[off:stop] for v in self.vals],
                         self.schema.prefix, combined_id=cid)
 
-    def num_keys(self) -> int:
-        if self._pt_frames is not None:
-            return sum(len(f) for f in self._pt_frames)
-        if self._table is not None:
-            self._flush_table_pending()
-            return self._table.rows and self._table.finish()[0].shape[0]
-        self._flush()
-        return 0 if self.keys is None else self.keys[0].shape[0]
-
 
 def _combine_once(keys: List[torch.Tensor], vals: List[torch.Tensor],
                   agg: Aggregation):
