@@ -6,7 +6,9 @@ and decode verifies the checksum (corruption detection parity with
 sliceio/codec.go:111,226-234).  Unlike gob, the wire layout here IS the device
 layout: numeric columns are raw little-endian column buffers that can be
 hipMemcpy'd straight into HBM with no per-row decode; object columns are
-pickled.
+pickled (object columns are only ever decoded from files this process
+wrote: spill, cache and checkpoint stores — the same trust model as
+the reference's gob).
 
 Batch layout:
     magic  u32 = 0xB16S11CE (0xB16511CE)
@@ -40,10 +42,9 @@ class CorruptionError(IOError):
 
 def _tensor_bytes(t: torch.Tensor) -> bytes:
     a = t.detach().cpu().contiguous()
-    if a.dtype == torch.bfloat16 or a.dtype == torch.float16:
+    if a.dtype in (torch.bfloat16, torch.float16):
+        # numpy has no bf16: ship the raw 16-bit pattern
         return a.view(torch.int16).numpy().tobytes()
-    if a.dtype in (torch.uint32, torch.uint64):
-        return a.numpy().tobytes()
     return a.numpy().tobytes()
 
 
