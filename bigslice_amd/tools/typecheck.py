@@ -83,12 +83,16 @@ def check_file(path: str) -> List[str]:
 
 
 def main():
-    if len(sys.argv) < 2:
+    args = [a for a in sys.argv[1:] if not a.startswith("-")]
+    if not args or "-h" in sys.argv or "--help" in sys.argv:
         print(__doc__)
         sys.exit(2)
     findings: List[str] = []
-    for path in sys.argv[1:]:
-        findings.extend(check_file(path))
+    for path in args:
+        try:
+            findings.extend(check_file(path))
+        except (OSError, SyntaxError) as e:
+            findings.append(f"{path}: {e}")
     for f in findings:
         print(f)
     sys.exit(1 if findings else 0)
