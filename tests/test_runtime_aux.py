@@ -407,3 +407,17 @@ def test_session_context_manager_and_result_iter():
     with bs.start(parallelism=2, device="cpu") as sess:
         rows = sorted(sess.run(bs.func(build)))
     assert rows == [(1, 5), (1, 7), (2, 6)]
+
+
+def test_machine_combiners_off_matches_on():
+    def build():
+        k = torch.arange(5000, dtype=torch.int64) % 37
+        v = torch.ones(5000, dtype=torch.int64)
+        return bs.Reduce(bs.Const(6, k, v, prefix=1), "sum")
+    outs = []
+    for mc in (True, False):
+        sess = bs.start(parallelism=4, device="cpu",
+                        machine_combiners=mc)
+        outs.append(sorted(sess.run(bs.func(build)).scan()))
+    assert outs[0] == outs[1]
+    assert len(outs[0]) == 37
