@@ -131,6 +131,15 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
                 rem += 1
         phase_remaining[g] = rem
 
+    # Driver wakeups are counter-based: notify only when the last root
+    # completes (or on error/lost), not on every task finish — at 5000
+    # shards the per-finish notify_all was the scheduler's top cost
+    # (one main-thread wakeup + O(roots) rescan per task).  The 1 s
+    # timed wait below remains the backstop for quiescent re-walks and
+    # state flipped by concurrent evaluations.
+    root_ids: Set[int] = {id(r) for r in roots}
+    roots_left = sum(1 for i in root_ids if i not in ok_counted)
+
     pool = getattr(executor, "pool", None)
     own_pool = None
     if pool is None:
@@ -182,9 +191,12 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
     def note_incomplete(h: Task):
         """A task previously counted OK went LOST/stale: restore its
         phase counter so consumers wait again."""
+        nonlocal roots_left
         if id(h) in ok_counted and h.state != TaskState.OK:
             ok_counted.discard(id(h))
             phase_remaining[gid(h)] = phase_remaining.get(gid(h), 0) + 1
+            if id(h) in root_ids:
+                roots_left += 1
 
     def start(t: Task):
         t.set_state(TaskState.WAITING)
@@ -202,6 +214,7 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
         _finish(t)
 
     def _finish(t: Task):
+        nonlocal roots_left
         st = t.state
         with lock:
             if st == TaskState.OK:
@@ -216,7 +229,10 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
                                     TaskState.INIT, TaskState.LOST) and \
                                     deps_ready(c):
                                 start(c)
-                done.notify_all()
+                    if id(t) in root_ids:
+                        roots_left -= 1
+                        if roots_left <= 0:
+                            done.notify_all()
             elif st == TaskState.LOST:
                 # resubmit, re-walking deps that were also lost:
                 # reconcile phase counters for dep members whose OK was

@@ -79,19 +79,23 @@ class Task:
 
     @property
     def state(self) -> TaskState:
-        with self._cond:
-            return self._state
+        # Lock-free: a single attribute read is atomic under the GIL
+        # and the value is an immutable enum.  Writers still lock (the
+        # condition ordering matters only for wait_state, which takes
+        # the lock).  State reads are the evaluator's hottest call
+        # (~3 per task per schedule pass); the lock here contended
+        # directly with every set_state notify.
+        return self._state
 
     @property
     def error(self) -> Optional[BaseException]:
-        with self._cond:
-            return self._err
+        return self._err
 
     def set_state(self, s: TaskState, err: BaseException = None) -> None:
         with self._cond:
-            self._state = s
             if s == TaskState.ERR:
-                self._err = err
+                self._err = err  # published before state for lock-free readers
+            self._state = s
             if s == TaskState.OK:
                 self.consecutive_lost = 0
             elif s == TaskState.LOST:
