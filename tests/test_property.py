@@ -112,7 +112,7 @@ def test_partition_property(keys, nparts):
                                st.integers(-100, 100)),
                      min_size=0, max_size=150),
        ops=st.lists(st.sampled_from(["mapadd", "filter", "reshuffle",
-                                     "reshard"]), max_size=4),
+                                     "reshard", "flatmap"]), max_size=4),
        nshard=st.integers(1, 4))
 @settings(max_examples=40, deadline=None)
 def test_random_pipeline_matches_oracle(rows, ops, nshard):
@@ -129,6 +129,10 @@ def test_random_pipeline_matches_oracle(rows, ops, nshard):
                 s = bs.Map(s, lambda k, v: (k, v + 3))
             elif op == "filter":
                 s = bs.Filter(s, lambda k, v: (k & 3) != 1)
+            elif op == "flatmap":
+                # vectorized: each row (k,v) -> (k,v),(k,-v)
+                s = bs.Flatmap(s, lambda k, v: (torch.cat([k, k]),
+                                                torch.cat([v, -v])))
             elif op == "reshuffle":
                 s = bs.Reshuffle(s)
             else:
@@ -136,16 +140,17 @@ def test_random_pipeline_matches_oracle(rows, ops, nshard):
         return bs.Reduce(s, "sum")
 
     # oracle
+    data = list(rows)
+    for op in ops:
+        if op == "mapadd":
+            data = [(k, v + 3) for k, v in data]
+        elif op == "filter":
+            data = [(k, v) for k, v in data if (k & 3) != 1]
+        elif op == "flatmap":
+            data = [p for k, v in data for p in ((k, v), (k, -v))]
     oracle = {}
-    for k, v in rows:
-        for op in ops:
-            if op == "mapadd":
-                v = v + 3
-            elif op == "filter":
-                if (k & 3) == 1:
-                    break
-        else:
-            oracle[k] = oracle.get(k, 0) + v
+    for k, v in data:
+        oracle[k] = oracle.get(k, 0) + v
 
     sess = bs.start(parallelism=3, device="cpu")
     got = dict(sess.run(bs.func(build)).scan())
