@@ -102,12 +102,46 @@ def memiter_stress(iters: int, device: str):
     print(f"memiter OK: {iters} iterations")
 
 
+def oom_stress(size: int, device: str):
+    """OOM *reporting* check (cmd/slicer/oom.go:20-31: the reference
+    deliberately OOMs a worker and only cares that the error is
+    reported cleanly).  MI355X analog: one absurd single-tensor
+    allocation inside a Map UDF — the HIP/host allocator rejects it
+    immediately (no gradual memory pressure, box-safe) — and the
+    session must surface it as an ordinary task error naming the
+    allocation, not hang or die."""
+
+    def build(nbytes):
+        src = bs.Const(2, torch.arange(4, dtype=torch.int64))
+
+        def boom(k):
+            torch.empty(nbytes, dtype=torch.uint8,
+                        device=device if device != "cpu" else "cpu")
+            raise AssertionError("not reached")
+        return bs.Map(src, boom, out_schema=(int,))
+
+    fv = bs.func(build)
+    sess = bs.start(parallelism=2, device=device)
+    try:
+        sess.run(fv, size)
+    except Exception as e:
+        msg = str(e).lower()
+        assert ("memory" in msg or "alloc" in msg), (
+            f"OOM surfaced as unrelated error: {e!r}")
+        print(f"oom OK: allocation of {size} bytes reported as task "
+              f"error: {type(e).__name__}")
+        return
+    raise AssertionError("oom run unexpectedly succeeded")
+
+
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("cmd", choices=["reduce", "cogroup", "memiter"])
+    ap.add_argument("cmd", choices=["reduce", "cogroup", "memiter", "oom"])
     ap.add_argument("--nshard", type=int, default=8)
     ap.add_argument("--nkey", type=int, default=100_000)
     ap.add_argument("--iters", type=int, default=5)
+    ap.add_argument("--size", type=int, default=1 << 48,
+                    help="oom allocation size in bytes")
     ap.add_argument("--device", type=str, default=None)
     args = ap.parse_args()
     device = args.device or (
@@ -116,6 +150,8 @@ def main():
         reduce_stress(args.nshard, args.nkey, device)
     elif args.cmd == "cogroup":
         cogroup_stress(args.nshard, args.nkey, device)
+    elif args.cmd == "oom":
+        oom_stress(args.size, device)
     else:
         memiter_stress(args.iters, device)
 

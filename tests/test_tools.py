@@ -40,6 +40,17 @@ def test_slicer_memiter_cpu():
     assert p.returncode == 0, p.stderr
 
 
+def test_slicer_oom_reported_cpu():
+    """OOM surfaces as a clean task error (cmd/slicer/oom.go analog);
+    the allocator rejects the absurd size fail-fast, so this is safe."""
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.slicer", "oom",
+         "--device", "cpu"],
+        capture_output=True, text=True)
+    assert p.returncode == 0, p.stderr
+    assert "oom OK" in p.stdout
+
+
 def test_badfuncs_late_registration():
     p = subprocess.run(
         [sys.executable, "-m", "bigslice_amd.tools.badfuncs", "late"],
