@@ -231,8 +231,11 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
                                 start(c)
                     if id(t) in root_ids:
                         roots_left -= 1
-                        if roots_left <= 0:
-                            done.notify_all()
+                if id(t) in root_ids and roots_left <= 0:
+                    # notify outside the freshly-counted check too: a
+                    # root re-finishing after an external LOST/re-run
+                    # may still be in ok_counted
+                    done.notify_all()
             elif st == TaskState.LOST:
                 # resubmit, re-walking deps that were also lost:
                 # reconcile phase counters for dep members whose OK was

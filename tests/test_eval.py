@@ -151,3 +151,38 @@ def test_concurrent_evals_share_tasks():
         t.join()
     assert not errs
     assert tasks[0].state == TaskState.OK
+
+
+def test_roots_already_ok_no_rerun():
+    # All tasks OK at entry: evaluate returns without running anything
+    # (counter-based root accounting must handle pre-counted roots).
+    tasks = chain(3)
+    for t in tasks:
+        t.set_state(TaskState.OK)
+    ex = FakeExecutor()
+    evaluate(ex, [tasks[0]])
+    assert ex.runs == {}
+
+
+def test_root_revoked_ok_recounted():
+    # r1 (a root) depends on r0 (also a root).  r1's first attempt
+    # discovers r0's output missing: it marks r0 LOST and reports
+    # itself LOST.  Both must re-run and evaluate must complete
+    # promptly (roots_left reconciliation in note_incomplete).
+    r0 = make_task("r0")
+    r1 = make_task("r1", [TaskDep([r0], 0)])
+
+    class RevokingExecutor(FakeExecutor):
+        def run(self, task):
+            with self.lock:
+                self.runs[task.name] = self.runs.get(task.name, 0) + 1
+                first_r1 = task.name == "r1" and self.runs["r1"] == 1
+            task.set_state(TaskState.RUNNING)
+            if first_r1:
+                r0.set_state(TaskState.LOST)  # dep output vanished
+                task.set_state(TaskState.LOST)
+            else:
+                task.set_state(TaskState.OK)
+
+    evaluate(RevokingExecutor(), [r0, r1])
+    assert r0.state == TaskState.OK and r1.state == TaskState.OK
