@@ -254,3 +254,54 @@ def test_dist_checkpoint_restart():
     assert rows1 == expect
     assert rows2 == expect  # restart serves from checkpoints
     assert n2 == n1  # no shard recomputation on restart
+
+
+def _exchange3_worker(rank, world, port, q):
+    """Drive Comm.exchange_buckets directly at world 3 with asymmetric
+    buckets, both metadata planes.  World 3 breaks the rank symmetry
+    that world 2 can hide (the round-end scaling bench runs N=4/8 on
+    this same code; only the transport differs under RCCL)."""
+    _init(rank, world, port)
+    from bigslice_amd.parallel.comm import Comm
+    from bigslice_amd.frame import Frame
+    from bigslice_amd import schema_of
+
+    comm = Comm(rank, world, "cpu")
+    assert comm.tensor_exchange_ok  # gloo drives the tensor path
+    schema = schema_of(int, int)
+    names = [f"t{d}" for d in range(world)]
+    name_index = {n: i for i, n in enumerate(names)}
+
+    def mk(dest, j):
+        # rows encode (src, dest, j) so routing is fully checkable
+        n = (rank + 1) * (j + 1)  # asymmetric sizes
+        k = torch.full((n,), rank * 100 + dest * 10 + j,
+                       dtype=torch.int64)
+        return Frame([k, torch.arange(n, dtype=torch.int64)], 1)
+
+    # rank r sends (r+dest) % 2 + 1 buckets to each dest
+    send = []
+    for dest in range(world):
+        nb = (rank + dest) % 2 + 1
+        send.append([(names[dest], dest * 4 + j, mk(dest, j))
+                     for j in range(nb)])
+
+    for kw in ({"name_index": name_index, "index_name": names}, {}):
+        got = comm.exchange_buckets(send, schema, **kw)
+        # expect: from each src, (src+rank)%2+1 buckets for me
+        expect = []
+        for src in range(world):
+            nb = (src + rank) % 2 + 1
+            for j in range(nb):
+                expect.append((names[rank], rank * 4 + j,
+                               src * 100 + rank * 10 + j,
+                               (src + 1) * (j + 1)))
+        summary = sorted((t, p, int(f.columns[0][0]), len(f))
+                         for (t, p, f) in got)
+        assert summary == sorted(expect), (kw, summary)
+    q.put((rank, "ok"))
+
+
+def test_dist_tensor_exchange_world3():
+    results = _run_workers(_exchange3_worker, world=3)
+    assert all(v == "ok" for v in results.values())
