@@ -155,7 +155,12 @@ def main():
                 "device": device,
             },
         }
-        print(json.dumps(out))
+        print(json.dumps(out), flush=True)
+    if distributed:
+        # Clean teardown: ranks exiting at different times with a live
+        # process group can SIGABRT in the backend's helper threads.
+        torch.distributed.barrier()
+        torch.distributed.destroy_process_group()
 
 
 if __name__ == "__main__":
