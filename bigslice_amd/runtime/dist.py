@@ -88,7 +88,12 @@ class DistExecutor(Executor):
         # or whose outputs survive in a persistent store (restart).
         if all(t.state == TaskState.OK for t in phase):
             return
-        if self._phase_checkpointed(phase):
+        # The checkpoint decision must be COLLECTIVE: a rank that owns
+        # no shards and no partitions of this phase (nshard < world) is
+        # vacuously "complete" locally, and skipping on that alone
+        # deadlocks the other ranks' phase collectives.  One tiny
+        # all_reduce: skip only if NO rank is missing outputs.
+        if not comm.any_flag(not self._phase_checkpointed(phase)):
             for t in phase:
                 t.set_state(TaskState.OK)
             return

@@ -305,3 +305,31 @@ def _exchange3_worker(rank, world, port, q):
 def test_dist_tensor_exchange_world3():
     results = _run_workers(_exchange3_worker, world=3)
     assert all(v == "ok" for v in results.values())
+
+
+def _narrow_phase_worker(rank, world, port, q):
+    """nshard < world: rank 2 owns no shards and no partitions of the
+    2-shard phases.  The checkpoint-skip decision must be collective or
+    that rank deadlocks the others' phase collectives (regression:
+    vacuously-true _phase_checkpointed at world 3)."""
+    _init(rank, world, port)
+    import bigslice_amd as bs
+
+    def b_cogroup(m):
+        a = bs.Const(m, torch.tensor([1, 2, 1], dtype=torch.int64),
+                     torch.tensor([10, 20, 30], dtype=torch.int64))
+        b = bs.Const(m, torch.tensor([2, 3], dtype=torch.int64),
+                     torch.tensor([5, 6], dtype=torch.int64))
+        return bs.Cogroup(a, b)
+
+    fv = bs.func(b_cogroup)
+    sess = bs.start(distributed=True, device="cpu")
+    r = sess.run(fv, 2)
+    q.put((rank, sorted((k, sorted(x), sorted(y))
+                        for k, x, y in r.scan())))
+
+
+def test_dist_nshard_less_than_world():
+    results = _run_workers(_narrow_phase_worker, world=3)
+    assert results[0] == [(1, [10, 30], []), (2, [20], [5]),
+                          (3, [], [6])]
