@@ -1,8 +1,9 @@
 """Distributed executor tests: SPMD over gloo, world_size=2, CPU.
 
-These exercise the same code paths the RCCL/GPU path uses (phase schedule,
-bucket exchange, registry check); only the tensor all-to-all is replaced by
-the object exchange on gloo.
+These exercise the same code paths the RCCL/GPU path uses (phase
+schedule, registry check, and — since gloo supports alltoall — the
+tensor all-to-allv bucket exchange itself); only the transport differs.
+Object/string-column pipelines fall back to the object exchange.
 """
 
 import os
