@@ -334,3 +334,29 @@ def test_dist_nshard_less_than_world():
     results = _run_workers(_narrow_phase_worker, world=3)
     assert results[0] == [(1, [10, 30], []), (2, [20], [5]),
                           (3, [], [6])]
+
+
+def _float_worker(rank, world, port, q):
+    """Mixed-dtype (int64 keys, float32 values) through the tensor
+    exchange: per-column all_to_all_single must handle heterogeneous
+    column dtypes."""
+    _init(rank, world, port)
+    import bigslice_amd as bs
+
+    def build(m):
+        keys = torch.arange(999, dtype=torch.int64) % 13
+        vals = keys.to(torch.float32) + 0.5
+        return bs.Reduce(bs.Const(m, keys, vals), "sum")
+
+    fv = bs.func(build)
+    sess = bs.start(distributed=True, device="cpu")
+    q.put((rank, dict(sess.run(fv, 4).scan())))
+
+
+def test_dist_mixed_dtype_exchange():
+    results = _run_workers(_float_worker)
+    keys = torch.arange(999, dtype=torch.int64) % 13
+    vals = keys.to(torch.float32) + 0.5
+    for k in range(13):
+        expect = float(vals[keys == k].sum())
+        assert abs(results[0][k] - expect) < 1e-2
