@@ -68,12 +68,30 @@ def _free_port():
         return s.getsockname()[1]
 
 
+def _worker_entry(fn, rank, world, port, q):
+    """Run a dist worker, then tear the process group down cleanly:
+    ranks exiting at different times with a live gloo group can
+    SIGABRT in its helper threads (same race as bench.py's teardown),
+    tripping the exitcode assertion below."""
+    try:
+        fn(rank, world, port, q)
+    finally:
+        import torch.distributed as dist
+        if dist.is_initialized():
+            try:
+                dist.barrier()
+            except Exception:
+                pass
+            dist.destroy_process_group()
+
+
 def _run_workers(fn, world=2, port=None):
     if port is None:
         port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=fn, args=(r, world, port, q))
+    procs = [ctx.Process(target=_worker_entry,
+                         args=(fn, r, world, port, q))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -197,7 +215,8 @@ def test_dist_error_propagates_to_all_ranks():
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     port = _free_port()
-    procs = [ctx.Process(target=_error_worker, args=(r, 2, port, q))
+    procs = [ctx.Process(target=_worker_entry,
+                         args=(_error_worker, r, 2, port, q))
              for r in range(2)]
     for p in procs:
         p.start()
