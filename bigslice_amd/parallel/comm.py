@@ -85,6 +85,15 @@ class Comm:
         dist.all_gather_object(out, obj)
         return out
 
+    def broadcast_obj(self, obj, src: int = 0):
+        """Broadcast one picklable object from src (CompileEnv
+        shipping — the Worker.Compile analog)."""
+        if self.world == 1:
+            return obj
+        box = [obj if self.rank == src else None]
+        dist.broadcast_object_list(box, src=src)
+        return box[0]
+
     def any_flag(self, flag: bool) -> bool:
         """Cheap collective OR (one small all_reduce — used as the
         per-phase error check so the common path avoids object

@@ -28,14 +28,29 @@ class CompileEnv:
     def __init__(self):
         self.cached: Dict[Tuple[str, int], bool] = {}
         self.frozen = False
+        self.sealed = False
 
     def freeze(self):
         self.frozen = True
+
+    def seal(self, decisions: Dict[Tuple[str, int], bool]):
+        """Install driver-broadcast decisions and forbid local probing.
+        In SPMD mode only rank 0 stats the filesystem; other ranks
+        compile against the sealed env so a racing cache write cannot
+        produce divergent graphs (exec/compile.go:125-184)."""
+        self.cached = dict(decisions)
+        self.sealed = True
 
     def cache_decision(self, key: str, shard: int,
                        compute: Callable[[], List[bool]]) -> bool:
         k = (key, shard)
         if k not in self.cached:
+            if self.sealed:
+                raise RuntimeError(
+                    f"cache decision for {k} missing from the "
+                    "driver-broadcast CompileEnv: this rank compiled a "
+                    "different graph than rank 0 (non-deterministic "
+                    "Func?)")
             if self.frozen:
                 return False
             decisions = compute()

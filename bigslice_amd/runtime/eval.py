@@ -234,7 +234,12 @@ def evaluate(executor: Executor, roots: Sequence[Task]) -> None:
                 if id(t) in root_ids and roots_left <= 0:
                     # notify outside the freshly-counted check too: a
                     # root re-finishing after an external LOST/re-run
-                    # may still be in ok_counted
+                    # may still be in ok_counted.  A root flipped to OK
+                    # entirely outside this evaluator's callbacks (a
+                    # concurrent evaluate sharing the task) is only
+                    # picked up by the 1s timed-wait backstop below —
+                    # up to ~1s extra latency on that rare path, traded
+                    # for keeping completion notification lock-free.
                     done.notify_all()
             elif st == TaskState.LOST:
                 # resubmit, re-walking deps that were also lost:

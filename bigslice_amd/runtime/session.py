@@ -234,13 +234,23 @@ class Session:
             slice_ = funcv.invoke(args)
             # fresh CompileEnv per run: cache decisions are re-checked
             # each invocation (exec/compile.go recomputes presence) and
-            # concurrent runs must not share frozen state.  NOTE for
-            # distributed mode: every rank computes decisions against
-            # the same filesystem; a rank-divergent cache view would
-            # need the reference's env-broadcast (tracked in NOTES.md).
+            # concurrent runs must not share frozen state.  In
+            # distributed mode rank 0 computes all cache decisions and
+            # broadcasts the frozen env; other ranks compile against it
+            # without touching the filesystem (exec/compile.go:125-184)
+            # so a racing cache write cannot yield divergent graphs.
             env = CompileEnv()
             compiler = Compiler(inv_index, env)
-            tasks = compiler.compile(slice_)
+            comm = getattr(self.executor, "comm", None)
+            if comm is not None and comm.world > 1:
+                if comm.rank == 0:
+                    tasks = compiler.compile(slice_)
+                    comm.broadcast_obj(env.cached, src=0)
+                else:
+                    env.seal(comm.broadcast_obj(None, src=0))
+                    tasks = compiler.compile(slice_)
+            else:
+                tasks = compiler.compile(slice_)
             if funcv.exclusive:
                 # Exclusive Funcs (func.go Exclusive): their tasks do
                 # not share the executor with others (locally: take all

@@ -84,7 +84,11 @@ class Task:
         # condition ordering matters only for wait_state, which takes
         # the lock).  State reads are the evaluator's hottest call
         # (~3 per task per schedule pass); the lock here contended
-        # directly with every set_state notify.
+        # directly with every set_state notify.  Caveat: on a
+        # free-threaded (PEP 703, no-GIL) build the err-before-state
+        # store ordering below is not a synchronization edge; if this
+        # ever runs under such a build, read state/error via the cond
+        # lock instead.
         return self._state
 
     @property

@@ -111,6 +111,13 @@ def oom_stress(size: int, device: str):
     session must surface it as an ordinary task error naming the
     allocation, not hang or die."""
 
+    if size < (1 << 45):
+        raise SystemExit(
+            f"--size {size} is small enough that the allocation could "
+            "succeed (and the UDF would then fail with a misleading "
+            "'not reached'); use >= 2**45 bytes for a guaranteed "
+            "immediate allocator rejection")
+
     def build(nbytes):
         src = bs.Const(2, torch.arange(4, dtype=torch.int64))
 

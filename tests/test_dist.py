@@ -355,6 +355,62 @@ def test_dist_nshard_less_than_world():
                           (3, [], [6])]
 
 
+def _compile_env_worker(rank, world, port, q):
+    """Driver-broadcast CompileEnv (exec/compile.go:125-184): only rank
+    0 may stat the filesystem for cache decisions; other ranks compile
+    against the sealed broadcast env.  Non-zero ranks get a poisoned
+    ShardCache.present to prove they never probe locally."""
+    import shutil
+    _init(rank, world, port)
+    import bigslice_amd as bs
+    from bigslice_amd.ops import cache as cache_mod
+
+    prefix = "/tmp/bigslice_envtest/c"
+    if rank == 0:
+        shutil.rmtree("/tmp/bigslice_envtest", ignore_errors=True)
+    else:
+        # a non-zero rank consulting the filesystem is the divergent-
+        # graph bug this feature exists to prevent
+        def poisoned(self):
+            raise AssertionError(
+                "rank != 0 computed cache decisions locally")
+        cache_mod.ShardCache.present = poisoned
+
+    computed = {"n": 0}
+
+    def build(m):
+        def gen(shard, ctx):
+            computed["n"] += 1
+            keys = torch.arange(40, dtype=torch.int64) % 4
+            yield (keys, torch.ones_like(keys))
+        src = bs.ReaderFunc(m, gen, bs.schema_of(int, int))
+        return bs.Reduce(bs.Cache(src, prefix), "sum")
+
+    fv = bs.func(build)
+    sess = bs.start(distributed=True, device="cpu")
+    # barrier so rank 0's rmtree lands before anyone compiles
+    import torch.distributed as dist
+    dist.barrier()
+    r1 = sess.run(fv, 4)
+    rows1 = sorted(r1.scan())
+    n_first = computed["n"]
+    dist.barrier()  # all cache writes visible before the second compile
+    r2 = sess.run(fv, 4)
+    rows2 = sorted(r2.scan())
+    q.put((rank, (rows1, rows2, n_first, computed["n"])))
+
+
+def test_dist_compile_env_broadcast():
+    results = _run_workers(_compile_env_worker)
+    expect = sorted((k, 40) for k in range(4))
+    rows1, rows2, n1, n2 = results[0]
+    assert rows1 == expect
+    assert rows2 == expect
+    assert n2 == n1  # second run read caches, no shard recompute
+    # rank 1 also recomputed nothing the second time
+    assert results[1][3] == results[1][2]
+
+
 def _float_worker(rank, world, port, q):
     """Mixed-dtype (int64 keys, float32 values) through the tensor
     exchange: per-column all_to_all_single must handle heterogeneous
