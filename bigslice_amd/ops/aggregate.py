@@ -64,6 +64,13 @@ class Aggregation:
         return tuple(a if isinstance(a, str) else id(a)
                      for a in self.aggs)
 
+    @property
+    def name(self) -> str:
+        """Stable human label (golden graphs, traces)."""
+        return "+".join(
+            a if isinstance(a, str) else getattr(a, "__name__", "fn")
+            for a in self.aggs)
+
     def __repr__(self):
         return f"Aggregation({self.aggs})"
 

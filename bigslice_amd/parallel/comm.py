@@ -106,16 +106,32 @@ class Comm:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         return bool(t.cpu().item())
 
-    def check_registry(self, digest: str):
+    def check_registry(self, digest: str, locations=None):
         """Cross-process Func-registry verification (the FuncLocations
-        diff, exec/slicemachine.go:689-702): all ranks must have built
-        identical registries or graphs diverge silently."""
+        diff, exec/slicemachine.go:689-702 + func.go:276-343): all
+        ranks must have built identical registries or graphs diverge
+        silently.  The common path compares digests only; on mismatch
+        the full location lists gather and the error carries a unified
+        location-level diff naming the Funcs that diverged."""
         digests = self.all_gather_obj(digest)
-        if any(d != digest for d in digests):
-            raise RuntimeError(
-                f"Func registry mismatch across ranks: {digests}; "
-                "register all Funcs at module import time in the same "
-                "order on every rank")
+        if not any(d != digest for d in digests):
+            return
+        if locations is None:
+            from ..runtime.session import func_locations
+            locations = func_locations()
+        all_locs = self.all_gather_obj(list(locations))
+        bad = next(r for r, d in enumerate(digests) if d != digests[0])
+        import difflib
+        diff = "\n".join(difflib.unified_diff(
+            all_locs[0], all_locs[bad],
+            fromfile="rank 0 Func registry",
+            tofile=f"rank {bad} Func registry", lineterm=""))
+        raise RuntimeError(
+            "Func registry mismatch across ranks; register all Funcs "
+            "at module import time in the same order on every rank.\n"
+            + (diff or f"(digests differ but locations match: "
+                       f"{digests} — location strings identical, "
+                       f"ordering or count diverged)"))
 
     # -- shuffle exchange -------------------------------------------------
 

@@ -128,17 +128,34 @@ def all_deps_ok(task: Task) -> bool:
                for dep in task.deps for t in dep.head_tasks)
 
 
-def graph_string(roots: Sequence[Task]) -> str:
+def graph_string(roots: Sequence[Task], detail: bool = False) -> str:
     """Debug rendering of the task graph (task.go GraphString), used by
-    compiler golden tests."""
+    compiler golden tests.  detail=True annotates each first visit with
+    the shuffle/partition/combiner attributes so golden diffs catch
+    fusion, naming, partitioning and combine-spec regressions
+    (exec/compile_test.go:23-137)."""
     lines: List[str] = []
     seen = set()
 
+    def annot(t: Task) -> str:
+        if not detail:
+            return ""
+        bits = []
+        if t.shuffle_out:
+            bits.append(f"shuffle nparts={t.num_partitions}")
+        if t.combiner is not None:
+            bits.append(f"comb={getattr(t.combiner, 'name', '?')}")
+        if t.pragma is not None and getattr(t.pragma, "materialize",
+                                            False):
+            bits.append("materialize")
+        return f"  [{' '.join(bits)}]" if bits else ""
+
     def visit(t: Task, depth: int):
-        lines.append("  " * depth + t.name)
         if t.name in seen:
+            lines.append("  " * depth + t.name)
             return
         seen.add(t.name)
+        lines.append("  " * depth + t.name + annot(t))
         for dep in t.deps:
             for h in dep.head_tasks:
                 visit(h, depth + 1)

@@ -411,6 +411,38 @@ def test_dist_compile_env_broadcast():
     assert results[1][3] == results[1][2]
 
 
+def _registry_diff_worker(rank, world, port, q):
+    """A divergent Func registry must fail with a location-level diff
+    naming the extra Func (func.go:276-343), not just digests."""
+    _init(rank, world, port)
+    import bigslice_amd as bs
+    from bigslice_amd.runtime.session import registry_digest
+
+    bs.func(lambda: bs.Const(1, torch.arange(2, dtype=torch.int64)))
+    if rank == 1:  # the divergence
+        extra = bs.func(
+            lambda: bs.Const(1, torch.arange(3, dtype=torch.int64)))
+        assert extra is not None
+    from bigslice_amd.parallel.comm import Comm
+    comm = Comm(rank, world, "cpu")
+    try:
+        comm.check_registry(registry_digest())
+        q.put((rank, None))
+    except RuntimeError as e:
+        q.put((rank, str(e)))
+
+
+def test_dist_registry_mismatch_location_diff():
+    results = _run_workers(_registry_diff_worker)
+    for rank in (0, 1):
+        msg = results[rank]
+        assert msg is not None, "mismatch not detected"
+        # the unified diff marks rank 1's extra registration line
+        assert "rank 1 Func registry" in msg
+        assert any(line.startswith("+") and "test_dist" in line
+                   for line in msg.splitlines()), msg
+
+
 def _float_worker(rank, world, port, q):
     """Mixed-dtype (int64 keys, float32 values) through the tensor
     exchange: per-column all_to_all_single must handle heterogeneous
