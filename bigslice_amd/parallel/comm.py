@@ -94,6 +94,23 @@ class Comm:
         dist.broadcast_object_list(box, src=src)
         return box[0]
 
+    def _reduce(self, t: torch.Tensor, op) -> torch.Tensor:
+        if self.world == 1:
+            return t
+        dev = t
+        if self.backend == "nccl":
+            dev = t.to(self.device)
+        dist.all_reduce(dev, op=op)
+        return dev.cpu() if dev is not t else t
+
+    def min_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        """Elementwise MIN across ranks (checkpoint-owner discovery)."""
+        return self._reduce(t.clone(), dist.ReduceOp.MIN)
+
+    def sum_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        """Elementwise SUM across ranks (global partition counts)."""
+        return self._reduce(t.clone(), dist.ReduceOp.SUM)
+
     def any_flag(self, flag: bool) -> bool:
         """Cheap collective OR (one small all_reduce — used as the
         per-phase error check so the common path avoids object

@@ -4,27 +4,36 @@ Role-parity with the reference's bigmachine executor (exec/bigmachine.go),
 redesigned for the MI355X execution model: instead of a driver pushing
 tasks to workers over RPC and workers pulling shuffle data through a
 storage layer, every rank compiles the identical task graph (the CompileEnv
-/ Func-registry discipline, verified by digest at start) and executes it
-phase-synchronously: each rank runs its own shards of a phase, partitions
-output with the fused K4 kernel (pre-combining when the consumer declares a
-combiner), and the phase boundary is ONE RCCL all-to-allv over xGMI.
+broadcast / Func-registry discipline) and executes it phase-synchronously.
+Each rank runs its shards of a phase, partitions output with the fused K4
+kernel (pre-combining when the consumer declares a combiner), and streams
+the phase boundary through a **windowed RCCL all-to-allv** overlapped with
+producer compute (parallel/exchange.py — the consumer-pull streaming data
+plane of exec/bigmachine.go:822-908, recast as collectives).
 
-Shard placement: shard s of every phase runs on rank s % world, and
-partition p of any shuffle output is owned by rank p % world — placement
-is static and known to all ranks, so there is no location metadata plane.
+Placement: partitions are assigned to ranks by a deterministic LPT packing
+of the first window's global partition sizes (the load-aware scheduling of
+exec/slicemachine.go:444-603 adapted to SPMD), falling back to
+``shard % world`` for phases with no counts yet.  Producer groups feeding
+one consumer (e.g. both cogroup inputs) share an *exchange domain* so
+partition p of every input lands on the same rank.
 """
 
 from __future__ import annotations
 
 from typing import Dict, List, Optional, Sequence, Tuple
 
+import torch
 
 from ..ops.slice_base import TaskContext
 from ..parallel.comm import Comm
+from ..schema import is_object
+from ..parallel.exchange import PARTIAL_COMBINE_ROWS, PhaseExchange, \
+    lpt_assign
 from ..utils import metrics
 from ..sliceio import MultiReader, Reader
 from .eval import Executor
-from .partition import PartitionWriter
+from .partition import PartitionWriter, split_frame
 from .store import MemoryStore
 from .task import Task, TaskState
 
@@ -51,14 +60,24 @@ class DistExecutor(Executor):
         self.device = comm.device
         self.scopes = {}
         self.tracer = None
+        # partition->rank maps keyed by producer group name; every
+        # group in an exchange domain records the shared map
+        self.placement: Dict[str, List[int]] = {}
+        # task name -> rank that ran (or checkpoint-holds) it
+        self._task_owner: Dict[str, int] = {}
+        self._group_domain: Dict[str, str] = {}
+        self._domain_members: Dict[str, List[str]] = {}
 
     # The session calls executor.evaluate when present (instead of the
     # generic task-pull evaluator).
     def evaluate(self, roots: Sequence[Task]) -> None:
+        self._build_domains(roots)
         phases = self._phase_order(roots)
         for phase in phases:
             self._run_phase(phase)
         self.comm.barrier()
+
+    # -- graph walks ------------------------------------------------------
 
     def _phase_order(self, roots: Sequence[Task]) -> List[List[Task]]:
         """Topological order of phase groups (deterministic across
@@ -81,6 +100,136 @@ class DistExecutor(Executor):
             visit(r)
         return order
 
+    def _build_domains(self, roots: Sequence[Task]) -> None:
+        """Union producer groups that feed a common consumer into one
+        exchange domain (domain key = min member group name), so e.g.
+        both cogroup inputs use one partition placement."""
+        parent: Dict[str, str] = {}
+
+        def find(x: str) -> str:
+            while parent.setdefault(x, x) != x:
+                parent[x] = parent[parent[x]]
+                x = parent[x]
+            return x
+
+        def union(a: str, b: str) -> None:
+            ra, rb = find(a), find(b)
+            if ra != rb:
+                # deterministic: smaller name becomes the root
+                lo, hi = sorted((ra, rb))
+                parent[hi] = lo
+
+        seen = set()
+
+        def visit(t: Task):
+            if id(t) in seen:
+                return
+            seen.add(id(t))
+            heads = [dep.head_tasks[0].group[0].name
+                     for dep in t.deps
+                     if dep.head_tasks and dep.head_tasks[0].shuffle_out]
+            for h in heads[1:]:
+                union(heads[0], h)
+            for h in heads:
+                find(h)
+            for dep in t.deps:
+                for h in dep.head_tasks:
+                    visit(h)
+
+        for r in roots:
+            visit(r)
+        members: Dict[str, List[str]] = {}
+        for g in parent:
+            members.setdefault(find(g), []).append(g)
+        for dom, gs in members.items():
+            gs.sort()
+            for g in gs:
+                self._group_domain[g] = dom
+            self._domain_members[dom] = gs
+
+    def _owner(self, t: Task) -> int:
+        """Which rank runs task t.  Consumer tasks follow their shuffle
+        dep's placement (data locality); non-shuffle chains follow
+        their dep's owner; sources use shard % world."""
+        got = self._task_owner.get(t.name)
+        if got is not None:
+            return got
+        for dep in t.deps:
+            if dep.head_tasks and dep.head_tasks[0].shuffle_out:
+                place = self.placement.get(
+                    dep.head_tasks[0].group[0].name)
+                if place is not None and t.shard < len(place):
+                    return place[t.shard]
+                return owner_rank(t.shard, self.comm.world)
+        if t.deps and t.deps[0].head_tasks:
+            return self._owner(t.deps[0].head_tasks[0])
+        return owner_rank(t.shard, self.comm.world)
+
+    def _placement_resolver(self, phase: List[Task]):
+        """Returns resolve(sizes) for PhaseExchange: an existing
+        placement for this phase's exchange domain, or — when sizes are
+        provided — a fresh LPT assignment recorded for every domain
+        member."""
+        key = phase[0].name
+        dom = self._group_domain.get(key, key)
+        mem = self._domain_members.get(dom, [key])
+
+        def resolve(sizes: Optional[List[int]]) -> Optional[List[int]]:
+            for m in mem:
+                if m in self.placement:
+                    place = self.placement[m]
+                    for mm in mem:
+                        self.placement.setdefault(mm, place)
+                    return place
+            if sizes is None:
+                return None
+            place = lpt_assign(sizes, self.comm.world)
+            for m in mem:
+                self.placement[m] = place
+            return place
+
+        return resolve
+
+    # -- checkpoint skip --------------------------------------------------
+
+    def _checkpoint_skip(self, phase: List[Task]) -> bool:
+        """Collective: skip the phase when every output partition is in
+        some rank's (persistent) store, discovering the placement those
+        outputs were stored under.  Gated on a persistent store so
+        fresh MemoryStore runs pay no per-phase collective."""
+        if not getattr(self.store, "persistent", False):
+            return False
+        comm, world = self.comm, self.comm.world
+        exemplar = phase[0]
+        if exemplar.shuffle_out:
+            nparts = exemplar.num_partitions
+            local = torch.full((nparts,), world, dtype=torch.int64)
+            for p in range(nparts):
+                if self.store.has(exemplar.name, p) or any(
+                        self.store.has(t.name, p) for t in phase):
+                    local[p] = comm.rank
+            owner = comm.min_reduce(local)
+            if bool((owner >= world).any()):
+                return False
+            place = [int(r) for r in owner]
+            key = phase[0].name
+            dom = self._group_domain.get(key, key)
+            for m in self._domain_members.get(dom, [key]):
+                self.placement.setdefault(m, place)
+            return True
+        local = torch.full((len(phase),), world, dtype=torch.int64)
+        for i, t in enumerate(phase):
+            if self.store.has(t.name, 0):
+                local[i] = comm.rank
+        owner = comm.min_reduce(local)
+        if bool((owner >= world).any()):
+            return False
+        for i, t in enumerate(phase):
+            self._task_owner[t.name] = int(owner[i])
+        return True
+
+    # -- phase execution --------------------------------------------------
+
     def _run_phase(self, phase: List[Task]) -> None:
         comm = self.comm
         world = comm.world
@@ -88,119 +237,202 @@ class DistExecutor(Executor):
         # or whose outputs survive in a persistent store (restart).
         if all(t.state == TaskState.OK for t in phase):
             return
-        # The checkpoint decision must be COLLECTIVE: a rank that owns
-        # no shards and no partitions of this phase (nshard < world) is
-        # vacuously "complete" locally, and skipping on that alone
-        # deadlocks the other ranks' phase collectives.  One tiny
-        # all_reduce: skip only if NO rank is missing outputs.
-        if not comm.any_flag(not self._phase_checkpointed(phase)):
+        if self._checkpoint_skip(phase):
             for t in phase:
                 t.set_state(TaskState.OK)
             return
+        for t in phase:
+            self._task_owner[t.name] = self._owner(t)
         my_tasks = [t for t in phase
-                    if owner_rank(t.shard, world) == comm.rank]
+                    if self._task_owner[t.name] == comm.rank]
         exemplar = phase[0]
-        shuffled = exemplar.shuffle_out
-        send: List[List[Tuple[str, int, object]]] = \
-            [[] for _ in range(world)]
+        if not exemplar.shuffle_out or world == 1:
+            self._run_phase_simple(phase, my_tasks)
+        elif comm.tensor_exchange_ok and not any(
+                is_object(dt) for dt in exemplar.schema.dtypes):
+            self._run_phase_windowed(phase, my_tasks)
+        else:
+            self._run_phase_single(phase, my_tasks)
+        for t in phase:
+            if t.state != TaskState.ERR:
+                t.set_state(TaskState.OK)
+
+    def _raise_collective(self, phase, err: Optional[BaseException],
+                          any_err: bool) -> None:
+        """Surface an error on every rank; discard the phase's partial
+        outputs so a restart cannot mix stale and fresh partitions."""
+        if not any_err:
+            return
+        errs = self.comm.all_gather_obj(repr(err) if err else None)
+        for t in phase:
+            self.store.discard_task(t.name)
+        e = err or RuntimeError(
+            "remote rank failed: "
+            f"{next((x for x in errs if x), '?')}")
+        for t in phase:
+            t.set_state(TaskState.ERR, e)
+        raise e
+
+    def _run_phase_simple(self, phase: List[Task], my_tasks: List[Task]
+                          ) -> None:
+        """Non-shuffle phases (store locally) and the world==1
+        fast path (buckets pass straight through to the store)."""
         err: Optional[BaseException] = None
-        # Machine-combiners mode (exec/session.go:166-176): this rank's
-        # producer tasks of the phase share one combiner/partitioner.
-        shared_writer = None
-        if (shuffled and len(my_tasks) > 1
-                and exemplar.combiner is not None
-                and exemplar.partitioner is None):
-            from ..ops.slice_base import TaskContext
-            shared_writer = PartitionWriter(
-                exemplar.num_partitions, None, exemplar.combiner,
-                exemplar.schema, self.device,
-                TaskContext(device=self.device).chunk)
+        shuffled = phase[0].shuffle_out
         try:
             for t in my_tasks:
-                buckets = self._run_task(t, shared_writer)
+                buckets = self._run_task(t)
                 if buckets is None:
                     continue
                 for p, frames in enumerate(buckets):
-                    if shuffled:
-                        d = owner_rank(p, world)
-                        for f in frames:
-                            send[d].append((t.name, p, f))
-                    else:
-                        rows = sum(len(f) for f in frames)
-                        self.store.put(t.name, p, frames, rows)
-            if shared_writer is not None and my_tasks:
-                first = my_tasks[0]
-                for p, frames in enumerate(shared_writer.finish()):
-                    d = owner_rank(p, world)
-                    for f in frames:
-                        send[d].append((first.name, p, f))
+                    rows = sum(len(f) for f in frames)
+                    self.store.put(t.name, p, frames, rows)
+            if shuffled:
+                # mark empty partitions so checkpoint discovery sees a
+                # complete phase
+                for t in phase:
+                    for p in range(t.num_partitions):
+                        if not self.store.has(t.name, p):
+                            self.store.put(t.name, p, [], 0)
         except BaseException as e:
             err = e
-        # Surface errors collectively so every rank raises; the common
-        # path costs one tiny all_reduce, details gather only on error.
-        if comm.any_flag(err is not None):
-            errs = comm.all_gather_obj(repr(err) if err else None)
-            e = err or RuntimeError(
-                "remote rank failed: "
-                f"{next((x for x in errs if x), '?')}")
-            for t in phase:
-                t.set_state(TaskState.ERR, e)
-            raise e
+        self._raise_collective(phase, err,
+                               self.comm.any_flag(err is not None))
 
-        if shuffled:
-            index_name = [t.name for t in phase]
-            name_index = {n: i for i, n in enumerate(index_name)}
-            recv = comm.exchange_buckets(send, exemplar.schema,
-                                         name_index, index_name)
-            # group received frames by (task, partition)
-            grouped: Dict[Tuple[str, int], List] = {}
-            for (tname, p, f) in recv:
-                grouped.setdefault((tname, p), []).append(f)
-            for (tname, p), frames in grouped.items():
-                if self.device != "cpu":
-                    frames = [f.to(self.device) for f in frames]
-                rows = sum(len(f) for f in frames)
-                self.store.put(tname, p, frames, rows)
-            # mark empty partitions we own so readers don't KeyError
-            for t in phase:
-                for p in range(t.num_partitions):
-                    if owner_rank(p, world) == comm.rank and \
-                            not self.store.has(t.name, p):
-                        self.store.put(t.name, p, [], 0)
-        for t in phase:
-            t.set_state(TaskState.OK)
-
-    def _phase_checkpointed(self, phase: List[Task]) -> bool:
-        """True when every output partition this rank owns is already
-        in the store (a previous job run completed the phase)."""
-        world = self.comm.world
-        rank = self.comm.rank
+    def _run_phase_single(self, phase: List[Task], my_tasks: List[Task]
+                          ) -> None:
+        """Fallback shuffle path (object columns / backends without
+        alltoall): compute everything, one placement all_reduce, one
+        object exchange."""
+        comm, world = self.comm, self.comm.world
         exemplar = phase[0]
-        if exemplar.shuffle_out:
-            for t in phase:
-                for p in range(t.num_partitions):
-                    if owner_rank(p, world) == rank and \
-                            not self.store.has(t.name, p):
-                        return False
-            return True
-        for t in phase:
-            if owner_rank(t.shard, world) == rank and \
-                    not self.store.has(t.name, 0):
-                return False
-        return True
+        nparts = exemplar.num_partitions
+        err: Optional[BaseException] = None
+        buckets_out: List[List] = [[] for _ in range(nparts)]
+        try:
+            for t in my_tasks:
+                buckets = self._run_task(t)
+                if buckets is None:
+                    continue
+                for p, frames in enumerate(buckets):
+                    buckets_out[p].extend(frames)
+        except BaseException as e:
+            err = e
+        self._raise_collective(phase, err,
+                               comm.any_flag(err is not None))
+        # placement from full global counts (exact LPT)
+        local = torch.tensor(
+            [sum(len(f) for f in fl) for fl in buckets_out],
+            dtype=torch.int64)
+        resolve = self._placement_resolver(phase)
+        place = resolve(None)
+        if place is None:
+            place = resolve(comm.sum_reduce(local).tolist())
+        send: List[List[Tuple[str, int, object]]] = \
+            [[] for _ in range(world)]
+        for p, frames in enumerate(buckets_out):
+            for f in frames:
+                send[place[p]].append((exemplar.name, p, f))
+        recv = comm.exchange_buckets(send, exemplar.schema)
+        grouped: Dict[int, List] = {}
+        for (_, p, f) in recv:
+            grouped.setdefault(p, []).append(f)
+        self._store_received(exemplar, grouped, place)
 
-    def _run_task(self, task: Task, shared_writer=None):
-        """Run one task; returns per-partition frame lists (or None for
-        terminal tasks or when a shared phase writer absorbs output)."""
-        task.set_state(TaskState.RUNNING)
-        scope = metrics.Scope()
-        self.scopes[task.name] = scope
-        self._scope_ctx = metrics.scoped(scope)
-        self._scope_ctx.__enter__()
-        if self.tracer:
-            self._span = self.tracer.span(task.name, pid=self.comm.rank)
-            self._span.__enter__()
-        ctx = TaskContext(device=self.device)
+    def _run_phase_windowed(self, phase: List[Task],
+                            my_tasks: List[Task]) -> None:
+        """The hot path: windowed all-to-allv overlapped with producer
+        compute; consumer-side streaming combine of received windows."""
+        comm = self.comm
+        exemplar = phase[0]
+        nparts = exemplar.num_partitions
+        schema = exemplar.schema
+        device = self.device
+        combine_mode = (exemplar.combiner is not None
+                        and exemplar.partitioner is None)
+        recv_aggs: Dict[int, object] = {}
+        recv_frames: Dict[int, List] = {}
+
+        def consume(p: int, frame):
+            if combine_mode:
+                agg = recv_aggs.get(p)
+                if agg is None:
+                    from ..ops.aggregate import make_aggregator
+                    agg = make_aggregator(schema, exemplar.combiner,
+                                          device)
+                    recv_aggs[p] = agg
+                agg.add(frame)
+            else:
+                recv_frames.setdefault(p, []).append(frame)
+
+        ex = PhaseExchange(comm, schema, nparts,
+                           self._placement_resolver(phase), consume)
+        ctx = TaskContext(device=device)
+        err: Optional[BaseException] = None
+        try:
+            if combine_mode:
+                # one shared combine table per GPU for the whole phase
+                # (machine-combiners, exec/session.go:166-176), flushed
+                # as partial combines past the row budget so high-
+                # cardinality phases still stream windows (K11)
+                from ..ops.aggregate import make_aggregator
+                agg = make_aggregator(schema, exemplar.combiner, device)
+                inserted = 0
+
+                def flush_agg(agg):
+                    for rf in agg.result_frames(ctx.chunk * 4):
+                        for p, pf in enumerate(split_frame(rf, nparts,
+                                                           None)):
+                            if pf is not None and len(pf):
+                                ex.add(p, pf)
+
+                for t in my_tasks:
+                    for f in self._task_frames(t, ctx):
+                        agg.add(f)
+                        inserted += len(f)
+                        if inserted >= PARTIAL_COMBINE_ROWS:
+                            flush_agg(agg)
+                            agg = make_aggregator(schema,
+                                                  exemplar.combiner,
+                                                  device)
+                            inserted = 0
+                flush_agg(agg)
+            else:
+                for t in my_tasks:
+                    for f in self._task_frames(t, ctx):
+                        parts = split_frame(f, nparts,
+                                            exemplar.partitioner)
+                        for p, pf in enumerate(parts):
+                            if pf is not None and len(pf):
+                                ex.add(p, pf)
+        except BaseException as e:
+            err = e
+        ex.finish(err=err is not None)
+        if combine_mode:
+            grouped = {p: list(a.result_frames(ctx.chunk * 4))
+                       for p, a in recv_aggs.items()}
+        else:
+            grouped = recv_frames
+        self._store_received(exemplar, grouped, ex.placement)
+        self._raise_collective(phase, err, ex.err_any)
+
+    def _store_received(self, exemplar: Task, grouped: Dict[int, List],
+                        place: List[int]) -> None:
+        """Store received buckets under (exemplar.name, p); mark owned
+        empty partitions so checkpoint discovery sees completeness."""
+        for p, frames in grouped.items():
+            if self.device != "cpu":
+                frames = [f.to(self.device) for f in frames]
+            rows = sum(len(f) for f in frames)
+            self.store.put(exemplar.name, p, frames, rows)
+        for p in range(exemplar.num_partitions):
+            if place[p] == self.comm.rank and \
+                    not self.store.has(exemplar.name, p):
+                self.store.put(exemplar.name, p, [], 0)
+
+    # -- task execution ---------------------------------------------------
+
+    def _open_deps(self, task: Task) -> List:
         dep_readers = []
         for dep in task.deps:
             readers = [self.store.open(h.name, dep.partition,
@@ -211,32 +443,44 @@ class DistExecutor(Executor):
                 dep_readers.append(readers)
             else:
                 dep_readers.append(MultiReader(readers))
-        out = task.do(dep_readers, ctx)
+        return dep_readers
+
+    def _task_frames(self, task: Task, ctx: TaskContext):
+        """Generator over one task's output frames, with scope/tracing
+        bracketing the full consumption (the streaming producer side of
+        the windowed exchange)."""
+        task.set_state(TaskState.RUNNING)
+        scope = metrics.Scope()
+        self.scopes[task.name] = scope
+        span = self.tracer.span(task.name, pid=self.comm.rank) \
+            if self.tracer else None
+        with metrics.scoped(scope):
+            if span:
+                span.__enter__()
+            try:
+                out = task.do(self._open_deps(task), ctx)
+                for f in out:
+                    yield f
+            finally:
+                if span:
+                    span.__exit__(None, None, None)
+
+    def _run_task(self, task: Task) -> Optional[List[List]]:
+        """Run one task to completion; returns per-partition frame
+        lists (or None for terminal tasks, which store their own
+        marker)."""
+        ctx = TaskContext(device=self.device)
         if task.num_out_columns == 0:
-            for _ in out:
+            for _ in self._task_frames(task, ctx):
                 pass
             self.store.put(task.name, 0, [], 0)
-            self._scope_ctx.__exit__(None, None, None)
-            if self.tracer:
-                self._span.__exit__(None, None, None)
-            return None
-        if shared_writer is not None:
-            for f in out:
-                shared_writer.add(f)
-            self._scope_ctx.__exit__(None, None, None)
-            if self.tracer:
-                self._span.__exit__(None, None, None)
             return None
         w = PartitionWriter(task.num_partitions, task.partitioner,
                             task.combiner, task.schema, self.device,
                             ctx.chunk)
-        for f in out:
+        for f in self._task_frames(task, ctx):
             w.add(f)
-        buckets = w.finish()
-        self._scope_ctx.__exit__(None, None, None)
-        if self.tracer:
-            self._span.__exit__(None, None, None)
-        return buckets
+        return w.finish()
 
     # -- Executor interface (driver-side readback) ------------------------
 
@@ -275,10 +519,12 @@ class DistExecutor(Executor):
         return out
 
     def gather_result(self, tasks: Sequence[Task], schema):
-        """Collective: gather all root-task outputs to rank 0."""
+        """Collective: gather all root-task outputs to rank 0.  A
+        partition lives in exactly one rank's store, so presence is
+        the ownership test."""
         frames = []
         for t in tasks:
-            if owner_rank(t.shard, self.comm.world) == self.comm.rank:
-                r = self.reader(t, 0)
+            if self.store.has(t.name, 0):
+                r = self.store.open(t.name, 0, device="cpu")
                 frames.extend(list(r))
         return self.comm.gather_frames(frames, schema)

@@ -25,6 +25,11 @@ from ..sliceio import Reader, codec
 
 
 class Store:
+    # True when contents survive a process restart (FileStore): gates
+    # the distributed executor's per-phase checkpoint-discovery
+    # collective so plain MemoryStore runs stay collective-free.
+    persistent = False
+
     def put(self, task_name: str, partition: int, frames: List[Frame],
             rows: int) -> None:
         raise NotImplementedError
@@ -190,6 +195,8 @@ class FileStore(Store):
     """File-backed store with the reference's layout and count footer
     (exec/store.go:173-261): {prefix}/{task}/p{partition} with 8-byte LE
     record count appended after the encoded stream."""
+
+    persistent = True
 
     def __init__(self, prefix: str):
         self.prefix = prefix
