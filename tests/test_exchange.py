@@ -216,6 +216,30 @@ def test_cogroup_shared_domain_world3():
         assert out[k] == (60, 20), (k, out[k])
 
 
+def test_bench_world8_cpu():
+    """CI form of the scaling bench: torchrun world 8 over gloo runs
+    bench.py end-to-end (the identical code path the driver's 8-GPU
+    RCCL run takes; only the transport differs)."""
+    import json
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, BIGSLICE_EXCHANGE_WINDOW_BYTES="65536")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", str(_free_port()), "bench.py", "--gpus", "8",
+         "--steps", "2", "--warmup", "1", "--rows-per-gpu", "100000",
+         "--nkeys", "3000"],
+        cwd=repo, env=env, capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = next(l for l in out.stdout.splitlines()
+                if l.startswith("{"))
+    parsed = json.loads(line)
+    assert parsed["n_gpus"] == 8
+    assert parsed["config"]["rows_total"] == 800000
+
+
 def _error_mid_window_worker(rank, world, port, q):
     """A UDF that fails after several windows have already exchanged:
     the failing rank must finish the window protocol (empty windows +
