@@ -17,6 +17,7 @@
 #include "hash_partition.hip"
 #include "groupby.hip"
 #include "sort.hip"
+#include "radix.hip"
 #include "vecagg.hip"
 #include "strings.hip"
 
@@ -396,38 +397,239 @@ std::vector<torch::Tensor> groupby_compact(torch::Tensor tkeys,
 
 // ---------------------------------------------------------------- sort
 
-// Reduced sort width: integer keys that are all non-negative order
-// correctly on their value bits alone (the sign-flip bit rocPRIM's key
-// codec adds is constant across them), so one cheap aminmax buys fewer
-// radix passes — the common case of dense ids/categories (20-bit keys:
-// 3 passes instead of 8).  Floats and negative keys sort full-width.
-static int sort_end_bit(const torch::Tensor& keys) {
+
+// ---------------------------------------------------------------------
+// Hand-written LSD radix sort orchestration (kernels in radix.hip):
+// one fused histogram prepass (all digit positions in one read), then
+// one decoupled-lookback scatter kernel per non-constant digit.
+// Set BIGSLICE_SORT_ROCPRIM=1 to force the rocPRIM path (A/B harness).
+
+// Re-read each call so one process can A/B implementations.
+static bool force_rocprim_sort() {
+  const char* e = getenv("BIGSLICE_SORT_ROCPRIM");
+  return e && e[0] == '1';
+}
+
+// Geometry A/B harness: BIGSLICE_RADIX_VARIANT selects (tile, block).
+// Re-read each call so one process can A/B variants back-to-back.
+static int radix_variant() {
+  const char* e = getenv("BIGSLICE_RADIX_VARIANT");
+  return e ? atoi(e) : 0;
+}
+
+template <typename K, int HAS_VAL, int TILE, int BLOCK, int SPLIT,
+          int PERSIST = 0, int NT = 0>
+static void hand_radix_passes(const torch::Tensor& keys,
+                              const torch::Tensor& vals,
+                              torch::Tensor& keys_out,
+                              torch::Tensor& vals_out,
+                              const std::vector<int>& passes,
+                              int npass, uint64_t bias,
+                              hipStream_t stream) {
+  const int64_t n = keys.size(0);
+  const int P = (int)passes.size();
+  const int64_t ntiles = (n + TILE - 1) / TILE;
+  auto state = torch::empty({ntiles * RDX_RADIX},
+                            keys.options().dtype(torch::kInt64));
+  // hist ping-pong + digit_base, all device-resident: after the first
+  // pass's histogram, each scatter tallies the NEXT pass's histogram
+  // itself, so no further prepass and no host syncs between passes.
+  auto i64 = keys.options().dtype(torch::kInt64);
+  auto hist_a = torch::zeros({RDX_RADIX}, i64);
+  auto hist_b = torch::zeros({RDX_RADIX}, i64);
+  auto dbase = torch::empty({RDX_RADIX}, i64);
+  {
+    const int64_t nb = std::min<int64_t>((n + 255) / 256, 4096);
+    hipLaunchKernelGGL((k_radix_hist_one<K>), dim3((int)nb), dim3(256),
+                       0, stream, (const K*)keys.data_ptr(), n,
+                       passes[0] * 8,
+                       (unsigned long long*)hist_a.data_ptr<int64_t>());
+    HIP_CHECK(hipGetLastError());
+  }
+  torch::Tensor keys_tmp, vals_tmp;
+  if (P > 1) {
+    keys_tmp = torch::empty_like(keys);
+    if (HAS_VAL) vals_tmp = torch::empty_like(vals);
+  }
+  const K* src_k = (const K*)keys.data_ptr();
+  const int64_t* src_v =
+      HAS_VAL ? (const int64_t*)vals.data_ptr() : nullptr;
+  torch::Tensor hist_cur = hist_a, hist_next = hist_b;
+  for (int i = 0; i < P; ++i) {
+    const bool to_out = ((P - i) % 2) == 1;
+    K* dst_k = (K*)(to_out ? keys_out : keys_tmp).data_ptr();
+    int64_t* dst_v =
+        HAS_VAL ? (int64_t*)(to_out ? vals_out : vals_tmp).data_ptr()
+                : nullptr;
+    const int xm = (passes[i] == npass - 1) ? 0x80 : 0;
+    hipLaunchKernelGGL(
+        k_radix_scan_hist, dim3(1), dim3(RDX_RADIX), 0, stream,
+        (const unsigned long long*)hist_cur.data_ptr<int64_t>(),
+        (unsigned long long*)dbase.data_ptr<int64_t>(), xm);
+    HIP_CHECK(hipGetLastError());
+    const int next_shift = (i + 1 < P) ? passes[i + 1] * 8 : -1;
+    if (next_shift >= 0)
+      HIP_CHECK(hipMemsetAsync(hist_next.data_ptr(), 0,
+                               RDX_RADIX * 8, stream));
+    HIP_CHECK(hipMemsetAsync(state.data_ptr(), 0,
+                             (size_t)ntiles * RDX_RADIX * 8, stream));
+    if (PERSIST) {
+      const int grid = (int)std::min<int64_t>(ntiles, 256);
+      hipLaunchKernelGGL(
+          (k_radix_scatter_persist<K, HAS_VAL, TILE, BLOCK>),
+          dim3(grid), dim3(BLOCK), 0, stream, src_k, dst_k, src_v,
+          dst_v, n, passes[i] * 8, bias,
+          (const unsigned long long*)dbase.data_ptr<int64_t>(),
+          (unsigned long long*)state.data_ptr<int64_t>(), next_shift,
+          (unsigned long long*)hist_next.data_ptr<int64_t>());
+    } else {
+      hipLaunchKernelGGL(
+          (k_radix_scatter<K, HAS_VAL, TILE, BLOCK, SPLIT, NT>),
+          dim3((int)ntiles),
+          dim3(BLOCK), 0, stream, src_k, dst_k, src_v, dst_v, n,
+          passes[i] * 8, bias,
+          (const unsigned long long*)dbase.data_ptr<int64_t>(),
+          (unsigned long long*)state.data_ptr<int64_t>(), next_shift,
+          (unsigned long long*)hist_next.data_ptr<int64_t>());
+    }
+    HIP_CHECK(hipGetLastError());
+    src_k = dst_k;
+    src_v = dst_v;
+    std::swap(hist_cur, hist_next);
+  }
+}
+
+// pass-skip probe: byte position p is constant across all keys iff the
+// bitwise AND and OR of the keys agree on that byte — one
+// pure-bandwidth reduction replaces a full histogram prepass AND the
+// old aminmax end_bit probe.
+template <typename K>
+static std::vector<int> probe_passes(const torch::Tensor& keys) {
+  const int64_t n = keys.size(0);
+  auto stream = current_stream();
+  const int npass = (int)keys.element_size();
+  auto i64 = keys.options().dtype(torch::kInt64);
+  auto andor = torch::empty({2}, i64);
+  andor[0] = -1;  // all ones
+  andor[1] = 0;
+  const int64_t nb = std::min<int64_t>((n + 255) / 256, 4096);
+  hipLaunchKernelGGL(
+      (k_radix_andor<K>), dim3((int)nb), dim3(256), 0, stream,
+      (const K*)keys.data_ptr(), n,
+      (unsigned long long*)andor.data_ptr<int64_t>(),
+      (unsigned long long*)(andor.data_ptr<int64_t>() + 1));
+  HIP_CHECK(hipGetLastError());
+  auto andor_h = andor.cpu();  // the one host sync
+  const uint64_t band = (uint64_t)andor_h[0].item<int64_t>();
+  const uint64_t bor = (uint64_t)andor_h[1].item<int64_t>();
+  std::vector<int> passes;
+  for (int p = 0; p < npass; ++p) {
+    if (((band >> (p * 8)) & 0xFF) != ((bor >> (p * 8)) & 0xFF))
+      passes.push_back(p);  // constant digit: pass skipped
+  }
+  return passes;
+}
+
+template <typename K, int HAS_VAL>
+static void hand_radix_sort(const torch::Tensor& keys,
+                            const torch::Tensor& vals,
+                            torch::Tensor& keys_out,
+                            torch::Tensor& vals_out,
+                            const std::vector<int>& passes) {
+  auto stream = current_stream();
   const int width = (int)keys.element_size() * 8;
-  if (keys.numel() < (1 << 20)) return width;
-  auto st = keys.scalar_type();
-  if (st != torch::kInt64 && st != torch::kInt32) return width;
-  auto mm = torch::aminmax(keys);
-  const int64_t mn = std::get<0>(mm).item<int64_t>();
-  if (mn < 0) return width;
-  const int64_t mx = std::get<1>(mm).item<int64_t>();
-  int bits = 1;
-  while (bits < width && (mx >> bits) != 0) ++bits;
-  return bits;
+  const int npass = width / 8;
+  const uint64_t bias = 1ull << (width - 1);
+  switch (radix_variant()) {
+    case 1:
+      hand_radix_passes<K, HAS_VAL, 4096, 512, 0>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 2:
+      hand_radix_passes<K, HAS_VAL, 8192, 1024, 0>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 5:
+      hand_radix_passes<K, HAS_VAL, 4096, 512, 1>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 6:
+      hand_radix_passes<K, HAS_VAL, 3584, 512, 1>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 8:
+      hand_radix_passes<K, HAS_VAL, 8192, 1024, 0, 0, 1>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 7:
+      hand_radix_passes<K, HAS_VAL, 8192, 1024, 0, 1>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 0:
+    case 3:
+    default:
+      // default: split reorder, 8K tiles, two workgroups per CU
+      hand_radix_passes<K, HAS_VAL, 8192, 512, 1>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+  }
+}
+
+// Production dispatch: the AND/OR probe picks the executed byte set.
+// A CONTIGUOUS set runs on the in-tree rocPRIM onesweep with the bit
+// range trimmed at both ends (measured fastest on uniform keys); a set
+// with interior constant bytes runs on the hand-written scatter, which
+// skips arbitrary bytes.  BIGSLICE_SORT_HAND=1 forces the hand path,
+// BIGSLICE_SORT_ROCPRIM=1 forces full-control rocPRIM (A/B harness).
+static bool force_hand_sort() {
+  const char* e = getenv("BIGSLICE_SORT_HAND");
+  return e && e[0] == '1';
+}
+
+static bool probed_sortable(const torch::Tensor& keys) {
+  return keys.scalar_type() == torch::kInt64 ||
+         keys.scalar_type() == torch::kInt32;
+}
+
+static bool use_hand_path(const std::vector<int>& passes) {
+  if (force_rocprim_sort()) return false;
+  if (force_hand_sort()) return true;
+  const bool contiguous =
+      (int)passes.size() == passes.back() - passes.front() + 1;
+  return !contiguous;
 }
 
 torch::Tensor radix_sort_keys(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
-  const int end_bit = sort_end_bit(keys);
   auto keys_out = torch::empty_like(keys);
+  int begin_bit = 0, end_bit = (int)keys.element_size() * 8;
+  if (n > 0 && probed_sortable(keys)) {
+    std::vector<int> passes =
+        keys.scalar_type() == torch::kInt64
+            ? probe_passes<int64_t>(keys) : probe_passes<int32_t>(keys);
+    if (passes.empty()) {
+      keys_out.copy_(keys);
+      return keys_out;
+    }
+    if (use_hand_path(passes)) {
+      torch::Tensor nil;
+      if (keys.scalar_type() == torch::kInt64)
+        hand_radix_sort<int64_t, 0>(keys, nil, keys_out, nil, passes);
+      else
+        hand_radix_sort<int32_t, 0>(keys, nil, keys_out, nil, passes);
+      return keys_out;
+    }
+    begin_bit = passes.front() * 8;
+    end_bit = (passes.back() + 1) * 8;
+  }
   auto run = [&](auto fn) {
     size_t temp_bytes = 0;
-    fn(keys.data_ptr(), keys_out.data_ptr(), n, end_bit, nullptr,
-       temp_bytes, current_stream());
+    fn(keys.data_ptr(), keys_out.data_ptr(), n, begin_bit, end_bit,
+       nullptr, temp_bytes, current_stream());
     auto temp = torch::empty({(int64_t)temp_bytes},
                              keys.options().dtype(torch::kUInt8));
-    fn(keys.data_ptr(), keys_out.data_ptr(), n, end_bit, temp.data_ptr(),
-       temp_bytes, current_stream());
+    fn(keys.data_ptr(), keys_out.data_ptr(), n, begin_bit, end_bit,
+       temp.data_ptr(), temp_bytes, current_stream());
   };
   switch (keys.scalar_type()) {
     case torch::kInt64: run(radix_sort_keys_int64_t); break;
@@ -449,19 +651,41 @@ std::vector<torch::Tensor> radix_sort_kv(torch::Tensor keys,
   TORCH_CHECK(vals.is_cuda() && vals.is_contiguous());
   TORCH_CHECK(vals.element_size() == 8, "8-byte values only");
   int64_t n = keys.size(0);
-  const int end_bit = sort_end_bit(keys);
   auto keys_out = torch::empty_like(keys);
   auto vals_out = torch::empty_like(vals);
+  int begin_bit = 0, end_bit = (int)keys.element_size() * 8;
+  if (n > 0 && probed_sortable(keys)) {
+    std::vector<int> passes =
+        keys.scalar_type() == torch::kInt64
+            ? probe_passes<int64_t>(keys) : probe_passes<int32_t>(keys);
+    if (passes.empty()) {
+      keys_out.copy_(keys);
+      vals_out.copy_(vals);
+      return {keys_out, vals_out};
+    }
+    if (use_hand_path(passes)) {
+      if (keys.scalar_type() == torch::kInt64)
+        hand_radix_sort<int64_t, 1>(keys, vals, keys_out, vals_out,
+                                    passes);
+      else
+        hand_radix_sort<int32_t, 1>(keys, vals, keys_out, vals_out,
+                                    passes);
+      return {keys_out, vals_out};
+    }
+    begin_bit = passes.front() * 8;
+    end_bit = (passes.back() + 1) * 8;
+  }
   auto run = [&](auto fn) {
     size_t temp_bytes = 0;
     fn(keys.data_ptr(), keys_out.data_ptr(),
        (const int64_t*)vals.data_ptr(), (int64_t*)vals_out.data_ptr(), n,
-       end_bit, nullptr, temp_bytes, current_stream());
+       begin_bit, end_bit, nullptr, temp_bytes, current_stream());
     auto temp = torch::empty({(int64_t)temp_bytes},
                              keys.options().dtype(torch::kUInt8));
     fn(keys.data_ptr(), keys_out.data_ptr(),
        (const int64_t*)vals.data_ptr(), (int64_t*)vals_out.data_ptr(), n,
-       end_bit, temp.data_ptr(), temp_bytes, current_stream());
+       begin_bit, end_bit, temp.data_ptr(), temp_bytes,
+       current_stream());
   };
   switch (keys.scalar_type()) {
     case torch::kInt64: run(radix_sort_pairs_int64_t); break;
@@ -580,20 +804,38 @@ torch::Tensor colsum16(torch::Tensor x, bool mfma) {
 torch::Tensor radix_argsort(torch::Tensor keys) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
-  const int end_bit = sort_end_bit(keys);
   auto perm_in = torch::arange(n, keys.options().dtype(torch::kInt64));
   auto perm_out = torch::empty_like(perm_in);
   auto keys_out = torch::empty_like(keys);
+  int begin_bit = 0, end_bit = (int)keys.element_size() * 8;
+  if (n > 0 && probed_sortable(keys)) {
+    std::vector<int> passes =
+        keys.scalar_type() == torch::kInt64
+            ? probe_passes<int64_t>(keys) : probe_passes<int32_t>(keys);
+    if (passes.empty()) return perm_in;
+    if (use_hand_path(passes)) {
+      if (keys.scalar_type() == torch::kInt64)
+        hand_radix_sort<int64_t, 1>(keys, perm_in, keys_out, perm_out,
+                                    passes);
+      else
+        hand_radix_sort<int32_t, 1>(keys, perm_in, keys_out, perm_out,
+                                    passes);
+      return perm_out;
+    }
+    begin_bit = passes.front() * 8;
+    end_bit = (passes.back() + 1) * 8;
+  }
   auto run = [&](auto fn) {
     size_t temp_bytes = 0;
     fn(keys.data_ptr(), keys_out.data_ptr(),
        perm_in.data_ptr<int64_t>(), perm_out.data_ptr<int64_t>(), n,
-       end_bit, nullptr, temp_bytes, current_stream());
+       begin_bit, end_bit, nullptr, temp_bytes, current_stream());
     auto temp = torch::empty({(int64_t)temp_bytes},
                              keys.options().dtype(torch::kUInt8));
     fn(keys.data_ptr(), keys_out.data_ptr(),
        perm_in.data_ptr<int64_t>(), perm_out.data_ptr<int64_t>(), n,
-       end_bit, temp.data_ptr(), temp_bytes, current_stream());
+       begin_bit, end_bit, temp.data_ptr(), temp_bytes,
+       current_stream());
   };
   switch (keys.scalar_type()) {
     case torch::kInt64: run(radix_sort_pairs_int64_t); break;

@@ -12,11 +12,12 @@
 template <typename K>
 static void radix_sort_pairs_impl(const K* keys_in, K* keys_out,
                                   const int64_t* vals_in, int64_t* vals_out,
-                                  int64_t n, int end_bit, void* temp,
+                                  int64_t n, int begin_bit, int end_bit,
+                                  void* temp,
                                   size_t& temp_bytes, hipStream_t stream) {
   hipError_t err = rocprim::radix_sort_pairs(
-      temp, temp_bytes, keys_in, keys_out, vals_in, vals_out, (size_t)n, 0,
-      end_bit, stream);
+      temp, temp_bytes, keys_in, keys_out, vals_in, vals_out, (size_t)n,
+      begin_bit, end_bit, stream);
   if (err != hipSuccess)
     throw std::runtime_error(std::string("rocprim::radix_sort_pairs: ") +
                              hipGetErrorString(err));
@@ -24,10 +25,10 @@ static void radix_sort_pairs_impl(const K* keys_in, K* keys_out,
 
 template <typename K>
 static void radix_sort_keys_impl(const K* keys_in, K* keys_out, int64_t n,
-                                 int end_bit, void* temp,
+                                 int begin_bit, int end_bit, void* temp,
                                  size_t& temp_bytes, hipStream_t stream) {
   hipError_t err = rocprim::radix_sort_keys(
-      temp, temp_bytes, keys_in, keys_out, (size_t)n, 0, end_bit,
+      temp, temp_bytes, keys_in, keys_out, (size_t)n, begin_bit, end_bit,
       stream);
   if (err != hipSuccess)
     throw std::runtime_error(std::string("rocprim::radix_sort_keys: ") +
@@ -37,11 +38,12 @@ static void radix_sort_keys_impl(const K* keys_in, K* keys_out, int64_t n,
 // Explicit type dispatch used by ext.hip.
 #define INSTANTIATE_SORT(K)                                               \
   void radix_sort_pairs_##K(const void* ki, void* ko, const int64_t* vi,  \
-                            int64_t* vo, int64_t n, int end_bit,          \
+                            int64_t* vo, int64_t n, int begin_bit,        \
+                            int end_bit,                                  \
                             void* temp, size_t& temp_bytes,               \
                             hipStream_t s) {                              \
-    radix_sort_pairs_impl<K>((const K*)ki, (K*)ko, vi, vo, n, end_bit,    \
-                             temp, temp_bytes, s);                        \
+    radix_sort_pairs_impl<K>((const K*)ki, (K*)ko, vi, vo, n, begin_bit,  \
+                             end_bit, temp, temp_bytes, s);               \
   }
 
 INSTANTIATE_SORT(int64_t)
@@ -51,10 +53,11 @@ INSTANTIATE_SORT(double)
 
 #define INSTANTIATE_SORT_KEYS(K)                                          \
   void radix_sort_keys_##K(const void* ki, void* ko, int64_t n,           \
-                           int end_bit, void* temp, size_t& temp_bytes,   \
+                           int begin_bit, int end_bit, void* temp,        \
+                           size_t& temp_bytes,                            \
                            hipStream_t s) {                               \
-    radix_sort_keys_impl<K>((const K*)ki, (K*)ko, n, end_bit, temp,       \
-                            temp_bytes, s);                               \
+    radix_sort_keys_impl<K>((const K*)ki, (K*)ko, n, begin_bit, end_bit,  \
+                            temp, temp_bytes, s);                         \
   }
 
 INSTANTIATE_SORT_KEYS(int64_t)

@@ -101,6 +101,34 @@ def test_radix_argsort_matches_torch(kernels):
         assert torch.equal(k[perm], k.sort().values)
 
 
+def test_hand_radix_sort_edges(kernels):
+    """Hand-written LSD radix sort (csrc/radix.hip): stability,
+    negatives, tail tiles, constant digits, int32, vs torch stable
+    sort."""
+    _C = kernels._C
+    g = torch.Generator(device="cuda").manual_seed(11)
+    cases = [
+        torch.randint(-2**62, 2**62, (n,), dtype=torch.int64,
+                      device="cuda", generator=g)
+        for n in (1, 63, 8191, 8192, 8193, 1_000_000)
+    ]
+    cases.append(torch.randint(0, 37, (500_000,), dtype=torch.int64,
+                               device="cuda", generator=g))
+    cases.append(torch.full((10_000,), -5, dtype=torch.int64,
+                            device="cuda"))
+    cases.append(torch.randint(-2**31, 2**31, (300_000,),
+                               dtype=torch.int32, device="cuda",
+                               generator=g))
+    for keys in cases:
+        vals = torch.arange(keys.numel(), dtype=torch.int64,
+                            device="cuda")
+        sk, sv = _C.radix_sort_kv(keys, vals)
+        ref_k, ref_i = torch.sort(keys, stable=True)
+        assert torch.equal(sk, ref_k)
+        assert torch.equal(sv, ref_i)  # stability
+        assert torch.equal(_C.radix_sort_keys(keys), ref_k)
+
+
 def test_hash_partition_large_nparts(kernels):
     from bigslice_amd.frame import Frame
     n, nparts = 300_000, 1024
