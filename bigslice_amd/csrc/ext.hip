@@ -338,6 +338,29 @@ void groupby_insert_packed(torch::Tensor keys, torch::Tensor vals,
   HIP_CHECK(hipGetLastError());
 }
 
+// Replicated-insert experiment (see k_groupby_insert_packed_rep):
+// `table` holds nrep packed tables back to back.
+void groupby_insert_packed_rep(torch::Tensor keys, torch::Tensor vals,
+                               torch::Tensor table, int64_t nrep,
+                               torch::Tensor flags, int64_t max_probes) {
+  TORCH_CHECK(keys.is_cuda() && keys.scalar_type() == torch::kInt64);
+  keys = keys.contiguous();
+  vals = vals.contiguous();
+  int64_t n = keys.size(0);
+  if (n == 0) return;
+  int64_t cap = table.size(0) / (2 * nrep) - 1;
+  TORCH_CHECK(cap > 0 && (cap & (cap - 1)) == 0);
+  const uint32_t seed = 0x9acb0442u;
+  int blocks = (int)std::min<int64_t>((n + THREADS - 1) / THREADS, 32768);
+  hipLaunchKernelGGL(k_groupby_insert_packed_rep, dim3(blocks),
+                     dim3(THREADS), 0, current_stream(),
+                     keys.data_ptr<int64_t>(), vals.data_ptr<int64_t>(), n,
+                     table.data_ptr<int64_t>(), cap, (int32_t)nrep, seed,
+                     flags.data_ptr<int32_t>(),
+                     flags.data_ptr<int32_t>() + 1, max_probes);
+  HIP_CHECK(hipGetLastError());
+}
+
 std::vector<torch::Tensor> groupby_compact_packed(torch::Tensor table,
                                                   torch::Tensor cursor) {
   int64_t cap = table.size(0) / 2 - 1;
@@ -862,6 +885,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "two-level LDS+global insert (int64 sum)");
   m.def("groupby_insert_mlp", &groupby_insert_mlp,
         "multi-row software-pipelined insert (int64 sum)");
+  m.def("groupby_insert_packed_rep", &groupby_insert_packed_rep,
+        "replicated packed insert (fabric-ceiling experiment)");
   m.def("groupby_insert_packed", &groupby_insert_packed,
         "packed-slot insert (int64 sum fast path)");
   m.def("groupby_compact_packed", &groupby_compact_packed,

@@ -341,6 +341,52 @@ extern "C" __global__ void k_groupby_insert_packed_sum_i64(
   }
 }
 
+// Sub-table replication experiment (NOTES.md item 1 / round-2 task:
+// confirm or refute the fabric/atomic-throughput ceiling): REP
+// replicas of the packed table, block b inserting into replica
+// b % REP (blocks land on XCD b % 8, so replica ~= XCD).  Relieves
+// per-line atomic contention at the cost of REP x the table footprint
+// (pushes a presized table out of the 256 MiB L3).  Merge = compact
+// each replica then re-aggregate (cheap at #distinct rows).
+extern "C" __global__ void k_groupby_insert_packed_rep(
+    const int64_t* keys, const int64_t* vals, int64_t n, int64_t* table,
+    int64_t cap, int32_t nrep, uint32_t seed, int32_t* sentinel_seen,
+    int32_t* overflow, int64_t max_probes) {
+  int64_t* mytab = table + (int64_t)(blockIdx.x % nrep) * 2 * (cap + 1);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t mask = (uint64_t)cap - 1;
+  for (; i < n; i += stride) {
+    int64_t k = keys[i];
+    int64_t slot;
+    if (k == GB_SENTINEL) {
+      atomicOr(sentinel_seen, 1);
+      slot = cap;
+    } else {
+      uint64_t h = mm3_u64((uint64_t)k, seed) & mask;
+      int64_t probes = 0;
+      for (;;) {
+        long long cur = ((volatile long long*)mytab)[2 * h];
+        if (cur == k) break;
+        if (cur == GB_SENTINEL) {
+          long long prev = atomicCAS((unsigned long long*)&mytab[2 * h],
+                                     (unsigned long long)GB_SENTINEL,
+                                     (unsigned long long)k);
+          if (prev == GB_SENTINEL || prev == k) break;
+        }
+        h = (h + 1) & mask;
+        if (++probes >= max_probes) {
+          atomicOr(overflow, 1);
+          return;
+        }
+      }
+      slot = (int64_t)h;
+    }
+    atomicAdd((unsigned long long*)&mytab[2 * slot + 1],
+              (unsigned long long)vals[i]);
+  }
+}
+
 extern "C" __global__ void k_groupby_compact_packed(
     const int64_t* table, int64_t cap, int64_t slots_per_block,
     int64_t* out_keys, int64_t* out_vals, unsigned long long* cursor) {

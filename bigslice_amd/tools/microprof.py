@@ -36,8 +36,12 @@ def timeit(fn, iters):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("which", choices=["groupby", "groupby_pp",
+                                      "insert", "insert_rep",
                                       "partition", "hash", "compact",
                                       "sortcombine", "hashbytes"])
+    ap.add_argument("--nrep", type=int, default=8)
+    ap.add_argument("--cap", type=int, default=0,
+                    help="table capacity (0 = 2x nkeys rounded up)")
     ap.add_argument("--rows", type=int, default=125_000_000)
     ap.add_argument("--nkeys", type=int, default=1_000_000)
     ap.add_argument("--nparts", type=int, default=8)
@@ -86,6 +90,32 @@ def main():
                 off += c
             t.finish()
         ms = timeit(run, args.iters)
+    elif args.which in ("insert", "insert_rep"):
+        # steady-state insert throughput into a presized packed table
+        # (all keys present after warmup: pure probe+atomicAdd path);
+        # insert_rep = the sub-table replication experiment
+        cap = args.cap
+        if not cap:
+            cap = 1024
+            while cap < 2 * args.nkeys:
+                cap <<= 1
+        nrep = args.nrep if args.which == "insert_rep" else 1
+        table = kernels._C.alloc_packed_table(
+            cap * nrep + nrep - 1,  # nrep*(cap+1) slots
+            torch.empty(0, dtype=torch.int64, device=dev))
+        flags = torch.zeros(2, dtype=torch.int32, device=dev)
+        out["cap"] = cap
+        out["nrep"] = nrep
+
+        def run():
+            if nrep > 1:
+                kernels._C.groupby_insert_packed_rep(
+                    keys, vals, table, nrep, flags, 4096)
+            else:
+                kernels._C.groupby_insert_packed(
+                    keys, vals, table, flags, 4096)
+        ms = timeit(run, args.iters)
+        assert int(flags[1].item()) == 0, "table overflow"
     elif args.which == "compact":
         t = kernels.GroupTable([torch.int64], ["sum"], dev)
         t.insert(keys, [vals])
