@@ -83,7 +83,13 @@ class MemoryStore(Store):
             return "none"
         budget = int(os.environ.get("BIGSLICE_STORE_HOST_BUDGET_BYTES",
                                     str(128 << 30)))
-        return "host" if self._host_bytes < budget else "disk"
+        if self._host_bytes >= budget:
+            return "disk"
+        # also charge the process-wide accountant shared with the
+        # spiller: the sum of all host tiers must fit the box's DRAM
+        from ..utils import hostmem
+        nb = sum(f.nbytes() for f in frames)
+        return "host" if hostmem.reserve(nb) else "disk"
 
     def _disk_path(self, task_name, partition) -> str:
         import tempfile
@@ -180,10 +186,14 @@ class MemoryStore(Store):
         return (sum(f.nbytes() for f in frames), rows)
 
     def discard_task(self, task_name):
+        from ..utils import hostmem
         with self._lock:
             for k in [k for k in self._data if k[0] == task_name]:
                 frames, _ = self._data.pop(k)
-                self._host_bytes -= self._host_keys.pop(k, 0)
+                nb = self._host_keys.pop(k, 0)
+                if nb:
+                    self._host_bytes -= nb
+                    hostmem.release(nb)
                 if isinstance(frames, _DiskEntry):
                     try:
                         os.unlink(frames.path)

@@ -95,6 +95,7 @@ class Spiller:
                  dir: Optional[str] = None, pin: bool = None):
         self.host_budget = host_budget_bytes
         self.host_used = 0
+        self._accounted = 0  # bytes charged to the global accountant
         self.batches: List[object] = []  # _HostBatch | Frame | _DiskBatch
         self.rows = 0
         self._dir = dir
@@ -113,17 +114,23 @@ class Spiller:
         out asynchronously on the copy stream; past
         SPILL_BACKPRESSURE_BYTES of un-drained device memory the caller
         blocks on the oldest copy (producers outrun the host link on
-        large jobs and the backlog would OOM HBM)."""
+        large jobs and the backlog would OOM HBM).  Host placement is
+        charged against BOTH this spiller's budget and the process-wide
+        accountant (utils.hostmem) shared with the store tier; either
+        refusal sends the batch to the unbounded disk tier."""
         self._release_completed()
         from .. import config
+        from ..utils import hostmem
         while _outstanding[0] > config.SPILL_BACKPRESSURE_BYTES:
             self._wait_oldest()
             self._release_completed()
         nbytes = frame.nbytes()
         self.rows += len(frame)
-        if self.host_used + nbytes <= self.host_budget:
+        if self.host_used + nbytes <= self.host_budget and \
+                hostmem.reserve(nbytes):
             self.batches.append(self._to_host(frame))
             self.host_used += nbytes
+            self._accounted += nbytes
             return nbytes
         d = self._ensure_dir()
         path = os.path.join(d, f"spill-{len(self.batches):06d}")
@@ -197,6 +204,10 @@ class Spiller:
 
     def close(self) -> None:
         self.batches.clear()
+        if self._accounted:
+            from ..utils import hostmem
+            hostmem.release(self._accounted)
+            self._accounted = 0
         if self._tmpdir is not None:
             self._tmpdir.cleanup()
             self._tmpdir = None

@@ -102,6 +102,33 @@ def test_spiller_roundtrip_memory_and_disk():
     sp.close()
 
 
+def test_spiller_global_host_accountant(monkeypatch):
+    """The process-wide host accountant (utils.hostmem) caps the SUM of
+    all spillers' host tiers: once exhausted, batches overflow to disk
+    even under a generous per-spiller budget, and close() releases the
+    reservation (VERDICT task 3 / reference spiller.go unbounded disk
+    tier)."""
+    from bigslice_amd.sliceio.spiller import _DiskBatch
+    from bigslice_amd.utils import hostmem
+    monkeypatch.setenv("BIGSLICE_HOST_BUDGET_BYTES", "600")
+    base = hostmem.used_bytes()
+    sp1 = Spiller(host_budget_bytes=1 << 30)
+    sp2 = Spiller(host_budget_bytes=1 << 30)
+    for sp in (sp1, sp2):
+        for _ in range(3):
+            sp.spill(make_frame(16))  # 16 rows x 2 cols x 8B = 256 B
+    # global budget 600 B admits ~2 host batches total; the rest are
+    # disk batches despite the large per-spiller budgets
+    disk = sum(isinstance(b, _DiskBatch)
+               for sp in (sp1, sp2) for b in sp.batches)
+    assert disk >= 4, disk
+    assert hostmem.used_bytes() - base <= 600
+    assert read_all(sp1.reader()) and read_all(sp2.reader())
+    sp1.close()
+    sp2.close()
+    assert hostmem.used_bytes() == base  # reservations released
+
+
 def test_frame_reader_chunks():
     f = make_frame(10)
     r = FrameReader(f, chunk=3)
