@@ -69,7 +69,150 @@ class SegmentedColumn:
         return [v[a:b].tolist() for a, b in zip(s, e)]
 
 
-Column = Union[torch.Tensor, list, SegmentedColumn]
+class BytesColumn:
+    """Device-resident variable-length byte rows: row i is
+    data[offsets[i]:offsets[i+1]] of one uint8 tensor.  The first-class
+    string/bytes column of the reference (frame/ops_builtin.go:143-164),
+    MI355X-native: hashing runs the K17 murmur3 kernel, partitioning and
+    shuffling move (lengths, bytes) tensor pairs, and ordering uses
+    64-bit two-seed dictionary ids (strings.string_ids) — grouping is by
+    id (a 2^-64 per-pair collision is accepted, like any dictionary
+    encoding)."""
+
+    __slots__ = ("data", "offsets")
+
+    def __init__(self, data: torch.Tensor, offsets: torch.Tensor):
+        self.data = data          # uint8 [total]
+        self.offsets = offsets    # int64 [n+1], absolute into data
+
+    @staticmethod
+    def from_list(rows, device: str = "cpu") -> "BytesColumn":
+        """Build from a list of bytes/str."""
+        import numpy as np
+        enc = [r.encode("utf-8") if isinstance(r, str) else bytes(r)
+               for r in rows]
+        offsets = np.zeros(len(enc) + 1, dtype=np.int64)
+        np.cumsum(np.fromiter((len(b) for b in enc), np.int64, len(enc)),
+                  out=offsets[1:])
+        data = np.frombuffer(b"".join(enc), dtype=np.uint8).copy()
+        return BytesColumn(
+            torch.from_numpy(data).to(device),
+            torch.from_numpy(offsets).to(device))
+
+    def __len__(self):
+        return self.offsets.shape[0] - 1
+
+    def __getitem__(self, idx):
+        if isinstance(idx, slice):
+            start, stop, step = idx.indices(len(self))
+            if step != 1:
+                raise TypeError("BytesColumn slicing requires step 1")
+            # zero-copy: offsets subrange stays absolute into data
+            return BytesColumn(self.data, self.offsets[start:stop + 1])
+        raise TypeError("BytesColumn supports slice indexing only")
+
+    @property
+    def device(self):
+        return self.data.device
+
+    def nbytes(self) -> int:
+        base = int(self.offsets[0]) if len(self) else 0
+        end = int(self.offsets[-1]) if len(self) else 0
+        return (end - base) + self.offsets.numel() * 8
+
+    def lengths(self) -> torch.Tensor:
+        return self.offsets[1:] - self.offsets[:-1]
+
+    def compacted(self) -> "BytesColumn":
+        """Materialize a zero-based copy (drops slack from slicing)."""
+        if len(self) == 0:
+            return BytesColumn(
+                torch.empty(0, dtype=torch.uint8,
+                            device=self.data.device),
+                torch.zeros(1, dtype=torch.int64,
+                            device=self.offsets.device))
+        base = self.offsets[0]
+        data = self.data[base:self.offsets[-1]].contiguous()
+        return BytesColumn(data, (self.offsets - base).contiguous())
+
+    def select(self, indices: torch.Tensor) -> "BytesColumn":
+        idx = indices.to(self.offsets.device)
+        lens = self.lengths()[idx]
+        n_out = idx.shape[0]
+        offs = torch.zeros(n_out + 1, dtype=torch.int64,
+                           device=self.offsets.device)
+        torch.cumsum(lens, 0, out=offs[1:])
+        total = int(offs[-1]) if n_out else 0
+        if total == 0:
+            return BytesColumn(
+                torch.empty(0, dtype=torch.uint8,
+                            device=self.data.device), offs)
+        # gather: absolute source index for every output byte
+        row_id = torch.repeat_interleave(
+            torch.arange(n_out, device=idx.device), lens)
+        within = torch.arange(total, device=idx.device) - offs[row_id]
+        src = self.offsets[idx][row_id] + within
+        return BytesColumn(self.data[src], offs)
+
+    def to(self, device, non_blocking=False) -> "BytesColumn":
+        c = self.compacted()
+        return BytesColumn(
+            c.data.to(device, non_blocking=non_blocking),
+            c.offsets.to(device, non_blocking=non_blocking))
+
+    def clone(self) -> "BytesColumn":
+        c = self.compacted()
+        return BytesColumn(c.data.clone(), c.offsets.clone())
+
+    def ids64(self) -> torch.Tensor:
+        """64-bit two-seed murmur3 dictionary ids per row (device
+        kernel on GPU, bit-identical host fallback): the sort/group
+        key for BYTES columns."""
+        from . import strings
+        c = self.compacted()
+        if c.data.is_cuda:
+            from . import kernels
+            return kernels._C.hash_bytes64(
+                c.data.contiguous(), c.offsets.contiguous(),
+                strings._ID_SEED_HI, strings._ID_SEED_LO)
+        from . import hashing
+        data = c.data.numpy().tobytes()
+        offs = c.offsets.tolist()
+        mm = hashing.murmur3_bytes
+        vals = []
+        for i in range(len(c)):
+            e = data[offs[i]:offs[i + 1]]
+            v = (mm(e, strings._ID_SEED_HI) << 32) | \
+                mm(e, strings._ID_SEED_LO)
+            vals.append(v - (1 << 64) if v >= (1 << 63) else v)
+        return torch.tensor(vals, dtype=torch.int64)
+
+    def hash32(self, seed: int) -> torch.Tensor:
+        """Row murmur3-32 (the K3/K17 partition hash), bit-identical to
+        the host OBJECT string path so placement is reproducible."""
+        c = self.compacted()
+        if c.data.is_cuda:
+            from . import kernels
+            return kernels._C.hash_bytes(c.data.contiguous(),
+                                         c.offsets.contiguous(), seed)
+        from . import hashing
+        data = c.data.numpy().tobytes()
+        offs = c.offsets.tolist()
+        import numpy as np
+        out = np.empty(len(c), dtype=np.uint32)
+        for i in range(len(c)):
+            out[i] = hashing.murmur3_bytes(data[offs[i]:offs[i + 1]],
+                                           seed)
+        return torch.from_numpy(out.astype(np.int64))
+
+    def tolists(self) -> list:
+        c = self.compacted()
+        data = c.data.cpu().numpy().tobytes()
+        offs = c.offsets.cpu().tolist()
+        return [data[offs[i]:offs[i + 1]] for i in range(len(c))]
+
+
+Column = Union[torch.Tensor, list, SegmentedColumn, BytesColumn]
 
 
 def _col_len(col: Column) -> int:
@@ -79,7 +222,12 @@ def _col_len(col: Column) -> int:
 
 
 def _col_dtype(col: Column):
-    return col.dtype if isinstance(col, torch.Tensor) else OBJECT
+    if isinstance(col, torch.Tensor):
+        return col.dtype
+    if isinstance(col, BytesColumn):
+        from .schema import BYTES
+        return BYTES
+    return OBJECT
 
 
 class Frame:
@@ -121,6 +269,9 @@ class Frame:
                 infer_dtype(c[0]) if c else OBJECT)
             if is_object(dt):
                 out.append(list(c))
+            elif dt == "bytes":
+                out.append(c if isinstance(c, BytesColumn)
+                           else BytesColumn.from_list(c, device))
             else:
                 out.append(torch.tensor(c, dtype=dt, device=device))
         if schema is not None and prefix is None:
@@ -133,6 +284,10 @@ class Frame:
         for dt in schema.dtypes:
             if is_object(dt):
                 cols.append([])
+            elif dt == "bytes":
+                cols.append(BytesColumn(
+                    torch.empty(0, dtype=torch.uint8, device=device),
+                    torch.zeros(1, dtype=torch.int64, device=device)))
             else:
                 cols.append(torch.empty(0, dtype=dt, device=device))
         return Frame(cols, schema.prefix)
@@ -161,7 +316,8 @@ class Frame:
     def has_objects(self) -> bool:
         """True when a column is host-only (Python lists); segmented
         device columns are NOT host-only."""
-        return any(not isinstance(c, (torch.Tensor, SegmentedColumn))
+        return any(not isinstance(c, (torch.Tensor, SegmentedColumn,
+                                      BytesColumn))
                    for c in self.columns)
 
     def nbytes(self) -> int:
@@ -169,7 +325,7 @@ class Frame:
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 total += c.numel() * c.element_size()
-            elif isinstance(c, SegmentedColumn):
+            elif isinstance(c, (SegmentedColumn, BytesColumn)):
                 total += c.nbytes()
             else:
                 total += sum(len(str(x)) for x in c)  # rough
@@ -189,7 +345,7 @@ class Frame:
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 cols.append(c[indices.to(c.device)])
-            elif isinstance(c, SegmentedColumn):
+            elif isinstance(c, (SegmentedColumn, BytesColumn)):
                 cols.append(c.select(indices))
             else:
                 idx = indices.cpu().tolist()
@@ -203,7 +359,7 @@ class Frame:
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 cols.append(c[keep.to(c.device)])
-            elif isinstance(c, SegmentedColumn):
+            elif isinstance(c, (SegmentedColumn, BytesColumn)):
                 cols.append(c.select(keep.nonzero().flatten()))
             else:
                 km = keep.cpu().numpy()
@@ -234,6 +390,17 @@ class Frame:
                 cols.append(SegmentedColumn(torch.cat(vals),
                                             torch.cat(starts),
                                             torch.cat(ends)))
+            elif isinstance(parts[0], BytesColumn):
+                datas, offs, off = [], [], 0
+                for p in parts:
+                    pc = p.compacted()
+                    datas.append(pc.data)
+                    offs.append(pc.offsets[:-1] + off)
+                    off += pc.data.shape[0]
+                offs.append(torch.tensor([off], dtype=torch.int64,
+                                         device=datas[0].device))
+                cols.append(BytesColumn(torch.cat(datas),
+                                        torch.cat(offs)))
             else:
                 merged: list = []
                 for p in parts:
@@ -244,7 +411,8 @@ class Frame:
     def to(self, device: str, non_blocking: bool = False) -> "Frame":
         cols: List[Column] = []
         for c in self.columns:
-            if isinstance(c, (torch.Tensor, SegmentedColumn)):
+            if isinstance(c, (torch.Tensor, SegmentedColumn,
+                              BytesColumn)):
                 cols.append(c.to(device, non_blocking=non_blocking))
             else:
                 if device != "cpu":
@@ -269,7 +437,7 @@ class Frame:
                                       device="cpu")
                 dst.copy_(c)
                 cols.append(dst)
-            elif isinstance(c, SegmentedColumn):
+            elif isinstance(c, (SegmentedColumn, BytesColumn)):
                 cols.append(c.to("cpu"))
             else:
                 cols.append(c)
@@ -279,6 +447,8 @@ class Frame:
         cols = []
         for c in self.columns:
             if isinstance(c, torch.Tensor):
+                cols.append(c.clone())
+            elif isinstance(c, BytesColumn):
                 cols.append(c.clone())
             elif isinstance(c, SegmentedColumn):
                 cols.append(SegmentedColumn(c.values.clone(),
@@ -309,7 +479,8 @@ class Frame:
         """Row permutation sorting lexicographically by prefix columns
         (reference frame.Less, frame/frame.go:375-385)."""
         n = len(self)
-        cols = self.columns[: self.prefix]
+        cols = [c.ids64() if isinstance(c, BytesColumn) else c
+                for c in self.columns[: self.prefix]]
         if any(not isinstance(c, torch.Tensor) for c in cols):
             from .hashing import custom_less_key
             keys = list(zip(*[list(c) if not isinstance(c, torch.Tensor)
@@ -335,7 +506,7 @@ class Frame:
         for c in self.columns:
             if isinstance(c, torch.Tensor):
                 out.append(c.cpu().tolist())
-            elif isinstance(c, SegmentedColumn):
+            elif isinstance(c, (SegmentedColumn, BytesColumn)):
                 out.append(c.tolists())
             else:
                 out.append(list(c))

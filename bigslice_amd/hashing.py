@@ -190,12 +190,26 @@ def hash_columns(cols: Sequence, seed: int = 0) -> torch.Tensor:
     frame.HashWithSeed).  Device tensors dispatch to the HIP kernel."""
     if not cols:
         raise ValueError("hash of zero key columns")
+    from .frame import BytesColumn
+    tensors = [c for c in cols if not isinstance(c, BytesColumn)]
+    bytes_cols = [c for c in cols if isinstance(c, BytesColumn)]
     first = cols[0]
-    if isinstance(first, torch.Tensor) and first.is_cuda:
+    on_device = (isinstance(first, torch.Tensor) and first.is_cuda) or (
+        isinstance(first, BytesColumn) and first.data.is_cuda)
+    if on_device:
         from . import kernels
-        return kernels.hash_columns_device(list(cols), seed)
+        h = None
+        if tensors:
+            h = kernels.hash_columns_device(tensors, seed)
+        for b in bytes_cols:
+            hb = b.hash32(seed)
+            h = hb if h is None else (h ^ hb)
+        return h
     h = None
     for c in cols:
-        hc = _hash_host_column(c, seed)
+        if isinstance(c, BytesColumn):
+            hc = c.hash32(seed).numpy().astype(np.uint32)
+        else:
+            hc = _hash_host_column(c, seed)
         h = hc if h is None else (h ^ hc)
     return torch.from_numpy(h.astype(np.int64))  # non-negative values

@@ -79,11 +79,77 @@ def make_aggregator(schema: Schema, agg: Aggregation, device: str,
                     seed: int = config.COMBINER_HASH_SEED):
     """Choose the right aggregator implementation for a schema:
     tensor path for any all-numeric schema (single- or multi-column
-    keys), host dict path for object keys / arbitrary combine fns."""
-    numeric = (not any(is_object(d) for d in schema.dtypes))
+    keys), device dictionary path for a single BYTES key, host dict
+    path for object keys / arbitrary combine fns."""
+    from ..schema import is_bytes
+    numeric = (not any(is_object(d) or is_bytes(d)
+                       for d in schema.dtypes))
     if agg.all_builtin and numeric and schema.prefix >= 1:
         return TensorAggregator(schema, agg, device)
+    if (agg.all_builtin and schema.prefix == 1
+            and is_bytes(schema.dtypes[0])
+            and not any(is_object(d) or is_bytes(d)
+                        for d in schema.dtypes[1:])):
+        return BytesKeyAggregator(schema, agg, device)
     return DictAggregator(schema, agg)
+
+
+class BytesKeyAggregator:
+    """Keyed reduce over a device varlen BYTES key: rows group by the
+    key's 64-bit dictionary id through the numeric TensorAggregator
+    (HIP GroupTable on GPU), while a per-aggregator dictionary maps
+    each id back to one exemplar byte row for the output.  This is the
+    device-resident string group-by of the reference's string ops
+    (frame/ops_builtin.go:143-164); id collisions are 2^-64 per pair,
+    the accepted dictionary-encoding tradeoff (see frame.BytesColumn).
+    """
+
+    def __init__(self, schema: Schema, agg: Aggregation, device: str):
+        from ..schema import Schema as S
+        self.schema = schema
+        inner_schema = S((torch.int64,) + tuple(schema.dtypes[1:]), 1)
+        self.inner = TensorAggregator(inner_schema, agg, device)
+        self.device = device
+        self.dict_ids: Optional[torch.Tensor] = None  # sorted
+        self.dict_keys = None  # BytesColumn exemplars, id-sorted
+
+    def add(self, frame: Frame) -> None:
+        if len(frame) == 0:
+            return
+        key_col = frame.columns[0]
+        ids = key_col.ids64()
+        uniq = torch.unique(ids)
+        if self.dict_ids is not None:
+            new = uniq[~torch.isin(uniq, self.dict_ids)]
+        else:
+            new = uniq
+        if new.numel():
+            # exemplar row per new id: first occurrence in this frame
+            sidx = torch.argsort(ids, stable=True)
+            pos = torch.searchsorted(ids[sidx], new)
+            exemplars = key_col.select(sidx[pos])
+            if self.dict_ids is None:
+                merged_ids, merged_keys = new, exemplars
+            else:
+                from ..frame import BytesColumn
+                cat_ids = torch.cat([self.dict_ids, new])
+                order = torch.argsort(cat_ids)
+                merged_ids = cat_ids[order]
+                both = Frame.concat([
+                    Frame([self.dict_keys], 1),
+                    Frame([exemplars], 1)]).columns[0]
+                merged_keys = both.select(order)
+            self.dict_ids = merged_ids
+            self.dict_keys = merged_keys
+        self.inner.add(Frame([ids] + list(frame.columns[1:]), 1))
+
+    def result_frames(self, chunk: int):
+        for f in self.inner.result_frames(chunk):
+            pos = torch.searchsorted(
+                self.dict_ids, f.columns[0].to(self.dict_ids.device))
+            keys = self.dict_keys.select(pos)
+            yield Frame([keys] + list(f.columns[1:]), 1,
+                        combined_id=f.combined_id)
 
 
 class TensorAggregator:

@@ -27,10 +27,14 @@ from .slice_base import Dep, Name, Slice, TaskContext
 
 def schema_of(*pytypes, prefix: int = None) -> Schema:
     """Build a Schema from Python types or torch dtypes:
-    str->OBJECT, int->int64, float->float64, bool->bool."""
+    str->OBJECT (host path), bytes->BYTES (device varlen byte rows),
+    int->int64, float->float64, bool->bool."""
+    from ..schema import BYTES
     dts = []
     for t in pytypes:
-        if t is str or t is bytes or t is object or t == OBJECT:
+        if t is bytes or t == BYTES:
+            dts.append(BYTES)
+        elif t is str or t is object or t == OBJECT:
             dts.append(OBJECT)
         elif t is int:
             dts.append(torch.int64)

@@ -21,9 +21,11 @@ def _columns_to_frame(cols, prefix=None) -> Frame:
     if isinstance(cols, Frame):
         return cols
     if isinstance(cols, (tuple, list)):
+        from ..frame import BytesColumn, SegmentedColumn
         conv = []
         for c in cols:
-            if isinstance(c, torch.Tensor):
+            if isinstance(c, (torch.Tensor, BytesColumn,
+                              SegmentedColumn)):
                 conv.append(c)
             elif isinstance(c, (list, tuple)):
                 if c and isinstance(c[0], (str, bytes, tuple)) or not c:

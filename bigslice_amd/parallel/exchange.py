@@ -43,8 +43,8 @@ from typing import Callable, Dict, List, Optional
 import torch
 import torch.distributed as dist
 
-from ..frame import Frame
-from ..schema import Schema
+from ..frame import BytesColumn, Frame
+from ..schema import Schema, is_bytes
 from .comm import Comm
 
 
@@ -92,14 +92,15 @@ def lpt_assign(sizes: List[int], world: int) -> List[int]:
 
 
 class _Window:
-    __slots__ = ("count_work", "gathered", "counts", "more_any",
-                 "err_any", "send_buf", "payload_works", "recv_cols",
-                 "my_meta", "launched", "consumed")
+    __slots__ = ("count_work", "gathered", "counts", "bcounts",
+                 "more_any", "err_any", "send_buf", "payload_works",
+                 "recv_cols", "my_meta", "launched", "consumed")
 
     def __init__(self):
         self.count_work = None
         self.gathered = None      # list of [nparts+2] tensors (transport)
         self.counts = None        # [world, nparts] host tensor after resolve
+        self.bcounts = None       # [world, nbcols, nparts] byte counts
         self.more_any = None
         self.err_any = False
         self.send_buf: Optional[Dict[int, List[Frame]]] = None
@@ -132,6 +133,10 @@ class PhaseExchange:
         self.comm = comm
         self.schema = schema
         self.nparts = nparts
+        # BYTES columns exchange as (lengths, data) tensor pairs; their
+        # per-partition byte counts ride extra lanes of the count plane
+        self.bytes_cols = [i for i, dt in enumerate(schema.dtypes)
+                           if is_bytes(dt)]
         self.resolve = resolve
         self.consume_cb = consume
         self.window_bytes = window_bytes or WINDOW_BYTES
@@ -188,9 +193,16 @@ class PhaseExchange:
         if k > 0:
             self._launch_payload(self.windows[k - 1])
         w = _Window()
-        counts = torch.zeros(self.nparts + 2, dtype=torch.int64)
+        nb = len(self.bytes_cols)
+        counts = torch.zeros(self.nparts * (1 + nb) + 2,
+                             dtype=torch.int64)
         for p, frames in self.buf.items():
             counts[p] = sum(len(f) for f in frames)
+            for j, c in enumerate(self.bytes_cols):
+                counts[(1 + j) * self.nparts + p] = sum(
+                    int(f.columns[c].offsets[-1] -
+                        f.columns[c].offsets[0])
+                    for f in frames)
         counts[-2] = 1 if err else 0
         counts[-1] = 1 if more else 0
         counts = counts.to(self._meta_device)
@@ -215,7 +227,11 @@ class PhaseExchange:
         w.gathered = None
         w.err_any = bool(host[:, -2].any())
         w.more_any = bool(host[:, -1].any())
-        w.counts = host[:, :-2]
+        w.counts = host[:, :self.nparts]
+        # per-bytes-col byte counts: [world, nbcols, nparts]
+        w.bcounts = host[:, self.nparts:-2].view(
+            self.comm.world, len(self.bytes_cols), self.nparts) \
+            if self.bytes_cols else None
 
     def _ensure_placement(self, w: _Window) -> None:
         if self.placement is not None:
@@ -240,14 +256,19 @@ class PhaseExchange:
         for p in order:
             in_splits[place[p]] += sum(len(f) for f in w.send_buf[p])
         my_meta, out_splits = [], [0] * world
+        nb = len(self.bytes_cols)
+        bout_splits = [[0] * world for _ in range(nb)]
         for src in range(world):
             for p in range(self.nparts):
                 if place[p] != me:
                     continue
                 r = int(w.counts[src, p])
+                bts = [int(w.bcounts[src, j, p]) for j in range(nb)]
                 if r:
-                    my_meta.append((src, p, r))
+                    my_meta.append((src, p, r, bts))
                     out_splits[src] += r
+                    for j in range(nb):
+                        bout_splits[j][src] += bts[j]
         w.my_meta = my_meta
         total_out = sum(out_splits)
         if int(w.counts.sum()) == 0:
@@ -258,18 +279,42 @@ class PhaseExchange:
             return
         recv_cols, works = [], []
         dev = torch.device(self.device)
-        for c, dt in enumerate(self.schema.dtypes):
-            parts = [f.columns[c].contiguous()
-                     for p in order for f in w.send_buf[p]]
+
+        def exchange(parts, dt, outs, ins):
             send = (torch.cat(parts) if parts else
                     torch.empty(0, dtype=dt, device=self.device))
             if send.device != dev:
                 send = send.to(self.device)
-            recv = torch.empty(total_out, dtype=dt, device=self.device)
+            recv = torch.empty(sum(outs), dtype=dt, device=self.device)
             works.append(dist.all_to_all_single(
-                recv, send, output_split_sizes=out_splits,
-                input_split_sizes=in_splits, async_op=True))
-            recv_cols.append(recv)
+                recv, send, output_split_sizes=outs,
+                input_split_sizes=ins, async_op=True))
+            return recv
+
+        for c, dt in enumerate(self.schema.dtypes):
+            if is_bytes(dt):
+                j = self.bytes_cols.index(c)
+                # lengths ride the row splits; data rides byte splits
+                lens = exchange(
+                    [f.columns[c].lengths()
+                     for p in order for f in w.send_buf[p]],
+                    torch.int64, out_splits, in_splits)
+                bins = [0] * world
+                for p in order:
+                    bins[place[p]] += sum(
+                        int(f.columns[c].offsets[-1] -
+                            f.columns[c].offsets[0])
+                        for f in w.send_buf[p])
+                data = exchange(
+                    [f.columns[c].compacted().data
+                     for p in order for f in w.send_buf[p]],
+                    torch.uint8, bout_splits[j], bins)
+                recv_cols.append((lens, data))
+            else:
+                recv_cols.append(exchange(
+                    [f.columns[c].contiguous()
+                     for p in order for f in w.send_buf[p]],
+                    dt, out_splits, in_splits))
         w.recv_cols = recv_cols
         w.payload_works = works
         w.send_buf = None  # the work object keeps the send tensors alive
@@ -282,8 +327,22 @@ class PhaseExchange:
             work.wait()
         w.payload_works = []
         off = 0
-        for (src, p, rows) in w.my_meta or []:
-            cols = [rc[off:off + rows] for rc in w.recv_cols]
+        boff = [0] * len(self.bytes_cols)
+        for (src, p, rows, bts) in w.my_meta or []:
+            cols = []
+            for c, rc in enumerate(w.recv_cols):
+                if isinstance(rc, tuple):
+                    j = self.bytes_cols.index(c)
+                    lens, data = rc
+                    l = lens[off:off + rows]
+                    offs = torch.zeros(rows + 1, dtype=torch.int64,
+                                       device=l.device)
+                    torch.cumsum(l, 0, out=offs[1:])
+                    cols.append(BytesColumn(
+                        data[boff[j]:boff[j] + bts[j]], offs))
+                    boff[j] += bts[j]
+                else:
+                    cols.append(rc[off:off + rows])
             self.consume_cb(p, Frame(cols, self.schema.prefix))
             off += rows
         w.recv_cols = []

@@ -16,6 +16,13 @@ import torch
 # Sentinel dtype for host-object columns (strings and arbitrary Python values).
 OBJECT = "object"
 
+# Sentinel dtype for DEVICE-RESIDENT variable-length byte rows
+# (frame.BytesColumn: one uint8 data tensor + int64 offsets).  Unlike
+# OBJECT, BYTES columns hash, partition, sort (by 64-bit dictionary
+# id), exchange and store on the device path — the first-class string
+# type of the reference (frame/ops_builtin.go:143-164), MI355X-native.
+BYTES = "bytes"
+
 _TORCH_DTYPES = {
     torch.int8, torch.uint8, torch.int16, torch.int32, torch.int64,
     torch.float16, torch.bfloat16, torch.float32, torch.float64,
@@ -28,7 +35,7 @@ _DTYPE_NAMES = {
     torch.int32: "i32", torch.int64: "i64", torch.uint32: "u32",
     torch.uint64: "u64", torch.float16: "f16", torch.bfloat16: "bf16",
     torch.float32: "f32", torch.float64: "f64", torch.bool: "b1",
-    OBJECT: "obj",
+    OBJECT: "obj", BYTES: "byt",
 }
 _NAME_DTYPES = {v: k for k, v in _DTYPE_NAMES.items()}
 
@@ -49,8 +56,12 @@ def is_object(dt) -> bool:
     return dt == OBJECT
 
 
+def is_bytes(dt) -> bool:
+    return dt == BYTES
+
+
 def check_dtype(dt):
-    if dt != OBJECT and dt not in _TORCH_DTYPES:
+    if dt != OBJECT and dt != BYTES and dt not in _TORCH_DTYPES:
         raise TypeError_(f"unsupported column dtype {dt!r}")
     return dt
 

@@ -30,7 +30,7 @@ from typing import Optional
 import numpy as np
 import torch
 
-from ..frame import Frame
+from ..frame import BytesColumn, Frame
 from ..schema import Schema
 
 MAGIC = 0xB16511CE
@@ -68,6 +68,13 @@ def encode_frame(frame: Frame, out: io.RawIOBase) -> int:
         if isinstance(c, torch.Tensor):
             payload = _tensor_bytes(c)
             body.write(struct.pack("<BQ", 0, len(payload)))
+        elif isinstance(c, BytesColumn):
+            # kind 2: varlen byte rows = offsets (i64) then data (u8)
+            cc = c.to("cpu").compacted()
+            off = _tensor_bytes(cc.offsets)
+            dat = _tensor_bytes(cc.data)
+            payload = struct.pack("<Q", len(off)) + off + dat
+            body.write(struct.pack("<BQ", 2, len(payload)))
         else:
             payload = pickle.dumps(c, protocol=pickle.HIGHEST_PROTOCOL)
             body.write(struct.pack("<BQ", 1, len(payload)))
@@ -118,6 +125,14 @@ def decode_frame(inp: io.RawIOBase, device: str = "cpu") -> Optional[Frame]:
             if device != "cpu":
                 t = t.to(device, non_blocking=True)
             cols.append(t)
+        elif kind == 2:
+            (olen,) = struct.unpack("<Q", payload[:8])
+            offs = _tensor_from_bytes(payload[8:8 + olen], torch.int64)
+            data = _tensor_from_bytes(payload[8 + olen:], torch.uint8)
+            col = BytesColumn(data, offs)
+            if device != "cpu":
+                col = col.to(device, non_blocking=True)
+            cols.append(col)
         else:
             cols.append(pickle.loads(payload))
     (want_crc,) = struct.unpack("<I", _read_exact(inp, 4))
