@@ -65,6 +65,93 @@ class Comm:
         self.world = world
         self.device = device
         self.backend = dist.get_backend() if is_initialized() else None
+        # stable identity across communicator rebuilds (rank/world
+        # change when the group shrinks after a peer loss)
+        self.orig_rank = rank
+        self.epoch = 0
+
+    def rebuild(self, coord_dir: str, grace_s: float = None,
+                max_wait_s: float = 120.0) -> None:
+        """Shrink-rebuild after a peer loss (in-run recovery, the
+        collective adaptation of the reference's transparent
+        lost-task recompute, exec/eval.go:352-376).
+
+        Protocol: every survivor announces itself under a per-epoch
+        coordination directory; once the alive set has been stable for
+        ``grace_s`` the lowest announced rank COMMITS the member list
+        with an exclusive create (first writer wins), and everyone
+        joins the committed set over a FileStore rendezvous with
+        survivors densely renumbered.  A survivor that detects the
+        failure too late to make the commit (its collective timeout
+        exceeded the grace window) fails cleanly instead of splitting
+        the group — size grace above the transport timeout to include
+        every survivor.  The old communicator is destroyed
+        unconditionally first (an in-flight peer death poisons it)."""
+        import json
+        import time
+        if grace_s is None:
+            try:
+                grace_s = float(os.environ.get(
+                    "BIGSLICE_RECOVERY_GRACE_S", "10"))
+            except ValueError:
+                grace_s = 10.0
+        self.epoch += 1
+        base = os.path.join(coord_dir, f"epoch{self.epoch:04d}")
+        os.makedirs(base, exist_ok=True)
+        try:
+            if dist.is_initialized():
+                dist.destroy_process_group()
+        except Exception:
+            pass
+        with open(os.path.join(base, f"alive-{self.orig_rank:04d}"),
+                  "w"):
+            pass
+        commit_path = os.path.join(base, "members.json")
+
+        def alive():
+            return sorted(
+                int(f.split("-")[1]) for f in os.listdir(base)
+                if f.startswith("alive-"))
+
+        deadline = time.monotonic() + max_wait_s
+        prev, stable_since = alive(), time.monotonic()
+        survivors = None
+        while time.monotonic() < deadline:
+            if os.path.exists(commit_path):
+                with open(commit_path) as fp:
+                    survivors = json.load(fp)
+                break
+            cur = alive()
+            now = time.monotonic()
+            if cur != prev:
+                prev, stable_since = cur, now
+            elif now - stable_since >= grace_s and \
+                    prev and prev[0] == self.orig_rank:
+                # I am the lowest stable survivor: commit the set
+                # (exclusive create — first writer wins any race)
+                try:
+                    with open(commit_path, "x") as fp:
+                        json.dump(prev, fp)
+                    survivors = prev
+                except FileExistsError:
+                    continue
+                break
+            time.sleep(0.1)
+        if survivors is None:
+            raise RuntimeError("recovery rendezvous timed out")
+        if self.orig_rank not in survivors:
+            raise RuntimeError(
+                f"rank {self.orig_rank} detected the peer loss after "
+                f"the surviving group {survivors} committed; cannot "
+                "rejoin this epoch")
+        new_rank = survivors.index(self.orig_rank)
+        store = dist.FileStore(os.path.join(base, "rdzv"),
+                               len(survivors))
+        dist.init_process_group(self.backend or "gloo", store=store,
+                                rank=new_rank,
+                                world_size=len(survivors))
+        self.rank = new_rank
+        self.world = len(survivors)
 
     @property
     def tensor_exchange_ok(self) -> bool:

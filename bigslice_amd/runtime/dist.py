@@ -77,6 +77,42 @@ class DistExecutor(Executor):
             self._run_phase(phase)
         self.comm.barrier()
 
+    # -- in-run rank-loss recovery ----------------------------------------
+
+    @property
+    def recoverable(self) -> bool:
+        """In-run shrink-recovery needs persistent phase checkpoints
+        (survivors re-run phases whose partitions died with the lost
+        rank) and a coordination directory."""
+        return getattr(self.store, "persistent", False) and \
+            getattr(self, "recovery_dir", None) is not None
+
+    def recover(self, roots: Sequence[Task]) -> None:
+        """After a peer loss poisons the communicator: rebuild the
+        group over the survivors, then reset every placement/ownership
+        decision and task state so the next evaluate() re-validates
+        each phase against the (persistent) stores — complete phases
+        skip via checkpoint discovery, phases whose partitions lived
+        on the lost rank re-run on the shrunk world."""
+        self.comm.rebuild(self.recovery_dir)
+        self.device = self.comm.device
+        self.placement.clear()
+        self._task_owner.clear()
+        self._group_domain.clear()
+        self._domain_members.clear()
+        seen = set()
+
+        def reset(t: Task):
+            if id(t) in seen:
+                return
+            seen.add(id(t))
+            t.set_state(TaskState.INIT)
+            for dep in t.deps:
+                for h in dep.head_tasks:
+                    reset(h)
+        for r in roots:
+            reset(r)
+
     # -- graph walks ------------------------------------------------------
 
     def _phase_order(self, roots: Sequence[Task]) -> List[List[Task]]:

@@ -196,6 +196,23 @@ class Result(Slice):
             visit(t)
 
 
+def _is_comm_failure(e: BaseException) -> bool:
+    """Transport-level failure (a peer died) vs an application error
+    that the engine already surfaced collectively on live ranks."""
+    try:
+        import torch.distributed as dist
+        if isinstance(e, dist.DistBackendError):
+            return True
+    except Exception:
+        pass
+    if not isinstance(e, RuntimeError):
+        return False
+    msg = str(e)
+    return any(s in msg for s in (
+        "onnection", "Socket", "Timed out", "imed out", "peer",
+        "NCCL", "Gloo", "aborted"))
+
+
 # -- Session ---------------------------------------------------------------
 
 class Session:
@@ -273,7 +290,24 @@ class Session:
             env.freeze()
             ev = getattr(self.executor, "evaluate", None)
             if ev is not None:
-                ev(tasks)
+                # In-run rank-loss recovery: a transport failure (a
+                # peer died inside a collective) triggers a shrink-
+                # rebuild + re-evaluate when the executor has
+                # persistent checkpoints; completed phases skip, lost
+                # partitions recompute on the survivors.
+                attempts = 0
+                while True:
+                    try:
+                        ev(tasks)
+                        break
+                    except BaseException as e:
+                        if (getattr(self.executor, "recoverable", False)
+                                and _is_comm_failure(e)
+                                and attempts < 3):
+                            attempts += 1
+                            self.executor.recover(tasks)
+                            continue
+                        raise
             else:
                 evaluate(self.executor, tasks)
             return Result(self, slice_, tasks)
@@ -332,6 +366,10 @@ def start(parallelism: int = None, device: str = None,
                 store = FileStore(os.path.join(
                     checkpoint_dir, f"rank{comm.rank:03d}"))
             executor = DistExecutor(comm, store=store)
+            if checkpoint_dir is not None:
+                # enables in-run shrink-recovery after a rank loss
+                executor.recovery_dir = os.path.join(checkpoint_dir,
+                                                     "recovery")
         else:
             from .local import LocalExecutor
             store = None

@@ -1,0 +1,101 @@
+"""In-run rank-loss recovery (VERDICT task 6): a rank dies mid-job;
+the surviving ranks rebuild the process group over a file rendezvous,
+reload phase checkpoints, and finish with the correct result — no
+operator intervention, no full restart.  The collective adaptation of
+the reference's transparent lost-task recompute
+(exec/eval.go:352-376, exec/slicemachine.go:216-227)."""
+
+import os
+import shutil
+
+import torch
+import torch.multiprocessing as mp
+
+from tests.test_dist import _free_port
+
+CKPT = "/tmp/bigslice_rankloss_test"
+
+
+def _init_fast(rank, world, port):
+    # short collective timeout so the straggling survivor detects the
+    # loss inside the recovery grace window (grace > timeout)
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank),
+        "BIGSLICE_RECOVERY_GRACE_S": "8",
+    })
+    from datetime import timedelta
+
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            timeout=timedelta(seconds=5))
+
+
+def _rankloss_worker(rank, world, port, q):
+    _init_fast(rank, world, port)
+    import bigslice_amd as bs
+
+    def build(m):
+        def gen(shard, ctx):
+            if os.environ.get("RANK") == str(world - 1) and shard == 3:
+                # die mid-phase: peers are (or will be) blocked in the
+                # exchange collectives
+                import time
+                time.sleep(0.5)
+                os._exit(17)
+            keys = torch.arange(300, dtype=torch.int64) % 17
+            yield (keys, torch.full_like(keys, shard + 1))
+        src = bs.ReaderFunc(m, gen, bs.schema_of(int, int))
+        return bs.Reduce(src, "sum")
+
+    fv = bs.func(build)
+    sess = bs.start(distributed=True, device="cpu",
+                    checkpoint_dir=CKPT)
+    res = sess.run(fv, 6)
+    rows = sorted(res.scan())
+    comm = sess.executor.comm
+    q.put((rank, rows, comm.world, comm.epoch))
+    # survivors tear down their (shrunk) group cleanly
+    import torch.distributed as dist
+    if dist.is_initialized():
+        try:
+            dist.barrier()
+        except Exception:
+            pass
+        dist.destroy_process_group()
+
+
+def test_rank_loss_mid_job_recovers():
+    world = 4
+    shutil.rmtree(CKPT, ignore_errors=True)
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_rankloss_worker,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world - 1):  # the killed rank never reports
+        rank, rows, new_world, epoch = q.get()
+        results[rank] = (rows, new_world, epoch)
+    for p in procs:
+        p.join(120)
+    assert procs[world - 1].exitcode == 17  # the victim died
+    for r in range(world - 1):
+        assert procs[r].exitcode == 0
+    # every shard's contribution arrives exactly once despite the loss
+    expect = {}
+    for shard in range(6):
+        keys = (torch.arange(300) % 17).tolist()
+        for k in keys:
+            expect[k] = expect.get(k, 0) + shard + 1
+    rows, new_world, epoch = results[0]
+    assert rows == sorted(expect.items())
+    assert new_world == world - 1  # the group shrank
+    assert epoch == 1              # exactly one rebuild
+    for r in (1, 2):
+        assert results[r][0] == []  # non-root ranks see no rows
+        assert results[r][1] == world - 1
