@@ -129,7 +129,7 @@ class PhaseExchange:
                  resolve: Callable[[Optional[List[int]]],
                                    Optional[List[int]]],
                  consume: Callable[[int, Frame], None],
-                 window_bytes: int = None):
+                 window_bytes: int = None, tracer=None, pid: int = 0):
         self.comm = comm
         self.schema = schema
         self.nparts = nparts
@@ -150,6 +150,8 @@ class PhaseExchange:
         self.buf_bytes = 0
         self.windows: List[_Window] = []
         self.err_any = False
+        self.tracer = tracer
+        self.pid = pid
 
     # -- producer side ----------------------------------------------------
 
@@ -209,6 +211,9 @@ class PhaseExchange:
         gathered = [torch.empty_like(counts)
                     for _ in range(self.comm.world)]
         w.count_work = dist.all_gather(gathered, counts, async_op=True)
+        if self.tracer:
+            self.tracer.emit(f"window{k}:counts-launch", "i",
+                             pid=self.pid, tid=90)
         w.gathered = gathered
         w.send_buf = self.buf
         self.buf = {}
@@ -317,12 +322,21 @@ class PhaseExchange:
                     dt, out_splits, in_splits))
         w.recv_cols = recv_cols
         w.payload_works = works
+        if self.tracer:
+            self.tracer.emit(
+                f"window{self.windows.index(w)}:payload-launch", "i",
+                pid=self.pid, tid=90)
         w.send_buf = None  # the work object keeps the send tensors alive
 
     def _consume(self, w: _Window) -> None:
         if w.consumed:
             return
         w.consumed = True
+        span = self.tracer.span(
+            f"window{self.windows.index(w)}:payload-wait+consume",
+            pid=self.pid, tid=90) if self.tracer else None
+        if span:
+            span.__enter__()
         for work in w.payload_works:
             work.wait()
         w.payload_works = []
@@ -346,3 +360,5 @@ class PhaseExchange:
             self.consume_cb(p, Frame(cols, self.schema.prefix))
             off += rows
         w.recv_cols = []
+        if span:
+            span.__exit__(None, None, None)

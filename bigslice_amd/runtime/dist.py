@@ -402,7 +402,8 @@ class DistExecutor(Executor):
                 recv_frames.setdefault(p, []).append(frame)
 
         ex = PhaseExchange(comm, schema, nparts,
-                           self._placement_resolver(phase), consume)
+                           self._placement_resolver(phase), consume,
+                           tracer=self.tracer, pid=comm.rank)
         ctx = TaskContext(device=device)
         err: Optional[BaseException] = None
         try:
@@ -443,7 +444,12 @@ class DistExecutor(Executor):
                                 ex.add(p, pf)
         except BaseException as e:
             err = e
-        ex.finish(err=err is not None)
+        if self.tracer:
+            with self.tracer.span(f"exchange-drain:{exemplar.name}",
+                                  pid=comm.rank, tid=90):
+                ex.finish(err=err is not None)
+        else:
+            ex.finish(err=err is not None)
         if combine_mode:
             grouped = {p: list(a.result_frames(ctx.chunk * 4))
                        for p, a in recv_aggs.items()}
