@@ -56,9 +56,21 @@ __global__ void k_radix_andor(const K* __restrict__ keys, int64_t n,
                               unsigned long long* __restrict__ out_or) {
   using U = std::conditional_t<sizeof(K) == 8, uint64_t, uint32_t>;
   U a = ~(U)0, o = 0;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
+  // 4 independent strided loads per iteration: the reduction is pure
+  // bandwidth, so keep enough loads in flight per lane
+  const int64_t stride4 = (int64_t)gridDim.x * blockDim.x * 4;
+  const int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t step = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = i0;
+  for (; i + 3 * step < n; i += stride4) {
+    const U u0 = (U)keys[i];
+    const U u1 = (U)keys[i + step];
+    const U u2 = (U)keys[i + 2 * step];
+    const U u3 = (U)keys[i + 3 * step];
+    a &= (u0 & u1) & (u2 & u3);
+    o |= (u0 | u1) | (u2 | u3);
+  }
+  for (; i < n; i += step) {
     const U u = (U)keys[i];
     a &= u;
     o |= u;

@@ -194,7 +194,8 @@ class GroupTable:
     # batch.  Launched WITHOUT a host sync (the device scalar is read
     # at the next insert or at finish, when the count is long done), so
     # the streaming hot path pays only ~0.1 ms of overlapped GPU time.
-    _CARD_SAMPLE = 1 << 19
+    _CARD_SAMPLE = int(os.environ.get("BIGSLICE_GB_SAMPLE_ROWS",
+                                      str(1 << 19)))
 
     def _start_sample(self, keys: torch.Tensor) -> None:
         forced = os.environ.get("BIGSLICE_GB_MODE")
