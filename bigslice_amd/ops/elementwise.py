@@ -47,6 +47,16 @@ def schema_of(*pytypes, prefix: int = None) -> Schema:
     return Schema(dts, prefix)
 
 
+def _warn_host_rowwise(op: str) -> None:
+    """One-time (per call site) notice that a rowwise UDF leaves the
+    device path on a GPU session: rows round-trip through Python."""
+    import warnings
+    warnings.warn(
+        f"{op}(rowwise=True) runs its UDF as a host row loop on a GPU "
+        "session (rows leave the device); prefer a vectorized UDF over "
+        "whole columns", RuntimeWarning, stacklevel=3)
+
+
 def _normalize_out(res, device_hint: str) -> List:
     """Normalize a UDF result into a column list."""
     if isinstance(res, Frame):
@@ -115,6 +125,8 @@ class Map(_PipelinedSlice):
     def reader(self, shard, dep_readers, ctx: TaskContext) -> Reader:
         src = dep_readers[0]
         fn, rowwise, schema = self.fn, self.rowwise, self.schema
+        if rowwise and ctx.device != "cpu":
+            _warn_host_rowwise("Map")
 
         def gen():
             for f in src:
@@ -139,6 +151,8 @@ class Filter(_PipelinedSlice):
     def reader(self, shard, dep_readers, ctx: TaskContext) -> Reader:
         src = dep_readers[0]
         fn, rowwise = self.fn, self.rowwise
+        if rowwise and ctx.device != "cpu":
+            _warn_host_rowwise("Filter")
 
         def gen():
             for f in src:
@@ -191,6 +205,8 @@ class Flatmap(_PipelinedSlice):
         fn, rowwise, schema = self.fn, self.rowwise, self.schema
         if self.fn_factory is not None:
             fn = self.fn_factory()
+        if rowwise and ctx.device != "cpu":
+            _warn_host_rowwise("Flatmap")
 
         def gen():
             for f in src:

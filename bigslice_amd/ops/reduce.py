@@ -85,6 +85,13 @@ class Fold(Slice):
         fn = self.fn
         schema = self.schema
         nkey = schema.prefix
+        if ctx.device != "cpu":
+            import warnings
+            warnings.warn(
+                "Fold runs its accumulator as a host row loop (rows "
+                "round-trip off the GPU); use Reduce with a builtin "
+                "combine ('sum'/'min'/'max'/'prod') for the device "
+                "path", RuntimeWarning, stacklevel=2)
 
         def gen():
             state = {}

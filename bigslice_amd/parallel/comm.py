@@ -25,7 +25,7 @@ import torch
 import torch.distributed as dist
 
 from ..frame import Frame
-from ..schema import Schema, is_object
+from ..schema import Schema, is_bytes, is_object
 
 
 def is_initialized() -> bool:
@@ -364,9 +364,15 @@ class Comm:
 
     def gather_frames(self, frames: List[Frame], schema: Schema,
                       root: int = 0) -> List[Frame]:
-        """Gather result frames to the root rank (scan readback)."""
+        """Gather result frames to the root rank (scan readback).
+        Numeric schemas ride tensor collectives (one size exchange +
+        per-column all_gather of padded device buffers) — no per-row
+        pickling; object schemas take the portable object path."""
         if self.world == 1:
             return frames
+        if self.tensor_exchange_ok and not any(
+                is_object(dt) or is_bytes(dt) for dt in schema.dtypes):
+            return self._gather_frames_tensor(frames, schema, root)
         payload = [f.to("cpu").column_lists() for f in frames]
         gathered = self.all_gather_obj(payload)
         if self.rank != root:
@@ -376,3 +382,40 @@ class Comm:
             for cols in g:
                 out.append(Frame.from_lists(cols, schema=schema))
         return out
+
+    def _gather_frames_tensor(self, frames: List[Frame],
+                              schema: Schema, root: int) -> List[Frame]:
+        device = self.device
+        if frames:
+            cat = Frame.concat(
+                [f.to(device) for f in frames]) if len(frames) > 1 \
+                else frames[0].to(device)
+            n = len(cat)
+        else:
+            cat = Frame.empty(schema, device)
+            n = 0
+        sizes = torch.tensor([n], dtype=torch.int64)
+        if self.backend == "nccl":
+            sizes = sizes.to(device)
+        all_sizes = [torch.empty_like(sizes) for _ in range(self.world)]
+        dist.all_gather(all_sizes, sizes)
+        counts = [int(s.cpu().item()) for s in all_sizes]
+        mx = max(counts)
+        if mx == 0:
+            return []
+        out_cols: List[torch.Tensor] = []
+        for c in range(schema.num_columns):
+            col = cat.columns[c]
+            pad = torch.empty(mx, dtype=schema.dtypes[c], device=device)
+            if n:
+                pad[:n] = col
+            bufs = [torch.empty_like(pad) for _ in range(self.world)]
+            # all_gather (supported on both gloo and nccl; plain
+            # gather is not collective-symmetric under nccl)
+            dist.all_gather(bufs, pad)
+            if self.rank == root:
+                out_cols.append(torch.cat(
+                    [b[:cnt] for b, cnt in zip(bufs, counts) if cnt]))
+        if self.rank != root:
+            return []
+        return [Frame(out_cols, schema.prefix).to("cpu")]

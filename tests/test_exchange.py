@@ -216,6 +216,54 @@ def test_cogroup_shared_domain_world3():
         assert out[k] == (60, 20), (k, out[k])
 
 
+def _gather_worker(rank, world, port, q):
+    """Tensor-path result gather: uneven per-rank sizes including an
+    empty rank; no object pickling on numeric schemas."""
+    _init(rank, world, port)
+    import torch as t
+    from bigslice_amd import schema_of
+    from bigslice_amd.frame import Frame
+    from bigslice_amd.parallel.comm import Comm
+    comm = Comm(rank, world, "cpu")
+    schema = schema_of(int, float)
+    frames = []
+    if rank != 1:  # rank 1 contributes nothing
+        n = 10 * (rank + 1)
+        frames = [Frame([t.arange(n, dtype=t.int64) + rank * 1000,
+                         t.full((n,), float(rank))], 1)]
+    out = comm.gather_frames(frames, schema)
+    q.put((rank, sorted(r for f in out for r in f.rows())))
+
+
+def test_tensor_result_gather_world3():
+    results = _run_workers(_gather_worker, world=3)
+    want = []
+    for rank in (0, 2):
+        want += [(rank * 1000 + i, float(rank))
+                 for i in range(10 * (rank + 1))]
+    assert results[0] == sorted(want)
+    assert results[1] == [] and results[2] == []
+
+
+def test_rowwise_and_fold_warn_on_gpu_session():
+    import warnings
+    import bigslice_amd as bs
+    from bigslice_amd.ops.slice_base import TaskContext
+    s = bs.Map(bs.Const(1, torch.arange(3, dtype=torch.int64)),
+               lambda x: (x,), out_schema=(int,), rowwise=True)
+    f = bs.Fold(bs.Const(1, torch.arange(3, dtype=torch.int64),
+                         torch.ones(3, dtype=torch.int64)),
+                lambda a, v: (a or 0) + v, out_schema=(int,))
+    ctx = TaskContext(device="cuda:0")  # no GPU needed to warn
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        s.reader(0, [iter([])], ctx)
+        f.reader(0, [iter([])], ctx)
+    msgs = [str(x.message) for x in w]
+    assert any("host row loop" in m for m in msgs), msgs
+    assert any("Fold" in m for m in msgs), msgs
+
+
 def test_bench_world8_cpu():
     """CI form of the scaling bench: torchrun world 8 over gloo runs
     bench.py end-to-end (the identical code path the driver's 8-GPU
