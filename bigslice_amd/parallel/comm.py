@@ -137,6 +137,11 @@ class Comm:
                     continue
                 break
             time.sleep(0.1)
+        if os.environ.get("BIGSLICE_RECOVERY_DEBUG"):
+            import sys
+            print(f"[recovery] orig_rank={self.orig_rank} "
+                  f"epoch={self.epoch} survivors={survivors}",
+                  file=sys.stderr, flush=True)
         if survivors is None:
             raise RuntimeError("recovery rendezvous timed out")
         if self.orig_rank not in survivors:

@@ -237,31 +237,43 @@ class DistExecutor(Executor):
             return False
         comm, world = self.comm, self.comm.world
         exemplar = phase[0]
+        has_local = getattr(self.store, "has_local", self.store.has)
+
+        def claim(name, p):
+            # local owners outrank peer-dir adoption (a lost/orphaned
+            # rank's checkpoint files, readable via FileStore peers —
+            # the Worker.Read pull-path role): rank r claims at r when
+            # the partition is in its own store, at world+r when only
+            # reachable through a sibling directory, 2*world = absent.
+            if has_local(name, p):
+                return comm.rank
+            if self.store.has(name, p):
+                return world + comm.rank
+            return 2 * world
+
         if exemplar.shuffle_out:
             nparts = exemplar.num_partitions
-            local = torch.full((nparts,), world, dtype=torch.int64)
+            local = torch.empty(nparts, dtype=torch.int64)
             for p in range(nparts):
-                if self.store.has(exemplar.name, p) or any(
-                        self.store.has(t.name, p) for t in phase):
-                    local[p] = comm.rank
+                c = min(claim(t.name, p) for t in [exemplar] + phase)
+                local[p] = c
             owner = comm.min_reduce(local)
-            if bool((owner >= world).any()):
+            if bool((owner >= 2 * world).any()):
                 return False
-            place = [int(r) for r in owner]
+            place = [int(r) % world for r in owner]
             key = phase[0].name
             dom = self._group_domain.get(key, key)
             for m in self._domain_members.get(dom, [key]):
                 self.placement.setdefault(m, place)
             return True
-        local = torch.full((len(phase),), world, dtype=torch.int64)
+        local = torch.empty(len(phase), dtype=torch.int64)
         for i, t in enumerate(phase):
-            if self.store.has(t.name, 0):
-                local[i] = comm.rank
+            local[i] = claim(t.name, 0)
         owner = comm.min_reduce(local)
-        if bool((owner >= world).any()):
+        if bool((owner >= 2 * world).any()):
             return False
         for i, t in enumerate(phase):
-            self._task_owner[t.name] = int(owner[i])
+            self._task_owner[t.name] = int(owner[i]) % world
         return True
 
     # -- phase execution --------------------------------------------------
@@ -561,12 +573,18 @@ class DistExecutor(Executor):
         return out
 
     def gather_result(self, tasks: Sequence[Task], schema):
-        """Collective: gather all root-task outputs to rank 0.  A
-        partition lives in exactly one rank's store, so presence is
-        the ownership test."""
+        """Collective: gather all root-task outputs to rank 0.  Each
+        partition is contributed by its recorded owner (peer-store
+        adoption means mere presence is visible to EVERY rank, so
+        presence alone would duplicate rows); local-only presence is
+        the fallback when no owner was recorded."""
+        has_local = getattr(self.store, "has_local", self.store.has)
         frames = []
         for t in tasks:
-            if self.store.has(t.name, 0):
+            owner = self._task_owner.get(t.name)
+            mine = (owner == self.comm.rank) if owner is not None \
+                else has_local(t.name, 0)
+            if mine and self.store.has(t.name, 0):
                 r = self.store.open(t.name, 0, device="cpu")
                 frames.extend(list(r))
         return self.comm.gather_frames(frames, schema)

@@ -10,6 +10,7 @@ iterative computing).
 from __future__ import annotations
 
 import hashlib
+import os
 import threading
 from typing import Callable, List, Sequence
 
@@ -301,9 +302,18 @@ class Session:
                         ev(tasks)
                         break
                     except BaseException as e:
-                        if (getattr(self.executor, "recoverable", False)
-                                and _is_comm_failure(e)
-                                and attempts < 3):
+                        rec = getattr(self.executor, "recoverable",
+                                      False)
+                        comm_fail = _is_comm_failure(e)
+                        if os.environ.get("BIGSLICE_RECOVERY_DEBUG"):
+                            import sys as _s
+                            print(f"[recovery-gate] {type(e).__name__} "
+                                  f"recoverable={rec} "
+                                  f"comm_fail={comm_fail} "
+                                  f"attempts={attempts}: "
+                                  f"{str(e)[:120]}",
+                                  file=_s.stderr, flush=True)
+                        if rec and comm_fail and attempts < 3:
                             attempts += 1
                             self.executor.recover(tasks)
                             continue
@@ -363,8 +373,10 @@ def start(parallelism: int = None, device: str = None,
             store = None
             if checkpoint_dir is not None:
                 from .store import FileStore
-                store = FileStore(os.path.join(
-                    checkpoint_dir, f"rank{comm.rank:03d}"))
+                store = FileStore(
+                    os.path.join(checkpoint_dir,
+                                 f"rank{comm.rank:03d}"),
+                    peers_dir=checkpoint_dir)
             executor = DistExecutor(comm, store=store)
             if checkpoint_dir is not None:
                 # enables in-run shrink-recovery after a rank loss

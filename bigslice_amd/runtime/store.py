@@ -204,16 +204,41 @@ class MemoryStore(Store):
 class FileStore(Store):
     """File-backed store with the reference's layout and count footer
     (exec/store.go:173-261): {prefix}/{task}/p{partition} with 8-byte LE
-    record count appended after the encoded stream."""
+    record count appended after the encoded stream.
+
+    ``peers_dir`` (the checkpoint root holding every rank's store)
+    enables the pull-path role of the reference's Worker.Read
+    (exec/bigmachine.go:1332-1402): reads fall back to SIBLING ranks'
+    directories, so after a rank loss or a restart at a different
+    world size the survivors ADOPT the orphaned rank's completed
+    partitions instead of recomputing them.  Writes are atomic
+    (tmp+rename) and footer-validated, so only complete partitions are
+    ever visible — the resumability the reference gets from byte
+    offsets, provided here at partition granularity."""
 
     persistent = True
 
-    def __init__(self, prefix: str):
+    def __init__(self, prefix: str, peers_dir: str = None):
         self.prefix = prefix
+        self.peers_dir = peers_dir
 
     def _path(self, task_name: str, partition: int) -> str:
         safe = task_name.replace("/", "_")
         return os.path.join(self.prefix, safe, f"p{partition:03d}")
+
+    def _resolve(self, task_name: str, partition: int):
+        """Own path if present, else the first sibling rank's copy."""
+        p = self._path(task_name, partition)
+        if os.path.exists(p):
+            return p
+        if self.peers_dir:
+            import glob
+            safe = task_name.replace("/", "_")
+            for cand in sorted(glob.glob(os.path.join(
+                    self.peers_dir, "rank*", safe,
+                    f"p{partition:03d}"))):
+                return cand
+        return None
 
     def put(self, task_name, partition, frames, rows):
         path = self._path(task_name, partition)
@@ -226,10 +251,17 @@ class FileStore(Store):
         os.replace(tmp, path)
 
     def has(self, task_name, partition):
+        return self._resolve(task_name, partition) is not None
+
+    def has_local(self, task_name, partition):
+        """Present in THIS rank's own directory (checkpoint discovery
+        prefers local owners; peer adoption is the fallback)."""
         return os.path.exists(self._path(task_name, partition))
 
     def open(self, task_name, partition, device="cpu"):
-        path = self._path(task_name, partition)
+        path = self._resolve(task_name, partition)
+        if path is None:
+            raise KeyError(f"no output for {task_name} p{partition}")
         size = os.path.getsize(path)
         fp = open(path, "rb")
 
@@ -250,8 +282,8 @@ class FileStore(Store):
         return _R()
 
     def stat(self, task_name, partition):
-        path = self._path(task_name, partition)
-        if not os.path.exists(path):
+        path = self._resolve(task_name, partition)
+        if path is None:
             return (0, 0)
         size = os.path.getsize(path)
         with open(path, "rb") as fp:

@@ -66,6 +66,85 @@ def _rankloss_worker(rank, world, port, q):
         dist.destroy_process_group()
 
 
+def _adoption_worker(rank, world, port, q):
+    """Invocation 1 completes and checkpoints on all ranks; the victim
+    dies during invocation 2.  Survivors must ADOPT the victim's
+    invocation-1 partitions from its orphaned store directory (the
+    FileStore peers pull path) instead of recomputing them."""
+    _init_fast(rank, world, port)
+    import bigslice_amd as bs
+    ck = CKPT + "_adopt"
+
+    def build1(m):
+        def gen(shard, ctx):
+            open(os.path.join(ck, f"gen1-{os.environ['RANK']}-{shard}-"
+                              f"{len(os.listdir(ck))}"), "w").close()
+            keys = torch.arange(100, dtype=torch.int64) % 9
+            yield (keys, torch.full_like(keys, shard + 1))
+        src = bs.ReaderFunc(m, gen, bs.schema_of(int, int))
+        return bs.Reduce(src, "sum")
+
+    def build2(prev):
+        def boom(k, v):
+            if os.environ.get("RANK") == str(world - 1):
+                import time
+                time.sleep(0.5)
+                os._exit(17)
+            return (k, v * 10)
+        # out_schema matters: without it, schema inference samples the
+        # UDF at build time (before the evaluate retry scope)
+        return bs.Map(bs.Reshuffle(prev), boom, out_schema=(int, int))
+
+    fv1, fv2 = bs.func(build1), bs.func(build2)
+    sess = bs.start(distributed=True, device="cpu",
+                    checkpoint_dir=ck)
+    r1 = sess.run(fv1, 4)
+    import torch.distributed as dist
+    dist.barrier()  # inv1 fully checkpointed everywhere
+    n_gen1 = len([f for f in os.listdir(ck) if f.startswith("gen1-")])
+    r2 = sess.run(fv2, r1)
+    rows = sorted(r2.scan())
+    n_gen1_after = len([f for f in os.listdir(ck)
+                        if f.startswith("gen1-")])
+    q.put((rank, rows, n_gen1, n_gen1_after, sess.executor.comm.world))
+    if dist.is_initialized():
+        try:
+            dist.barrier()
+        except Exception:
+            pass
+        dist.destroy_process_group()
+
+
+def test_rank_loss_adopts_checkpointed_partitions():
+    world = 4
+    shutil.rmtree(CKPT + "_adopt", ignore_errors=True)
+    os.makedirs(CKPT + "_adopt")
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_adoption_worker,
+                         args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world - 1):
+        rank, rows, n1, n1b, w = q.get()
+        results[rank] = (rows, n1, n1b, w)
+    for p in procs:
+        p.join(120)
+    assert procs[world - 1].exitcode == 17
+    rows, n1, n1b, w = results[0]
+    expect = {}
+    for shard in range(4):
+        for k in (torch.arange(100) % 9).tolist():
+            expect[k] = expect.get(k, 0) + (shard + 1)
+    assert rows == sorted((k, v * 10) for k, v in expect.items())
+    assert w == world - 1
+    # the victim's invocation-1 shard was ADOPTED, not recomputed
+    assert n1b == n1, (n1, n1b)
+
+
 def test_rank_loss_mid_job_recovers():
     world = 4
     shutil.rmtree(CKPT, ignore_errors=True)
