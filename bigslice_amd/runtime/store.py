@@ -133,47 +133,8 @@ class MemoryStore(Store):
             raise KeyError(f"no output for {task_name} p{partition}")
         frames, _ = entry
         if isinstance(frames, _DiskEntry):
-            path = frames.path
-
-            class _D(Reader):
-                def __init__(self):
-                    self._fp = open(path, "rb")
-                    self._size = os.path.getsize(path)
-
-                def read(self):
-                    if self._fp.tell() >= self._size:
-                        return None
-                    return codec.decode_frame(self._fp, device)
-
-                def close(self):
-                    self._fp.close()
-            return _D()
-
-        class _R(Reader):
-            def __init__(self):
-                self.i = 0
-
-            def read(self):
-                if self.i >= len(frames):
-                    return None
-                f = frames[self.i]
-                self.i += 1
-                if f.device != "cpu" and _EVENT_MODE and \
-                        torch.cuda.is_available():
-                    # cross-stream consumers: keep the caching
-                    # allocator from reusing these blocks until the
-                    # reading stream passes this point (the producer
-                    # recorded its event; ordering is via wait_event
-                    # in the executor)
-                    cur = torch.cuda.current_stream()
-                    for c in f.columns:
-                        if isinstance(c, torch.Tensor) and c.is_cuda:
-                            c.record_stream(cur)
-                if device != "cpu" and f.device != device and \
-                        not f.has_objects:
-                    f = f.to(device, non_blocking=True)
-                return f
-        return _R()
+            return _DiskReader(frames.path, device)
+        return _FrameListReader(frames, device)
 
     def stat(self, task_name, partition):
         with self._lock:
@@ -199,6 +160,48 @@ class MemoryStore(Store):
                         os.unlink(frames.path)
                     except OSError:
                         pass
+
+
+class _DiskReader(Reader):
+    def __init__(self, path, device):
+        self._fp = open(path, "rb")
+        self._size = os.path.getsize(path)
+        self._device = device
+
+    def read(self):
+        if self._fp.tell() >= self._size:
+            return None
+        return codec.decode_frame(self._fp, self._device)
+
+    def close(self):
+        self._fp.close()
+
+
+class _FrameListReader(Reader):
+    def __init__(self, frames, device):
+        self.frames = frames
+        self.device = device
+        self.i = 0
+
+    def read(self):
+        if self.i >= len(self.frames):
+            return None
+        f = self.frames[self.i]
+        self.i += 1
+        if f.device != "cpu" and _EVENT_MODE and \
+                torch.cuda.is_available():
+            # cross-stream consumers: keep the caching allocator from
+            # reusing these blocks until the reading stream passes
+            # this point (the producer recorded its event; ordering is
+            # via wait_event in the executor)
+            cur = torch.cuda.current_stream()
+            for c in f.columns:
+                if isinstance(c, torch.Tensor) and c.is_cuda:
+                    c.record_stream(cur)
+        if self.device != "cpu" and f.device != self.device and \
+                not f.has_objects:
+            f = f.to(self.device, non_blocking=True)
+        return f
 
 
 class FileStore(Store):
