@@ -535,7 +535,8 @@ static std::vector<int> probe_passes(const torch::Tensor& keys) {
   auto andor = torch::empty({2}, i64);
   andor[0] = -1;  // all ones
   andor[1] = 0;
-  const int64_t nb = std::min<int64_t>((n + 255) / 256, 4096);
+  // fill the chip: 256 CUs x 8 workgroups, grid-stride the rest
+  const int64_t nb = std::min<int64_t>((n + 2047) / 2048, 2048);
   hipLaunchKernelGGL(
       (k_radix_andor<K>), dim3((int)nb), dim3(256), 0, stream,
       (const K*)keys.data_ptr(), n,

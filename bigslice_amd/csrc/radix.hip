@@ -56,19 +56,22 @@ __global__ void k_radix_andor(const K* __restrict__ keys, int64_t n,
                               unsigned long long* __restrict__ out_or) {
   using U = std::conditional_t<sizeof(K) == 8, uint64_t, uint32_t>;
   U a = ~(U)0, o = 0;
-  // 4 independent strided loads per iteration: the reduction is pure
-  // bandwidth, so keep enough loads in flight per lane
-  const int64_t stride4 = (int64_t)gridDim.x * blockDim.x * 4;
-  const int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  // 8 independent strided loads per iteration: the reduction is pure
+  // bandwidth, so keep many loads in flight per lane
   const int64_t step = (int64_t)gridDim.x * blockDim.x;
-  int64_t i = i0;
-  for (; i + 3 * step < n; i += stride4) {
-    const U u0 = (U)keys[i];
-    const U u1 = (U)keys[i + step];
-    const U u2 = (U)keys[i + 2 * step];
-    const U u3 = (U)keys[i + 3 * step];
-    a &= (u0 & u1) & (u2 & u3);
-    o |= (u0 | u1) | (u2 | u3);
+  const int64_t stride8 = step * 8;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i + 7 * step < n; i += stride8) {
+    U u0 = (U)keys[i];
+    U u1 = (U)keys[i + step];
+    U u2 = (U)keys[i + 2 * step];
+    U u3 = (U)keys[i + 3 * step];
+    U u4 = (U)keys[i + 4 * step];
+    U u5 = (U)keys[i + 5 * step];
+    U u6 = (U)keys[i + 6 * step];
+    U u7 = (U)keys[i + 7 * step];
+    a &= (u0 & u1) & (u2 & u3) & (u4 & u5) & (u6 & u7);
+    o |= (u0 | u1) | (u2 | u3) | (u4 | u5) | (u6 | u7);
   }
   for (; i < n; i += step) {
     const U u = (U)keys[i];
