@@ -104,6 +104,35 @@ def _local_worker(q):
     q.put(("local", run_all(sess, bs)))
 
 
+def _local_inline_worker(q):
+    os.environ["BIGSLICE_INLINE"] = "1"
+    import bigslice_amd as bs
+    sess = bs.start(parallelism=4, device="cpu")
+    assert sess.executor.inline
+    q.put(("inline", run_all(sess, bs)))
+
+
+def test_inline_executor_matches_pool():
+    """The inline execution mode (GPU default) must agree with the
+    pool on every fixture."""
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    out = {}
+    for target in (_local_worker, _local_inline_worker):
+        q = ctx.SimpleQueue()
+        p = ctx.Process(target=target, args=(q,))
+        p.start()
+        tag, rows = q.get()
+        p.join(120)
+        assert p.exitcode == 0
+        out[tag] = rows
+    for name in out["local"]:
+        if name == "head":
+            assert len(out["inline"][name]) == len(out["local"][name])
+            continue
+        assert out["inline"][name] == out["local"][name], name
+
+
 def _dist_worker(rank, world, port, q):
     _init(rank, world, port)
     import bigslice_amd as bs
