@@ -622,12 +622,15 @@ static bool use_hand_path(const std::vector<int>& passes) {
   return !contiguous;
 }
 
-torch::Tensor radix_sort_keys(torch::Tensor keys) {
+torch::Tensor radix_sort_keys(torch::Tensor keys, bool probe) {
   TORCH_CHECK(keys.is_cuda() && keys.is_contiguous());
   int64_t n = keys.size(0);
   auto keys_out = torch::empty_like(keys);
   int begin_bit = 0, end_bit = (int)keys.element_size() * 8;
-  if (n > 0 && probed_sortable(keys)) {
+  // probe=false: full-width rocPRIM with no byte-constancy probe (the
+  // probe's host readback is a stream sync — callers that must stay
+  // async, like the cardinality sample, opt out)
+  if (probe && n > 0 && probed_sortable(keys)) {
     std::vector<int> passes =
         keys.scalar_type() == torch::kInt64
             ? probe_passes<int64_t>(keys) : probe_passes<int32_t>(keys);
@@ -898,7 +901,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hash-aggregate table compaction (K9)");
   m.def("agg_identity", &agg_identity, "aggregation identity fill");
   m.def("radix_argsort", &radix_argsort, "device radix argsort (K6)");
-  m.def("radix_sort_keys", &radix_sort_keys, "device radix key sort");
+  m.def("radix_sort_keys", &radix_sort_keys, "device radix key sort",
+        py::arg("keys"), py::arg("probe") = true);
   m.def("segment_sum_sorted", &segment_sum_sorted_i64,
         "reduce-by-key sum over sorted int64 pairs (K16)");
   m.def("segment_reduce_sorted", &segment_reduce_sorted,

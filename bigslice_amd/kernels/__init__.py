@@ -207,7 +207,10 @@ class GroupTable:
             return
         self._sample_forced = None
         prefix = min(keys.shape[0], self._CARD_SAMPLE)
-        sk = _C.radix_sort_keys(keys[:prefix].contiguous())
+        # probe=False: the probe's count readback would sync the
+        # stream; the sample must stay fire-and-forget
+        sk = _C.radix_sort_keys(keys[:prefix].contiguous(),
+                                probe=False)
         self._sample_n = prefix
         self._sample_ne = (sk[1:] != sk[:-1]).sum()  # device scalar
 
