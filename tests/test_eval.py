@@ -133,6 +133,34 @@ def test_stress_random_loss():
     assert all(t.state == TaskState.OK for t in roots)
 
 
+def test_stress_random_loss_inline():
+    """The same convergence under the INLINE drain loop (the GPU
+    default): losses requeue on the evaluating thread with no pool."""
+    rng = random.Random(7)
+    producers = [make_task(f"ip{i}") for i in range(8)]
+    mids = [make_task(f"im{i}", [TaskDep(producers, i)])
+            for i in range(4)]
+    roots = [make_task(f"ir{i}", [TaskDep(mids, i)]) for i in range(2)]
+
+    class LossyInline(FakeExecutor):
+        inline = True
+
+        def run(self, task):
+            with self.lock:
+                self.runs[task.name] = self.runs.get(task.name, 0) + 1
+            task.set_state(TaskState.RUNNING)
+            if task.consecutive_lost < 2 and rng.random() < 0.35:
+                task.set_state(TaskState.LOST)
+            else:
+                task.set_state(TaskState.OK)
+
+    ex = LossyInline()
+    evaluate(ex, roots)
+    assert all(t.state == TaskState.OK for t in roots)
+    # everything ran on this thread: more runs than tasks (losses)
+    assert sum(ex.runs.values()) >= 14
+
+
 def test_concurrent_evals_share_tasks():
     # Two Evals over the same graph coordinate through task state.
     tasks = chain(3)
