@@ -141,7 +141,7 @@ def _dist_worker(rank, world, port, q):
     q.put((rank, out))
 
 
-@pytest.mark.parametrize("world", [2, 3])
+@pytest.mark.parametrize("world", [2, 3, 8])
 def test_dual_executor_matrix(world):
     import torch.multiprocessing as mp
     ctx = mp.get_context("spawn")
