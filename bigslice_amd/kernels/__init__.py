@@ -225,6 +225,7 @@ class GroupTable:
             return self._sample_forced
         distinct = 1 + int(self._sample_ne.item())
         s = self._sample_n
+        dbg = os.environ.get("BIGSLICE_GB_DEBUG")
         # Invert d = K(1-e^(-s/K)) to estimate the key-space size and
         # presize the table: the 10M-key hash insert measured 6.2 ms
         # properly sized vs 22 ms through the overflow-regrow grind.
@@ -236,6 +237,13 @@ class GroupTable:
             if self.cap_hint is None:
                 self.cap_hint = _next_pow2(
                     min(max(int(4 * K), 1024), 1 << 30))
+        if dbg:
+            import sys
+            mode = ("lds" if distinct * 32 <= s else
+                    "global" if distinct * 100 <= s * 88 else "sort")
+            print(f"[gb] sample={s} distinct={distinct} "
+                  f"cap_hint={self.cap_hint} mode={mode}",
+                  file=sys.stderr, flush=True)
         if distinct * 32 <= s:
             return "lds"
         if distinct * 100 <= s * 88:

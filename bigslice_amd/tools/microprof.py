@@ -21,16 +21,25 @@ from bigslice_amd.frame import Frame
 
 
 def timeit(fn, iters):
-    s = torch.cuda.Event(enable_timing=True)
-    e = torch.cuda.Event(enable_timing=True)
-    fn()  # warmup
+    """Median of per-iteration times (2 warmups).  Median, not mean:
+    on leases where a prior process ran, the FIRST touch of each
+    freshly-committed GPU allocation pays a one-time ~40-80 ms
+    driver-side cost (measured: iteration k hits it when the caching
+    allocator grows, all later iterations are clean) — a mean smears
+    that one-time cost into a phantom 3-4x 'regression'."""
+    import time
+    for _ in range(2):
+        fn()  # warmups absorb allocator-growth first-touch costs
     torch.cuda.synchronize()
-    s.record()
-    for _ in range(iters):
+    ts = []
+    for _ in range(max(iters, 3)):
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
         fn()
-    e.record()
-    torch.cuda.synchronize()
-    return s.elapsed_time(e) / iters
+        torch.cuda.synchronize()
+        ts.append((time.perf_counter() - t0) * 1000)
+    ts.sort()
+    return ts[len(ts) // 2]
 
 
 def main():
