@@ -441,7 +441,7 @@ static int radix_variant() {
 }
 
 template <typename K, int HAS_VAL, int TILE, int BLOCK, int SPLIT,
-          int PERSIST = 0, int NT = 0>
+          int PERSIST = 0, int NT = 0, int CHUNKED = 0, int WPS = 8>
 static void hand_radix_passes(const torch::Tensor& keys,
                               const torch::Tensor& vals,
                               torch::Tensor& keys_out,
@@ -496,7 +496,15 @@ static void hand_radix_passes(const torch::Tensor& keys,
                                RDX_RADIX * 8, stream));
     HIP_CHECK(hipMemsetAsync(state.data_ptr(), 0,
                              (size_t)ntiles * RDX_RADIX * 8, stream));
-    if (PERSIST) {
+    if (CHUNKED) {
+      hipLaunchKernelGGL(
+          (k_radix_scatter_chunked<K, HAS_VAL, TILE, BLOCK, WPS>),
+          dim3((int)ntiles), dim3(BLOCK), 0, stream, src_k, dst_k,
+          src_v, dst_v, n, passes[i] * 8, bias,
+          (const unsigned long long*)dbase.data_ptr<int64_t>(),
+          (unsigned long long*)state.data_ptr<int64_t>(), next_shift,
+          (unsigned long long*)hist_next.data_ptr<int64_t>());
+    } else if (PERSIST) {
       const int grid = (int)std::min<int64_t>(ntiles, 256);
       hipLaunchKernelGGL(
           (k_radix_scatter_persist<K, HAS_VAL, TILE, BLOCK>),
@@ -587,6 +595,16 @@ static void hand_radix_sort(const torch::Tensor& keys,
       break;
     case 7:
       hand_radix_passes<K, HAS_VAL, 8192, 1024, 0, 1>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 9:
+      // chunked reorder (rocPRIM-occupancy experiment, measured
+      // SLOWER: see k_radix_scatter_chunked's header comment)
+      hand_radix_passes<K, HAS_VAL, 4096, 512, 0, 0, 0, 1>(
+          keys, vals, keys_out, vals_out, passes, npass, bias, stream);
+      break;
+    case 10:
+      hand_radix_passes<K, HAS_VAL, 8192, 1024, 0, 0, 0, 1>(
           keys, vals, keys_out, vals_out, passes, npass, bias, stream);
       break;
     case 0:

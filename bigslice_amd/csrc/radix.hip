@@ -655,3 +655,247 @@ __launch_bounds__(BLOCK, 1) __global__ void k_radix_scatter_persist(
     if (s) atomicAdd(&next_hist[threadIdx.x], (unsigned long long)s);
   }
 }
+
+// Chunked-reorder variant (MEASURED SLOWER — kept as the record of a
+// tested hypothesis, variants 9/10): instead of staging the WHOLE
+// tile's pairs in LDS (150 KiB -> one workgroup per CU), slots stream
+// through a BLOCK-sized LDS window in TILE/BLOCK chunks (conditional
+// writes into the window, one coalesced read per thread per chunk).
+// LDS drops to ~30 KiB so 2-4 workgroups co-reside (up to 32
+// waves/CU) — the occupancy structure of rocPRIM's onesweep.
+// Outcome on 500 M int64 kv pairs (profiles/sort_variants.txt):
+// 21-22 ms (20-bit) / 54-58 ms (full) vs the monolithic SPLIT
+// default's 15.6 / 41.2 ms — the chunked window makes every thread
+// re-scan all IPT items per chunk (IPT^2 = 64 conditional LDS writes
+// instead of IPT) and pays 2*IPT barriers per tile, which costs more
+// than the 2-4x occupancy buys on this bandwidth-bound kernel; a
+// 6-waves/SIMD spill-free build was slower still (58-61 ms).  So
+// rocPRIM's uniform-key edge is NOT its occupancy; the production
+// dispatch (probe -> rocPRIM for contiguous byte sets, hand kernel
+// for interior constant bytes) stands.
+template <typename K, int HAS_VAL, int TILE, int BLOCK, int WPS = 8>
+__launch_bounds__(BLOCK, WPS) __global__
+void k_radix_scatter_chunked(
+    const K* __restrict__ keys_in, K* __restrict__ keys_out,
+    const int64_t* __restrict__ vals_in, int64_t* __restrict__ vals_out,
+    int64_t n, int shift, uint64_t bias,
+    const unsigned long long* __restrict__ digit_base,
+    unsigned long long* __restrict__ state, int next_shift,
+    unsigned long long* __restrict__ next_hist) {
+  using U = std::conditional_t<sizeof(K) == 8, uint64_t, uint32_t>;
+  constexpr int WAVES = BLOCK / 64;
+  constexpr int IPT = TILE / BLOCK;
+  __shared__ __align__(16) int64_t lds_win[BLOCK + 1];  // +1 spill slot
+  __shared__ unsigned short wavehist[WAVES][RDX_RADIX];
+  __shared__ unsigned short wavehist2[WAVES][RDX_RADIX];
+  __shared__ unsigned int digit_start[RDX_RADIX];
+  __shared__ unsigned int wave_tot[WAVES > 4 ? WAVES : 4];
+  __shared__ unsigned long long win_base[RDX_RADIX];
+
+  const int tile = blockIdx.x;
+  const int64_t base = (int64_t)tile * TILE;
+  const int64_t rem = n - base;
+  const int cnt = rem < TILE ? (int)rem : TILE;
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const uint64_t lt = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  const int wrow0 = wave * (64 * IPT);
+
+  for (int i = threadIdx.x; i < WAVES * RDX_RADIX; i += BLOCK) {
+    ((unsigned short*)wavehist)[i] = 0;
+    ((unsigned short*)wavehist2)[i] = 0;
+  }
+  __syncthreads();
+
+  // ---- burst loads + stable ballot-match rank + next-pass tally ----
+  K my_keys[IPT];
+  unsigned short my_rank[IPT];
+#pragma unroll
+  for (int i = 0; i < IPT; ++i) {
+    const int r = wrow0 + i * 64 + lane;
+    my_keys[i] = (r < cnt) ? keys_in[base + r] : K(0);
+  }
+#pragma unroll
+  for (int i = 0; i < IPT; ++i) {
+    const int r = wrow0 + i * 64 + lane;
+    const bool active = r < cnt;
+    const K k = my_keys[i];
+    const unsigned int d =
+        (unsigned int)((((U)k ^ (U)bias) >> shift) & 0xFF);
+    uint64_t peers = __ballot(active);
+    for (int b = 0; b < 8; ++b) {
+      const uint64_t bb = __ballot((d >> b) & 1);
+      peers &= ((d >> b) & 1) ? bb : ~bb;
+    }
+    const unsigned int before = wavehist[wave][d];
+    if (active && (peers & lt) == 0)
+      wavehist[wave][d] =
+          (unsigned short)(before + (unsigned int)__popcll(peers));
+    my_rank[i] =
+        (unsigned short)(before + (unsigned int)__popcll(peers & lt));
+    if (next_shift >= 0) {
+      const unsigned int d2 = (unsigned int)(((U)k >> next_shift) &
+                                             0xFF);
+      uint64_t p2 = __ballot(active);
+      for (int b = 0; b < 8; ++b) {
+        const uint64_t bb = __ballot((d2 >> b) & 1);
+        p2 &= ((d2 >> b) & 1) ? bb : ~bb;
+      }
+      if (active && (p2 & lt) == 0)
+        wavehist2[wave][d2] = (unsigned short)(
+            wavehist2[wave][d2] + (unsigned int)__popcll(p2));
+    }
+  }
+  __syncthreads();
+
+  // ---- digit scan + early publish ----
+  unsigned int my_total = 0;
+  if (threadIdx.x < RDX_RADIX) {
+    const int d = threadIdx.x;
+    unsigned int sum = 0;
+#pragma unroll
+    for (int w = 0; w < WAVES; ++w) {
+      const unsigned int c = wavehist[w][d];
+      wavehist[w][d] = (unsigned short)sum;
+      sum += c;
+    }
+    my_total = sum;
+    const unsigned long long pub =
+        ((unsigned long long)(tile == 0 ? RDX_FLAG_PREFIX : RDX_FLAG_AGG)
+         << 62) | (unsigned long long)sum;
+    __hip_atomic_store(&state[(uint64_t)tile * RDX_RADIX + d], pub,
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    unsigned int incl = sum;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      const unsigned int up = __shfl_up(incl, off, 64);
+      if (lane >= off) incl += up;
+    }
+    digit_start[d] = incl - sum;
+    if (lane == 63) wave_tot[d / 64] = incl;
+  }
+  __syncthreads();
+  if (threadIdx.x < RDX_RADIX) {
+    unsigned int carry = 0;
+    const int sw = threadIdx.x / 64;
+    for (int w = 0; w < sw; ++w) carry += wave_tot[w];
+    digit_start[threadIdx.x] += carry;
+  }
+  __syncthreads();
+
+  // full tile rank (position within the tile's digit-major order)
+#pragma unroll
+  for (int i = 0; i < IPT; ++i) {
+    const int r = wrow0 + i * 64 + lane;
+    if (r < cnt) {
+      const unsigned int d =
+          (unsigned int)((((U)my_keys[i] ^ (U)bias) >> shift) & 0xFF);
+      my_rank[i] = (unsigned short)(
+          digit_start[d] + wavehist[wave][d] + my_rank[i]);
+    } else {
+      my_rank[i] = (unsigned short)TILE;  // parked outside any window
+    }
+  }
+
+  // ---- lookback: global exclusive tile prefix per digit ----
+  unsigned long long lb_excl = 0;
+  if (threadIdx.x < RDX_RADIX && tile > 0) {
+    const int d = threadIdx.x;
+    int j = tile - 1;
+    bool done = false;
+    while (!done) {
+      unsigned long long v[RDX_LOOKBACK_BATCH];
+      const int m = (j + 1) < RDX_LOOKBACK_BATCH ? (j + 1)
+                                                 : RDX_LOOKBACK_BATCH;
+#pragma unroll
+      for (int q = 0; q < RDX_LOOKBACK_BATCH; ++q) {
+        if (q < m)
+          v[q] = __hip_atomic_load(
+              &state[(uint64_t)(j - q) * RDX_RADIX + d],
+              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      }
+      for (int q = 0; q < m; ++q) {
+        unsigned long long x = v[q];
+        while ((x >> 62) == 0) {
+          __builtin_amdgcn_s_sleep(1);
+          x = __hip_atomic_load(
+              &state[(uint64_t)(j - q) * RDX_RADIX + d],
+              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+        lb_excl += x & ((1ull << 62) - 1);
+        if ((x >> 62) == RDX_FLAG_PREFIX) {
+          done = true;
+          break;
+        }
+      }
+      j -= m;
+      if (j < 0) done = true;
+    }
+    __hip_atomic_store(
+        &state[(uint64_t)tile * RDX_RADIX + d],
+        (RDX_FLAG_PREFIX << 62) | (lb_excl + my_total),
+        __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+  if (threadIdx.x < RDX_RADIX)
+    // store (global digit base - tile digit start): slot j of digit d
+    // lands at win_base[d] + j directly
+    win_base[threadIdx.x] = digit_base[threadIdx.x] + lb_excl -
+                            digit_start[threadIdx.x];
+  __syncthreads();
+
+  // ---- chunked key reorder through the BLOCK-sized LDS window ----
+  unsigned char my_dig2[IPT];  // digit per item, for the value pass
+#pragma unroll
+  for (int j = 0; j < IPT; ++j) {
+    const unsigned int x = j * BLOCK;
+#pragma unroll
+    for (int i = 0; i < IPT; ++i) {
+      // clamped write: ranks outside this window hit the spill slot
+      const unsigned int off = (unsigned int)my_rank[i] - x;
+      lds_win[off < BLOCK ? off : BLOCK] = (int64_t)my_keys[i];
+    }
+    __syncthreads();
+    const unsigned int slot = x + threadIdx.x;
+    if ((int)slot < cnt) {
+      const K k = (K)lds_win[threadIdx.x];
+      const unsigned int d =
+          (unsigned int)((((U)k ^ (U)bias) >> shift) & 0xFF);
+      keys_out[win_base[d] + slot] = k;
+      if (HAS_VAL) my_dig2[j] = (unsigned char)d;
+    }
+    __syncthreads();
+  }
+
+  // ---- value pass through the same window ----
+  if (HAS_VAL) {
+    int64_t my_vals[IPT];
+#pragma unroll
+    for (int i = 0; i < IPT; ++i) {
+      const int r = wrow0 + i * 64 + lane;
+      my_vals[i] = (r < cnt) ? vals_in[base + r] : 0;
+    }
+#pragma unroll
+    for (int j = 0; j < IPT; ++j) {
+      const unsigned int x = j * BLOCK;
+#pragma unroll
+      for (int i = 0; i < IPT; ++i) {
+        const unsigned int off = (unsigned int)my_rank[i] - x;
+        lds_win[off < BLOCK ? off : BLOCK] = my_vals[i];
+      }
+      __syncthreads();
+      const unsigned int slot = x + threadIdx.x;
+      if ((int)slot < cnt) {
+        vals_out[win_base[my_dig2[j]] + slot] = lds_win[threadIdx.x];
+      }
+      __syncthreads();
+    }
+  }
+
+  // flush the next pass's digit tallies
+  if (next_shift >= 0 && threadIdx.x < RDX_RADIX) {
+    unsigned int s = 0;
+#pragma unroll
+    for (int w = 0; w < WAVES; ++w) s += wavehist2[w][threadIdx.x];
+    if (s) atomicAdd(&next_hist[threadIdx.x], (unsigned long long)s);
+  }
+}
