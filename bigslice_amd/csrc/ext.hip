@@ -543,8 +543,9 @@ static std::vector<int> probe_passes(const torch::Tensor& keys) {
   auto andor = torch::empty({2}, i64);
   andor[0] = -1;  // all ones
   andor[1] = 0;
-  // fill the chip: 256 CUs x 8 workgroups, grid-stride the rest
-  const int64_t nb = std::min<int64_t>((n + 2047) / 2048, 2048);
+  // 512 grid-strided blocks saturate HBM with the 8-deep load burst
+  // while keeping the same-address atomic tail at ~1 op/block
+  const int64_t nb = std::min<int64_t>((n + 2047) / 2048, 512);
   hipLaunchKernelGGL(
       (k_radix_andor<K>), dim3((int)nb), dim3(256), 0, stream,
       (const K*)keys.data_ptr(), n,
@@ -789,6 +790,32 @@ std::vector<torch::Tensor> segment_sum_sorted_i64(torch::Tensor keys,
   return segment_reduce_sorted(keys, vals, 0);
 }
 
+// K18: (unique_keys, run_starts, count) of a SORTED key array in one
+// pass (sort.hip: reduce-by-key over a counting iterator).  count is a
+// 1-element device tensor; the caller slices after one sync.
+std::vector<torch::Tensor> runs_sorted(torch::Tensor keys) {
+  TORCH_CHECK(keys.is_cuda() && keys.is_contiguous() &&
+              keys.scalar_type() == torch::kInt64,
+              "runs_sorted: contiguous int64 device keys required");
+  const int64_t n = keys.size(0);
+  auto uniq = torch::empty_like(keys);
+  auto starts = torch::empty_like(keys);
+  auto count = torch::zeros({1}, keys.options());
+  if (n == 0) return {uniq, starts, count};
+  size_t temp_bytes = 0;
+  runs_sorted_i64(keys.data_ptr<int64_t>(), n,
+                  uniq.data_ptr<int64_t>(), starts.data_ptr<int64_t>(),
+                  count.data_ptr<int64_t>(), nullptr, temp_bytes,
+                  current_stream());
+  auto temp = torch::empty({(int64_t)temp_bytes},
+                           keys.options().dtype(torch::kUInt8));
+  runs_sorted_i64(keys.data_ptr<int64_t>(), n,
+                  uniq.data_ptr<int64_t>(), starts.data_ptr<int64_t>(),
+                  count.data_ptr<int64_t>(), temp.data_ptr(), temp_bytes,
+                  current_stream());
+  return {uniq, starts, count};
+}
+
 // K17: murmur3 over variable-length byte rows (device strings).
 torch::Tensor hash_bytes(torch::Tensor bytes, torch::Tensor offsets,
                          int64_t seed) {
@@ -923,6 +950,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("keys"), py::arg("probe") = true);
   m.def("segment_sum_sorted", &segment_sum_sorted_i64,
         "reduce-by-key sum over sorted int64 pairs (K16)");
+  m.def("runs_sorted", &runs_sorted,
+        "run boundaries of sorted keys (K18)");
   m.def("segment_reduce_sorted", &segment_reduce_sorted,
         "typed reduce-by-key over sorted pairs (K16; deterministic)");
   m.def("hash_bytes", &hash_bytes,

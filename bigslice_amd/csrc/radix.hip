@@ -82,10 +82,27 @@ __global__ void k_radix_andor(const K* __restrict__ keys, int64_t n,
     a &= (U)__shfl_xor((unsigned long long)a, off, 64);
     o |= (U)__shfl_xor((unsigned long long)o, off, 64);
   }
+  // Block-level LDS reduction -> ONE atomic pair per block.  The
+  // earlier per-wave form issued 2 contended same-address atomics per
+  // wave; at a 2048-block grid that tail alone ran ~0.4 ms (the whole
+  // 0.5 GB read should take ~70 us), measured as 1.1 TB/s on a pure
+  // streaming reduction.
+  __shared__ unsigned long long red_a[8], red_o[8];
+  const int wv = threadIdx.x / 64;
   if ((threadIdx.x % 64) == 0) {
-    atomicAnd(out_and, (unsigned long long)a |
-                           ~(unsigned long long)(U)~(U)0);
-    atomicOr(out_or, (unsigned long long)o);
+    red_a[wv] = (unsigned long long)a |
+                ~(unsigned long long)(U)~(U)0;
+    red_o[wv] = (unsigned long long)o;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    unsigned long long ba = red_a[0], bo = red_o[0];
+    for (int w = 1; w < (int)(blockDim.x / 64); ++w) {
+      ba &= red_a[w];
+      bo |= red_o[w];
+    }
+    atomicAnd(out_and, ba);
+    atomicOr(out_or, bo);
   }
 }
 

@@ -114,3 +114,24 @@ INSTANTIATE_RBK(int64_t)
 INSTANTIATE_RBK(int32_t)
 INSTANTIATE_RBK(float)
 INSTANTIATE_RBK(double)
+
+// K18 (cogroup assembly): run boundaries of a SORTED key array in ONE
+// rocPRIM reduce-by-key pass — values are a counting iterator and the
+// reduction is min, so each run's aggregate IS its start index.
+// Replaces the ne-mask -> nonzero -> gather chain (3 reads of the key
+// array + a bool mask round trip + rocPRIM partition) with a single
+// 8 B/row read.
+#include <rocprim/iterator/counting_iterator.hpp>
+
+void runs_sorted_i64(const int64_t* keys, int64_t n, int64_t* uniq_out,
+                     int64_t* starts_out, int64_t* count_out, void* temp,
+                     size_t& temp_bytes, hipStream_t s) {
+  hipError_t err = rocprim::reduce_by_key(
+      temp, temp_bytes, keys,
+      rocprim::counting_iterator<int64_t>(0), (size_t)n, uniq_out,
+      starts_out, count_out, rocprim::minimum<int64_t>(),
+      rocprim::equal_to<int64_t>(), s);
+  if (err != hipSuccess)
+    throw std::runtime_error(std::string("rocprim::runs_sorted: ") +
+                             hipGetErrorString(err));
+}

@@ -308,6 +308,25 @@ def test_radix_sort_reduced_bits(kernels):
         assert torch.equal(kernels.radix_sort_keys(k), ref_k), (lo, hi)
 
 
+def test_runs_sorted_matches_fallback(kernels):
+    # K18 one-pass run boundaries vs the diff-mask chain, including
+    # all-equal, all-distinct and negative keys
+    from bigslice_amd.ops.cogroup import _runs
+    for keys in (
+            torch.randint(0, 97, (1_000_003,), dtype=torch.int64),
+            torch.zeros(4097, dtype=torch.int64),
+            torch.arange(5000, dtype=torch.int64),
+            torch.randint(-5, 5, (300_000,), dtype=torch.int64),
+            torch.empty(0, dtype=torch.int64)):
+        sk = torch.sort(keys).values.to("cuda:0")
+        uniq, starts, ends = _runs(sk)
+        # fallback path on the same data (force via CPU then move)
+        m_uniq, m_starts, m_ends = _runs(sk.cpu())
+        assert torch.equal(uniq.cpu(), m_uniq)
+        assert torch.equal(starts.cpu(), m_starts)
+        assert torch.equal(ends.cpu(), m_ends)
+
+
 def test_grouptable_sort_combine_paths(kernels):
     # Force both combine paths and check they agree with torch.
     import os
