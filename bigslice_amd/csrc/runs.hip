@@ -43,15 +43,17 @@ __global__ __launch_bounds__(RUNS_BLOCK) void k_runs_sorted(
   // registers: boundaries are ~1 row in 50+, so pass B re-reads
   // keys[r] for flagged lanes only (a few MB) instead of pinning
   // IPT int64 registers per thread.
+  // Consecutive lanes hold consecutive rows, so keys[r-1] is a wave
+  // shuffle: only lane 0 issues a second load (1/64 of rows).
   unsigned int my_flags = 0;  // bit k = row base + k*BLOCK + tid
 #pragma unroll
   for (int k = 0; k < RUNS_IPT; ++k) {
     const int64_t r = base + (int64_t)k * RUNS_BLOCK + threadIdx.x;
-    bool flag = false;
-    if (r < n) {
-      const int64_t kv = keys[r];
-      flag = (r == 0) || (kv != keys[r - 1]);
-    }
+    const bool in = r < n;
+    const int64_t kv = in ? keys[r] : 0;
+    int64_t kprev = __shfl_up(kv, 1, 64);
+    if (lane == 0 && in && r > 0) kprev = keys[r - 1];
+    const bool flag = in && ((r == 0) || (kv != kprev));
     my_flags |= flag ? (1u << k) : 0u;
     const uint64_t b = __ballot(flag);
     if (lane == 0) cnt[k * WAVES + wave] = (unsigned int)__popcll(b);
