@@ -25,7 +25,9 @@
 #define RUNS_FLAG_PREFIX 2ull
 #define RUNS_LOOKBACK_BATCH 8
 
-__global__ __launch_bounds__(RUNS_BLOCK) void k_runs_sorted(
+// 4 waves/SIMD floor (<=128 VGPRs): the unrolled slab loops otherwise
+// hold 175 VGPRs of live addresses and occupancy drops to 8 waves/CU.
+__global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
     const int64_t* __restrict__ keys, int64_t n,
     int64_t* __restrict__ uniq_out, int64_t* __restrict__ starts_out,
     int64_t* __restrict__ count_out,
@@ -44,19 +46,33 @@ __global__ __launch_bounds__(RUNS_BLOCK) void k_runs_sorted(
   // keys[r] for flagged lanes only (a few MB) instead of pinning
   // IPT int64 registers per thread.
   // Consecutive lanes hold consecutive rows, so keys[r-1] is a wave
-  // shuffle: only lane 0 issues a second load (1/64 of rows).
-  unsigned int my_flags = 0;  // bit k = row base + k*BLOCK + tid
+  // shuffle; lane 0's real neighbor load is issued inside the same
+  // burst.  Loads burst 8-deep before the shuffle/ballot chain — the
+  // interleaved form serializes every load behind a waitcnt (same
+  // fix as the radix scatter's burst loads).
+  uint64_t my_flags = 0;  // bit k = row base + k*BLOCK + tid
 #pragma unroll
-  for (int k = 0; k < RUNS_IPT; ++k) {
-    const int64_t r = base + (int64_t)k * RUNS_BLOCK + threadIdx.x;
-    const bool in = r < n;
-    const int64_t kv = in ? keys[r] : 0;
-    int64_t kprev = __shfl_up(kv, 1, 64);
-    if (lane == 0 && in && r > 0) kprev = keys[r - 1];
-    const bool flag = in && ((r == 0) || (kv != kprev));
-    my_flags |= flag ? (1u << k) : 0u;
-    const uint64_t b = __ballot(flag);
-    if (lane == 0) cnt[k * WAVES + wave] = (unsigned int)__popcll(b);
+  for (int k0 = 0; k0 < RUNS_IPT; k0 += 8) {
+    int64_t kvb[8], kpb[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int64_t r =
+          base + (int64_t)(k0 + j) * RUNS_BLOCK + threadIdx.x;
+      kvb[j] = (r < n) ? keys[r] : 0;
+      if (lane == 0) kpb[j] = (r > 0 && r < n) ? keys[r - 1] : 0;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = k0 + j;
+      const int64_t r = base + (int64_t)k * RUNS_BLOCK + threadIdx.x;
+      const bool in = r < n;
+      int64_t kprev = __shfl_up(kvb[j], 1, 64);
+      if (lane == 0) kprev = kpb[j];
+      const bool flag = in && ((r == 0) || (kvb[j] != kprev));
+      my_flags |= flag ? (1ull << k) : 0ull;
+      const uint64_t b = __ballot(flag);
+      if (lane == 0) cnt[k * WAVES + wave] = (unsigned int)__popcll(b);
+    }
   }
   __syncthreads();
   // exclusive scan of the 2*WAVES..IPT*WAVES counters in (k, wave)
