@@ -34,6 +34,7 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
     unsigned long long* __restrict__ state) {
   constexpr int WAVES = RUNS_BLOCK / 64;
   __shared__ unsigned int cnt[RUNS_IPT * WAVES];  // slab-major
+  __shared__ unsigned int wtot[RUNS_BLOCK / 64];  // per-wave totals
   __shared__ unsigned long long lds_base[1];
   const int tile = blockIdx.x;
   const int64_t base = (int64_t)tile * RUNS_TILE;
@@ -51,6 +52,7 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
   // interleaved form serializes every load behind a waitcnt (same
   // fix as the radix scatter's burst loads).
   uint64_t my_flags = 0;  // bit k = row base + k*BLOCK + tid
+  unsigned int my_wtot = 0;
 #pragma unroll
   for (int k0 = 0; k0 < RUNS_IPT; k0 += 8) {
     int64_t kvb[8], kpb[8];
@@ -71,63 +73,95 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
       const bool flag = in && ((r == 0) || (kvb[j] != kprev));
       my_flags |= flag ? (1ull << k) : 0ull;
       const uint64_t b = __ballot(flag);
-      if (lane == 0) cnt[k * WAVES + wave] = (unsigned int)__popcll(b);
+      if (lane == 0) {
+        const unsigned int c = (unsigned int)__popcll(b);
+        cnt[k * WAVES + wave] = c;
+        my_wtot += c;
+      }
     }
   }
+  if (lane == 0) wtot[wave] = my_wtot;
   __syncthreads();
-  // exclusive scan of the 2*WAVES..IPT*WAVES counters in (k, wave)
-  // order by one lane (tiny: IPT*WAVES adds)
-  if (threadIdx.x == 0) {
-    unsigned int run = 0;
-    for (int i = 0; i < RUNS_IPT * WAVES; ++i) {
-      const unsigned int v = cnt[i];
-      cnt[i] = run;
-      run += v;
+  // Split the inter-tile critical path across waves: wave 0 reduces
+  // the counters (7 shuffle steps), publishes AGG IMMEDIATELY, then
+  // walks 64 predecessors per round with lane-parallel loads; wave 1
+  // meanwhile builds the exclusive scan of the counters.  The serial
+  // form (one thread: 128-add scan, then a batched walk) left every
+  // successor tile spinning on a publish that sat behind the scan.
+  constexpr int NCNT = RUNS_IPT * WAVES;
+  unsigned int run = 0;
+  if (wave == 0) {
+    // total from the per-wave accumulators (NOT from cnt — wave 1 is
+    // overwriting cnt with its exclusive scan concurrently)
+    unsigned int s = (lane < WAVES) ? wtot[lane] : 0;
+    for (int off = 32; off; off >>= 1)
+      s += (unsigned int)__shfl_xor((int)s, off, 64);
+    run = s;  // wave-uniform total
+    if (lane == 0) {
+      const unsigned long long pub =
+          ((tile == 0 ? RUNS_FLAG_PREFIX : RUNS_FLAG_AGG) << 62) |
+          (unsigned long long)run;
+      __hip_atomic_store(&state[tile], pub, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
     }
-    // publish this tile's total, then look back for the prefix
-    const unsigned long long pub =
-        ((tile == 0 ? RUNS_FLAG_PREFIX : RUNS_FLAG_AGG) << 62) |
-        (unsigned long long)run;
-    __hip_atomic_store(&state[tile], pub, __ATOMIC_RELAXED,
-                       __HIP_MEMORY_SCOPE_AGENT);
     unsigned long long excl = 0;
     if (tile > 0) {
-      int j = tile - 1;
+      int j0 = tile - 1;
       bool done = false;
       while (!done) {
-        unsigned long long v[RUNS_LOOKBACK_BATCH];
-        const int m = (j + 1) < RUNS_LOOKBACK_BATCH
-                          ? (j + 1) : RUNS_LOOKBACK_BATCH;
-#pragma unroll
-        for (int q = 0; q < RUNS_LOOKBACK_BATCH; ++q) {
-          if (q < m)
-            v[q] = __hip_atomic_load(&state[j - q], __ATOMIC_RELAXED,
-                                     __HIP_MEMORY_SCOPE_AGENT);
-        }
-        for (int q = 0; q < m; ++q) {
-          unsigned long long x = v[q];
+        const int idx = j0 - lane;
+        unsigned long long x = 0;
+        if (idx >= 0) {
+          x = __hip_atomic_load(&state[idx], __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
           while ((x >> 62) == 0) {
             __builtin_amdgcn_s_sleep(1);
-            x = __hip_atomic_load(&state[j - q], __ATOMIC_RELAXED,
+            x = __hip_atomic_load(&state[idx], __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_AGENT);
           }
-          excl += x & ((1ull << 62) - 1);
-          if ((x >> 62) == RUNS_FLAG_PREFIX) {
-            done = true;
-            break;
-          }
         }
-        j -= m;
-        if (j < 0) done = true;
+        const uint64_t pmask =
+            __ballot(idx >= 0 && (x >> 62) == RUNS_FLAG_PREFIX);
+        unsigned long long c = x & ((1ull << 62) - 1);
+        if (pmask) {
+          const int lp = __ffsll((unsigned long long)pmask) - 1;
+          if (lane > lp) c = 0;
+          done = true;
+        } else if (idx < 0) {
+          c = 0;
+        }
+        for (int off = 32; off; off >>= 1)
+          c += __shfl_xor((unsigned long long)c, off, 64);
+        excl += c;
+        j0 -= 64;
+        if (j0 < 0) done = true;
       }
-      __hip_atomic_store(
-          &state[tile],
-          (RUNS_FLAG_PREFIX << 62) | (excl + (unsigned long long)run),
-          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      if (lane == 0)
+        __hip_atomic_store(
+            &state[tile],
+            (RUNS_FLAG_PREFIX << 62) | (excl + (unsigned long long)run),
+            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
-    lds_base[0] = excl;
-    if (tile == (int)gridDim.x - 1)
-      *count_out = (int64_t)(excl + run);
+    if (lane == 0) {
+      lds_base[0] = excl;
+      if (tile == (int)gridDim.x - 1)
+        *count_out = (int64_t)(excl + run);
+    }
+  } else if (wave == 1) {
+    // exclusive scan of cnt[NCNT] in place: shuffle scan per 64-chunk
+    // with a running carry
+    unsigned int carry = 0;
+    for (int i0 = 0; i0 < NCNT; i0 += 64) {
+      const unsigned int v = cnt[i0 + lane];
+      unsigned int incl = v;
+      for (int off = 1; off < 64; off <<= 1) {
+        const unsigned int up =
+            (unsigned int)__shfl_up((int)incl, off, 64);
+        if (lane >= off) incl += up;
+      }
+      cnt[i0 + lane] = carry + incl - v;
+      carry += (unsigned int)__shfl((int)incl, 63, 64);
+    }
   }
   __syncthreads();
   const unsigned long long tb = lds_base[0];
