@@ -821,18 +821,27 @@ std::vector<torch::Tensor> runs_sorted(torch::Tensor keys) {
     return {uniq, starts, count};
   }
   auto stream = current_stream();
-  const int64_t ntiles = (n + RUNS_TILE - 1) / RUNS_TILE;
+  // 64K-row tiles amortize per-block fixed costs on large inputs;
+  // 8K-row tiles keep small inputs on >=2 blocks/CU
+  const int ipt = (n >= (16 << 20)) ? 256 : 32;
+  const int64_t tile = (int64_t)RUNS_BLOCK * ipt;
+  const int64_t ntiles = (n + tile - 1) / tile;
   auto state = torch::empty({ntiles},
                             keys.options().dtype(torch::kInt64));
   HIP_CHECK(hipMemsetAsync(state.data_ptr(), 0, (size_t)ntiles * 8,
                            stream));
-  hipLaunchKernelGGL(k_runs_sorted, dim3((int)ntiles),
-                     dim3(RUNS_BLOCK), 0, stream,
-                     keys.data_ptr<int64_t>(), n,
-                     uniq.data_ptr<int64_t>(),
-                     starts.data_ptr<int64_t>(),
-                     count.data_ptr<int64_t>(),
-                     (unsigned long long*)state.data_ptr<int64_t>());
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3((int)ntiles), dim3(RUNS_BLOCK), 0,
+                       stream, keys.data_ptr<int64_t>(), n,
+                       uniq.data_ptr<int64_t>(),
+                       starts.data_ptr<int64_t>(),
+                       count.data_ptr<int64_t>(),
+                       (unsigned long long*)state.data_ptr<int64_t>());
+  };
+  if (ipt == 256)
+    launch(k_runs_sorted<256>);
+  else
+    launch(k_runs_sorted<32>);
   HIP_CHECK(hipGetLastError());
   return {uniq, starts, count};
 }

@@ -6,9 +6,20 @@
 // propagation moves key+aggregate pairs through the lookback): a
 // boundary flag here is just keys[r] != keys[r-1], so a tile's state
 // granule is ONE {flag|count} word and the compaction is ballot
-// arithmetic.  Geometry: 256-thread blocks, each thread IPT rows
-// strided by BLOCK (coalesced); ordering inside a tile is
-// (slab k, wave, lane), scanned in LDS once per tile.
+// arithmetic.
+//
+// Geometry notes (measured on the way here):
+// * consecutive lanes hold consecutive rows, so keys[r-1] is a wave
+//   shuffle; lane 0's real neighbor load rides in the same burst;
+// * loads burst 8-deep ahead of the shuffle/ballot chain (interleaved
+//   form serializes each load behind a waitcnt);
+// * per-block fixed costs (two barriers, the lookback's L2 round
+//   trips) were ~half the runtime of an 8K-row tile whose streaming
+//   part is ~2.5 us, capping the kernel at 1.4 TB/s.  Large inputs
+//   therefore run 64K-row tiles: the outer slab loop is NOT unrolled
+//   (a fully unrolled 64-slab loop held 175 VGPRs of live addresses)
+//   and pass B recomputes boundary flags from L2-warm keys instead of
+//   pinning a >64-bit flag register per thread.
 //
 // Reference parity: this is the assembly step of the cogroup reader
 // (cogroup.go:150-214 groups equal keys per dep); the sort-merge
@@ -19,42 +30,31 @@
 #include <cstdint>
 
 #define RUNS_BLOCK 256
-#define RUNS_IPT 32
-#define RUNS_TILE (RUNS_BLOCK * RUNS_IPT)
 #define RUNS_FLAG_AGG 1ull
 #define RUNS_FLAG_PREFIX 2ull
-#define RUNS_LOOKBACK_BATCH 8
 
-// 4 waves/SIMD floor (<=128 VGPRs): the unrolled slab loops otherwise
-// hold 175 VGPRs of live addresses and occupancy drops to 8 waves/CU.
+template <int IPT>
 __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
     const int64_t* __restrict__ keys, int64_t n,
     int64_t* __restrict__ uniq_out, int64_t* __restrict__ starts_out,
     int64_t* __restrict__ count_out,
     unsigned long long* __restrict__ state) {
   constexpr int WAVES = RUNS_BLOCK / 64;
-  __shared__ unsigned int cnt[RUNS_IPT * WAVES];  // slab-major
-  __shared__ unsigned int wtot[RUNS_BLOCK / 64];  // per-wave totals
+  constexpr int64_t TILE = (int64_t)RUNS_BLOCK * IPT;
+  __shared__ unsigned int cnt[IPT * WAVES];  // slab-major
+  __shared__ unsigned int wtot[WAVES];       // per-wave totals
+  __shared__ unsigned char fbits[IPT / 8][RUNS_BLOCK];  // flag bytes
   __shared__ unsigned long long lds_base[1];
   const int tile = blockIdx.x;
-  const int64_t base = (int64_t)tile * RUNS_TILE;
+  const int64_t base = (int64_t)tile * TILE;
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
   const uint64_t lt = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
 
-  // pass A: flags + per-(slab, wave) counts.  Keys are NOT kept in
-  // registers: boundaries are ~1 row in 50+, so pass B re-reads
-  // keys[r] for flagged lanes only (a few MB) instead of pinning
-  // IPT int64 registers per thread.
-  // Consecutive lanes hold consecutive rows, so keys[r-1] is a wave
-  // shuffle; lane 0's real neighbor load is issued inside the same
-  // burst.  Loads burst 8-deep before the shuffle/ballot chain — the
-  // interleaved form serializes every load behind a waitcnt (same
-  // fix as the radix scatter's burst loads).
-  uint64_t my_flags = 0;  // bit k = row base + k*BLOCK + tid
+  // pass A: boundary flags -> per-(slab, wave) counts
   unsigned int my_wtot = 0;
-#pragma unroll
-  for (int k0 = 0; k0 < RUNS_IPT; k0 += 8) {
+#pragma unroll 1
+  for (int k0 = 0; k0 < IPT; k0 += 8) {
     int64_t kvb[8], kpb[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -63,6 +63,7 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
       kvb[j] = (r < n) ? keys[r] : 0;
       if (lane == 0) kpb[j] = (r > 0 && r < n) ? keys[r - 1] : 0;
     }
+    unsigned int fb = 0;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int k = k0 + j;
@@ -71,7 +72,7 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
       int64_t kprev = __shfl_up(kvb[j], 1, 64);
       if (lane == 0) kprev = kpb[j];
       const bool flag = in && ((r == 0) || (kvb[j] != kprev));
-      my_flags |= flag ? (1ull << k) : 0ull;
+      fb |= flag ? (1u << j) : 0u;
       const uint64_t b = __ballot(flag);
       if (lane == 0) {
         const unsigned int c = (unsigned int)__popcll(b);
@@ -79,20 +80,19 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
         my_wtot += c;
       }
     }
+    fbits[k0 / 8][threadIdx.x] = (unsigned char)fb;
   }
   if (lane == 0) wtot[wave] = my_wtot;
   __syncthreads();
+
   // Split the inter-tile critical path across waves: wave 0 reduces
-  // the counters (7 shuffle steps), publishes AGG IMMEDIATELY, then
+  // the per-wave totals (7 shuffles), publishes AGG IMMEDIATELY, then
   // walks 64 predecessors per round with lane-parallel loads; wave 1
-  // meanwhile builds the exclusive scan of the counters.  The serial
-  // form (one thread: 128-add scan, then a batched walk) left every
-  // successor tile spinning on a publish that sat behind the scan.
-  constexpr int NCNT = RUNS_IPT * WAVES;
+  // meanwhile builds the exclusive scan of the slab counters (totals
+  // come from wtot, NOT cnt, which wave 1 is overwriting).
+  constexpr int NCNT = IPT * WAVES;
   unsigned int run = 0;
   if (wave == 0) {
-    // total from the per-wave accumulators (NOT from cnt — wave 1 is
-    // overwriting cnt with its exclusive scan concurrently)
     unsigned int s = (lane < WAVES) ? wtot[lane] : 0;
     for (int off = 32; off; off >>= 1)
       s += (unsigned int)__shfl_xor((int)s, off, 64);
@@ -166,17 +166,23 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
   __syncthreads();
   const unsigned long long tb = lds_base[0];
 
-  // pass B: compacted ordered writes
+  // pass B: flags from LDS, compacted ordered writes.  Only flagged
+  // lanes (~1 row in 50+) re-read their key from global.
+#pragma unroll 1
+  for (int k0 = 0; k0 < IPT; k0 += 8) {
+    const unsigned int fb = fbits[k0 / 8][threadIdx.x];
 #pragma unroll
-  for (int k = 0; k < RUNS_IPT; ++k) {
-    const bool flag = (my_flags >> k) & 1u;
-    const uint64_t b = __ballot(flag);
-    if (flag) {
-      const int64_t r = base + (int64_t)k * RUNS_BLOCK + threadIdx.x;
-      const unsigned int pos = cnt[k * WAVES + wave] +
-                               (unsigned int)__popcll(b & lt);
-      uniq_out[tb + pos] = keys[r];
-      starts_out[tb + pos] = r;
+    for (int j = 0; j < 8; ++j) {
+      const int k = k0 + j;
+      const bool flag = (fb >> j) & 1u;
+      const uint64_t b = __ballot(flag);
+      if (flag) {
+        const int64_t r = base + (int64_t)k * RUNS_BLOCK + threadIdx.x;
+        const unsigned int pos = cnt[k * WAVES + wave] +
+                                 (unsigned int)__popcll(b & lt);
+        uniq_out[tb + pos] = keys[r];
+        starts_out[tb + pos] = r;
+      }
     }
   }
 }
