@@ -47,7 +47,8 @@ def main():
     ap.add_argument("which", choices=["groupby", "groupby_pp",
                                       "insert", "insert_rep",
                                       "partition", "hash", "compact",
-                                      "sortcombine", "hashbytes"])
+                                      "sortcombine", "hashbytes",
+                                      "runs"])
     ap.add_argument("--nrep", type=int, default=8)
     ap.add_argument("--cap", type=int, default=0,
                     help="table capacity (0 = 2x nkeys rounded up)")
@@ -135,6 +136,13 @@ def main():
                 kernels._C.groupby_compact_packed(t.table, cursor)
             else:
                 kernels._C.groupby_compact(t.tkeys, t.tabs, cursor)
+        ms = timeit(run, args.iters)
+    elif args.which == "runs":
+        # K18 run-boundary compaction over sorted keys (PMC target)
+        sk = torch.sort(keys).values.contiguous()
+
+        def run():
+            kernels._C.runs_sorted(sk)
         ms = timeit(run, args.iters)
     elif args.which == "partition":
         f = Frame([keys, vals], prefix=1)
