@@ -308,23 +308,27 @@ def test_radix_sort_reduced_bits(kernels):
         assert torch.equal(kernels.radix_sort_keys(k), ref_k), (lo, hi)
 
 
-def test_runs_sorted_matches_fallback(kernels):
+def test_runs_sorted_matches_fallback(kernels, monkeypatch):
     # K18 one-pass run boundaries vs the diff-mask chain, including
-    # all-equal, all-distinct and negative keys
+    # all-equal, all-distinct, negative keys and the large-tile
+    # geometry (>16M rows); the rocPRIM A/B arm must agree too
     from bigslice_amd.ops.cogroup import _runs
     for keys in (
             torch.randint(0, 97, (1_000_003,), dtype=torch.int64),
             torch.zeros(4097, dtype=torch.int64),
             torch.arange(5000, dtype=torch.int64),
             torch.randint(-5, 5, (300_000,), dtype=torch.int64),
+            torch.randint(0, 1 << 18, (17_000_001,), dtype=torch.int64),
             torch.empty(0, dtype=torch.int64)):
         sk = torch.sort(keys).values.to("cuda:0")
-        uniq, starts, ends = _runs(sk)
         # fallback path on the same data (force via CPU then move)
         m_uniq, m_starts, m_ends = _runs(sk.cpu())
-        assert torch.equal(uniq.cpu(), m_uniq)
-        assert torch.equal(starts.cpu(), m_starts)
-        assert torch.equal(ends.cpu(), m_ends)
+        for env in ("0", "1"):
+            monkeypatch.setenv("BIGSLICE_RUNS_ROCPRIM", env)
+            uniq, starts, ends = _runs(sk)
+            assert torch.equal(uniq.cpu(), m_uniq), env
+            assert torch.equal(starts.cpu(), m_starts), env
+            assert torch.equal(ends.cpu(), m_ends), env
 
 
 def test_grouptable_sort_combine_paths(kernels):
