@@ -186,3 +186,36 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
     }
   }
 }
+
+// K16 fast path: segmented reduction over SORTED values given the
+// run boundaries from k_runs_sorted.  One THREAD per run: lanes of a
+// wave own 64 ADJACENT runs, so the wave's serial walks collectively
+// stream one contiguous region (every cache line fully used).  The
+// host guards this kernel to max run length <= 4096 and falls back to
+// rocPRIM reduce_by_key for skewed runs and for float sums (which
+// need its fixed reduction tree for run-to-run determinism; integer
+// adds are exact in any order).  codes: 0=sum 1=min 2=max.
+__global__ __launch_bounds__(256) void k_segreduce_i64(
+    const int64_t* __restrict__ vals,
+    const int64_t* __restrict__ starts,
+    const int64_t* __restrict__ ends, int64_t m, int code,
+    int64_t* __restrict__ out) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < m; i += stride) {
+    const int64_t s = starts[i];
+    const int64_t e = ends[i];
+    int64_t acc;
+    if (code == 0) {
+      acc = 0;
+      for (int64_t j = s; j < e; ++j) acc += vals[j];
+    } else if (code == 1) {
+      acc = vals[s];
+      for (int64_t j = s + 1; j < e; ++j) acc = min(acc, vals[j]);
+    } else {
+      acc = vals[s];
+      for (int64_t j = s + 1; j < e; ++j) acc = max(acc, vals[j]);
+    }
+    out[i] = acc;
+  }
+}

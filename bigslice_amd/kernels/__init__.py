@@ -402,17 +402,40 @@ class GroupTable:
                 col = (batches[0][1][c] if len(batches) == 1
                        else torch.cat([v[c] for _, v in batches]))
                 vcols.append(col[perm])
-        # K16: one decoupled-lookback reduce-by-key pass per value
-        # column — deterministic, exact for ints (wraps like atomics)
+        # K16: segmented reduction per value column.  Fast path for
+        # int64 columns: boundaries come from ONE k_runs_sorted pass
+        # (reused across columns) and k_segreduce walks each run with
+        # one thread — exact for ints in any order.  Guards: skewed
+        # runs (max length > 4096, one thread would serialize) and
+        # float sums (need reduce_by_key's fixed reduction tree for
+        # run-to-run determinism) take the rocPRIM path.
         uk = None
         outs = []
-        for c, v in enumerate(vcols):
-            uq, aggs, cnt = _C.segment_reduce_sorted(
-                ks, v.contiguous(), self.codes[c])
-            if uk is None:
-                m = int(cnt.item())
-                uk = uq[:m]
-            outs.append(aggs[:m])
+        runs = None
+        if all(dt == torch.int64 for dt in self.val_dtypes):
+            uq, starts, cnt = _C.runs_sorted(ks.contiguous())
+            m = int(cnt.item())
+            uk, starts = uq[:m], starts[:m]
+            n_t = torch.tensor([ks.shape[0]], dtype=torch.int64,
+                               device=ks.device)
+            ends = torch.cat([starts[1:], n_t])
+            if m == 0 or int((ends - starts).max().item()) <= 4096:
+                runs = (starts, ends)
+            else:
+                uk = None
+        if runs is not None:
+            for c, v in enumerate(vcols):
+                outs.append(_C.segment_reduce_runs(
+                    v.contiguous(), runs[0], runs[1], self.codes[c]))
+        else:
+            uk = None
+            for c, v in enumerate(vcols):
+                uq, aggs, cnt = _C.segment_reduce_sorted(
+                    ks, v.contiguous(), self.codes[c])
+                if uk is None:
+                    m = int(cnt.item())
+                    uk = uq[:m]
+                outs.append(aggs[:m])
         if self.cap is not None:
             # merge with the provisionally hash-inserted first batch
             self._insert_now(uk, outs, "global")

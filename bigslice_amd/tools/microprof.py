@@ -47,7 +47,7 @@ def main():
     ap.add_argument("which", choices=["groupby", "groupby_pp",
                                       "insert", "insert_rep",
                                       "partition", "hash", "compact",
-                                      "sortcombine", "hashbytes",
+                                      "sortcombine", "sortcombine2", "hashbytes",
                                       "runs"])
     ap.add_argument("--nrep", type=int, default=8)
     ap.add_argument("--cap", type=int, default=0,
@@ -166,6 +166,21 @@ def main():
             ks, vs = _C.radix_sort_kv(keys, vals)
             uq, sm, cnt = _C.segment_reduce_sorted(ks, vs, 0)
             int(cnt.item())
+        out["ms"] = timeit(run, args.iters)
+        out["grows_per_s"] = args.rows / out["ms"] / 1e6
+    if args.which == "sortcombine2":
+        # sort + one k_runs boundary pass + thread-per-run segsum
+        from bigslice_amd.kernels import _C
+
+        def run():
+            ks, vs = _C.radix_sort_kv(keys, vals)
+            uq, starts, cnt = _C.runs_sorted(ks)
+            m = int(cnt.item())
+            starts = starts[:m]
+            n_t = torch.tensor([ks.shape[0]], dtype=torch.int64,
+                               device=ks.device)
+            ends = torch.cat([starts[1:], n_t])
+            _C.segment_reduce_runs(vs, starts, ends, 0)
         out["ms"] = timeit(run, args.iters)
         out["grows_per_s"] = args.rows / out["ms"] / 1e6
     if args.which == "hashbytes":

@@ -331,6 +331,40 @@ def test_runs_sorted_matches_fallback(kernels, monkeypatch):
             assert torch.equal(ends.cpu(), m_ends), env
 
 
+def test_segment_reduce_runs_fast_path(kernels, monkeypatch):
+    # K16 fast path: forced sort-combine must agree with a torch
+    # reference for sum/min/max, including the skewed-run case that
+    # triggers the rocPRIM fallback (max run length > 4096)
+    monkeypatch.setenv("BIGSLICE_GB_COMBINE", "sort")
+    for skew in (False, True):
+        n, nk = 2_000_000, 50_000
+        g = torch.Generator().manual_seed(11 + skew)
+        keys = torch.randint(0, nk, (n,), dtype=torch.int64, generator=g)
+        if skew:
+            hot = torch.rand(n, generator=g) < 0.5
+            keys = torch.where(hot, torch.zeros_like(keys), keys)
+        v1 = torch.randint(-100, 100, (n,), dtype=torch.int64,
+                           generator=g)
+        v2 = torch.randint(-100, 100, (n,), dtype=torch.int64,
+                           generator=g)
+        t = kernels.GroupTable([torch.int64, torch.int64],
+                               ["sum", "max"], "cuda:0")
+        t.insert(keys.cuda(), [v1.cuda(), v2.cuda()])
+        uk, outs = t.finish()
+        order = torch.argsort(uk.cpu())
+        uk_s = uk.cpu()[order]
+        ref_k = torch.unique(keys)
+        assert torch.equal(uk_s, ref_k), skew
+        import torch as _t
+        ref_sum = _t.zeros(nk, dtype=_t.int64).scatter_add_(
+            0, keys, v1)[ref_k]
+        ref_max = _t.full((nk,), -(1 << 62), dtype=_t.int64)
+        ref_max.scatter_reduce_(0, keys, v2, reduce="amax")
+        ref_max = ref_max[ref_k]
+        assert torch.equal(outs[0].cpu()[order], ref_sum), skew
+        assert torch.equal(outs[1].cpu()[order], ref_max), skew
+
+
 def test_grouptable_sort_combine_paths(kernels):
     # Force both combine paths and check they agree with torch.
     import os
