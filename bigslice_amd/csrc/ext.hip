@@ -796,13 +796,12 @@ std::vector<torch::Tensor> segment_sum_sorted_i64(torch::Tensor keys,
 // once via runs_sorted and reuses them for every value column (the
 // rocPRIM reduce_by_key path redoes the key scan per column).
 torch::Tensor segment_reduce_runs(torch::Tensor vals,
-                                  torch::Tensor starts,
-                                  torch::Tensor ends, int64_t code) {
+                                  torch::Tensor starts, int64_t n,
+                                  int64_t code) {
   TORCH_CHECK(vals.is_cuda() && vals.is_contiguous() &&
               vals.scalar_type() == torch::kInt64,
               "segment_reduce_runs: contiguous int64 values required");
-  TORCH_CHECK(starts.is_cuda() && starts.is_contiguous() &&
-              ends.is_cuda() && ends.is_contiguous());
+  TORCH_CHECK(starts.is_cuda() && starts.is_contiguous());
   TORCH_CHECK(code >= 0 && code <= 2,
               "segment_reduce_runs: sum/min/max only");
   const int64_t m = starts.size(0);
@@ -811,9 +810,25 @@ torch::Tensor segment_reduce_runs(torch::Tensor vals,
   const int64_t nb = std::min<int64_t>((m + 255) / 256, 4096);
   hipLaunchKernelGGL(k_segreduce_i64, dim3((int)nb), dim3(256), 0,
                      current_stream(), vals.data_ptr<int64_t>(),
-                     starts.data_ptr<int64_t>(),
-                     ends.data_ptr<int64_t>(), m, (int)code,
+                     starts.data_ptr<int64_t>(), m, n, (int)code,
                      out.data_ptr<int64_t>());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
+// (count, max_run_length) of a runs_sorted result in ONE device
+// tensor, so the host guard pays a single readback.
+torch::Tensor runs_guard(torch::Tensor starts, torch::Tensor cnt,
+                         int64_t n) {
+  TORCH_CHECK(starts.is_cuda() && starts.is_contiguous() &&
+              cnt.is_cuda());
+  auto out = torch::zeros({2}, starts.options());
+  out.index_put_({0}, cnt[0]);
+  hipLaunchKernelGGL(k_runs_guard, dim3(512), dim3(256), 0,
+                     current_stream(),
+                     starts.data_ptr<int64_t>(),
+                     cnt.data_ptr<int64_t>(), n,
+                     (unsigned long long*)(out.data_ptr<int64_t>() + 1));
   HIP_CHECK(hipGetLastError());
   return out;
 }
@@ -1011,6 +1026,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "run boundaries of sorted keys (K18)");
   m.def("segment_reduce_runs", &segment_reduce_runs,
         "int64 segmented reduce with precomputed runs (K16 fast path)");
+  m.def("runs_guard", &runs_guard,
+        "(count, max run length) of a runs_sorted result");
   m.def("segment_reduce_sorted", &segment_reduce_sorted,
         "typed reduce-by-key over sorted pairs (K16; deterministic)");
   m.def("hash_bytes", &hash_bytes,

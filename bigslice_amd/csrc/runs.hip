@@ -197,14 +197,13 @@ __global__ __launch_bounds__(RUNS_BLOCK, 4) void k_runs_sorted(
 // adds are exact in any order).  codes: 0=sum 1=min 2=max.
 __global__ __launch_bounds__(256) void k_segreduce_i64(
     const int64_t* __restrict__ vals,
-    const int64_t* __restrict__ starts,
-    const int64_t* __restrict__ ends, int64_t m, int code,
-    int64_t* __restrict__ out) {
+    const int64_t* __restrict__ starts, int64_t m, int64_t n,
+    int code, int64_t* __restrict__ out) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < m; i += stride) {
     const int64_t s = starts[i];
-    const int64_t e = ends[i];
+    const int64_t e = (i + 1 < m) ? starts[i + 1] : n;
     int64_t acc;
     if (code == 0) {
       acc = 0;
@@ -218,4 +217,27 @@ __global__ __launch_bounds__(256) void k_segreduce_i64(
     }
     out[i] = acc;
   }
+}
+
+// Guard helper: max run length + run count in one 2-word device
+// scalar so the host pays a single readback.  cnt_dev is
+// runs_sorted's count output (read on device — no extra sync).
+__global__ void k_runs_guard(const int64_t* __restrict__ starts,
+                             const int64_t* __restrict__ cnt_dev,
+                             int64_t n,
+                             unsigned long long* __restrict__ out_max) {
+  const int64_t m = *cnt_dev;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  unsigned long long mx = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < m; i += stride) {
+    const int64_t e = (i + 1 < m) ? starts[i + 1] : n;
+    const unsigned long long len = (unsigned long long)(e - starts[i]);
+    mx = mx > len ? mx : len;
+  }
+  for (int off = 32; off; off >>= 1) {
+    const unsigned long long o = __shfl_xor(mx, off, 64);
+    mx = mx > o ? mx : o;
+  }
+  if ((threadIdx.x % 64) == 0) atomicMax(out_max, mx);
 }

@@ -215,12 +215,15 @@ class GroupTable:
         self._sample_ne = (sk[1:] != sk[:-1]).sum()  # device scalar
 
     def _read_sample(self) -> str:
-        """Combine strategy from the sample (measured at 125M rows:
-        LDS-tier hash 2.7 ms at 1k keys; global streaming insert 5.7 ms
-        at 1M keys, degrading to 22 ms at 10M keys where
-        sort+segment-reduce stays ~6 ms).  Thresholds in terms of
-        distinct d over s=512k draws: d<=s/32 ~ <=16k keys -> lds;
-        d<=0.88s ~ <=2M keys -> global; else sort."""
+        """Combine strategy from the sample.  Measured at 125M rows
+        (profiles/sortcombine_crossover_r2.txt): LDS-tier hash 2.7 ms
+        at 1k keys; sort+segment-reduce 3.9-5.1 ms at 64k-10M keys;
+        global streaming insert LOSES that whole band (10.3 ms at 64k
+        keys — ~1900 rows/key pile onto the same table slots and the
+        memory-side atomic unit serializes; 6.0 ms at 1M; 22 ms
+        unsized at 10M).  So: distinct d over s=512k draws d<=s/32
+        ~ <=16k keys -> lds, else sort; "global" remains for forced
+        mode, non-sortable aggregates and table merges."""
         if getattr(self, "_sample_forced", None):
             return self._sample_forced
         distinct = 1 + int(self._sample_ne.item())
@@ -239,15 +242,12 @@ class GroupTable:
                     min(max(int(4 * K), 1024), 1 << 30))
         if dbg:
             import sys
-            mode = ("lds" if distinct * 32 <= s else
-                    "global" if distinct * 100 <= s * 88 else "sort")
+            mode = "lds" if distinct * 32 <= s else "sort"
             print(f"[gb] sample={s} distinct={distinct} "
                   f"cap_hint={self.cap_hint} mode={mode}",
                   file=sys.stderr, flush=True)
         if distinct * 32 <= s:
             return "lds"
-        if distinct * 100 <= s * 88:
-            return "global"
         return "sort"
 
     def insert(self, keys: torch.Tensor, vals: List[torch.Tensor]):
@@ -413,20 +413,17 @@ class GroupTable:
         outs = []
         runs = None
         if all(dt == torch.int64 for dt in self.val_dtypes):
+            nrows = ks.shape[0]
             uq, starts, cnt = _C.runs_sorted(ks.contiguous())
-            m = int(cnt.item())
-            uk, starts = uq[:m], starts[:m]
-            n_t = torch.tensor([ks.shape[0]], dtype=torch.int64,
-                               device=ks.device)
-            ends = torch.cat([starts[1:], n_t])
-            if m == 0 or int((ends - starts).max().item()) <= 4096:
-                runs = (starts, ends)
-            else:
-                uk = None
+            guard = _C.runs_guard(starts, cnt, nrows).cpu()  # one sync
+            m, maxlen = int(guard[0]), int(guard[1])
+            if m == 0 or maxlen <= 4096:
+                uk = uq[:m]
+                runs = starts[:m]
         if runs is not None:
             for c, v in enumerate(vcols):
                 outs.append(_C.segment_reduce_runs(
-                    v.contiguous(), runs[0], runs[1], self.codes[c]))
+                    v.contiguous(), runs, nrows, self.codes[c]))
         else:
             uk = None
             for c, v in enumerate(vcols):

@@ -175,12 +175,9 @@ def main():
         def run():
             ks, vs = _C.radix_sort_kv(keys, vals)
             uq, starts, cnt = _C.runs_sorted(ks)
-            m = int(cnt.item())
-            starts = starts[:m]
-            n_t = torch.tensor([ks.shape[0]], dtype=torch.int64,
-                               device=ks.device)
-            ends = torch.cat([starts[1:], n_t])
-            _C.segment_reduce_runs(vs, starts, ends, 0)
+            g = _C.runs_guard(starts, cnt, ks.shape[0]).cpu()
+            _C.segment_reduce_runs(vs, starts[:int(g[0])],
+                                   ks.shape[0], 0)
         out["ms"] = timeit(run, args.iters)
         out["grows_per_s"] = args.rows / out["ms"] / 1e6
     if args.which == "hashbytes":
