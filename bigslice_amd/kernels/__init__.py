@@ -212,18 +212,21 @@ class GroupTable:
         sk = _C.radix_sort_keys(keys[:prefix].contiguous(),
                                 probe=False)
         self._sample_n = prefix
+        self._sample_batch_n = keys.shape[0]
         self._sample_ne = (sk[1:] != sk[:-1]).sum()  # device scalar
 
     def _read_sample(self) -> str:
-        """Combine strategy from the sample.  Measured at 125M rows
-        (profiles/sortcombine_crossover_r2.txt): LDS-tier hash 2.7 ms
-        at 1k keys; sort+segment-reduce 3.9-5.1 ms at 64k-10M keys;
-        global streaming insert LOSES that whole band (10.3 ms at 64k
-        keys — ~1900 rows/key pile onto the same table slots and the
-        memory-side atomic unit serializes; 6.0 ms at 1M; 22 ms
-        unsized at 10M).  So: distinct d over s=512k draws d<=s/32
-        ~ <=16k keys -> lds, else sort; "global" remains for forced
-        mode, non-sortable aggregates and table merges."""
+        """Combine strategy from the sample.  Three regimes (measured
+        at 125M rows, profiles/sortcombine_crossover_r2.txt):
+        * few keys (d <= s/32 ~ 16k): LDS-tier hash, 2.7 ms at 1k;
+        * contended hash (rows/key >~ 500): the memory-side atomic
+          unit serializes same-slot adds — insert 10.3 ms vs
+          sort+segment-reduce 3.9 ms at 64k keys x 1900 rows/key;
+        * wide keys (sample ~saturated, K >~ 2M): sort; a right-sized
+          insert is 6.2 ms at 10M keys but the sort path stays ~5.1.
+        In between (e.g. the north-star consumer shape, ~125 rows/key)
+        the streaming insert wins IN CONTEXT (it overlaps with reads;
+        the sort path bunches all work at finish): stay "global"."""
         if getattr(self, "_sample_forced", None):
             return self._sample_forced
         distinct = 1 + int(self._sample_ne.item())
@@ -240,15 +243,20 @@ class GroupTable:
             if self.cap_hint is None:
                 self.cap_hint = _next_pow2(
                     min(max(int(4 * K), 1024), 1 << 30))
+        if distinct * 32 <= s:
+            mode = "lds"
+        elif distinct * 100 > s * 88:
+            mode = "sort"  # sample ~saturated: K at least ~2M
+        else:
+            K = max(K, 1.0)
+            rows_per_key = self._sample_batch_n / K
+            mode = "sort" if rows_per_key > 500 else "global"
         if dbg:
             import sys
-            mode = "lds" if distinct * 32 <= s else "sort"
             print(f"[gb] sample={s} distinct={distinct} "
                   f"cap_hint={self.cap_hint} mode={mode}",
                   file=sys.stderr, flush=True)
-        if distinct * 32 <= s:
-            return "lds"
-        return "sort"
+        return mode
 
     def insert(self, keys: torch.Tensor, vals: List[torch.Tensor]):
         n = keys.shape[0]
