@@ -154,6 +154,14 @@ class _RunCursor:
         c = self.buf.columns[0]
         return c[-1]  # tensor scalar for tensor columns, value for lists
 
+    def _take_n(self, n: int) -> Optional[Frame]:
+        if n == 0:
+            return None
+        out = self.buf.slice(0, n)
+        self.buf = self.buf.slice(n, len(self.buf)) \
+            if n < len(self.buf) else None
+        return out
+
     def take_upto(self, cutoff) -> Optional[Frame]:
         """Rows with key <= cutoff (buffer is sorted)."""
         c = self.buf.columns[0]
@@ -166,12 +174,22 @@ class _RunCursor:
         else:
             import bisect
             n = bisect.bisect_right(c, cutoff)
-        if n == 0:
-            return None
-        out = self.buf.slice(0, n)
-        self.buf = self.buf.slice(n, len(self.buf)) \
-            if n < len(self.buf) else None
-        return out
+        return self._take_n(n)
+
+    def take_below(self, cutoff) -> Optional[Frame]:
+        """Rows with key < cutoff (buffer is sorted)."""
+        c = self.buf.columns[0]
+        if isinstance(c, torch.Tensor):
+            n = int(torch.searchsorted(c, cutoff, right=False).item()) \
+                if isinstance(cutoff, torch.Tensor) else \
+                int(torch.searchsorted(
+                    c, torch.tensor(cutoff, dtype=c.dtype,
+                                    device=c.device),
+                    right=False).item())
+        else:
+            import bisect
+            n = bisect.bisect_left(c, cutoff)
+        return self._take_n(n)
 
 
 class MergeReader(Reader):
@@ -197,36 +215,44 @@ class MergeReader(Reader):
             out = live[0].buf
             live[0].buf = None
             return out
-        # cutoff = min over runs of the last buffered first-key.  Rows
-        # with first-key <= cutoff are globally mergeable ONLY once every
-        # run whose buffer ends exactly at the cutoff either has more
-        # buffered beyond it or is at EOF — otherwise a later batch of
-        # that run could continue the same first-key with smaller
-        # secondary key columns.
+        # cutoff = min over NON-EOF runs of the last buffered first-key.
+        # Rows with first-key STRICTLY BELOW the cutoff are globally
+        # complete (every unbuffered row of run i has key >= run i's
+        # buffered max >= cutoff); rows AT the cutoff may continue in a
+        # later batch of a cutoff run (same first-key, smaller secondary
+        # columns), so they stay buffered for the next window.  Only
+        # when no run holds anything below the cutoff — the cutoff
+        # run's buffer is one giant equal-key block — does that run
+        # grow, bounded by the block, not the dataset.  (An earlier
+        # version grew the minimum run on EVERY window: the minimum is
+        # at its own cutoff by definition, so merges of many runs
+        # buffered entire datasets and a 10B-row external sort OOMed.)
         while True:
-            maxes = [c.max_buffered_key() for c in live]
+            non_eof = [c for c in live if not c.eof]
+            if not non_eof:
+                parts = [c.buf for c in live]
+                for c in live:
+                    c.buf = None
+                return sort_frame(Frame.concat(parts))
+            maxes = [c.max_buffered_key() for c in non_eof]
             if isinstance(maxes[0], torch.Tensor):
                 cutoff = torch.stack(list(maxes)).min()
-                at_cut = [bool((m == cutoff).item()) for m in maxes]
             else:
                 cutoff = min(maxes)
-                at_cut = [m == cutoff for m in maxes]
-            grew = False
-            for c, ac in zip(live, at_cut):
-                if ac and not c.eof:
-                    want = len(c.buf) * 2
-                    c.fill(want)
-                    if len(c.buf) > want // 2 or c.eof:
-                        grew = True
-            if not grew:
-                break
-        parts = []
-        for c in live:
-            p = c.take_upto(cutoff)
-            if p is not None:
-                parts.append(p)
-        merged = sort_frame(Frame.concat(parts))
-        return merged
+            parts = []
+            for c in live:
+                p = c.take_upto(cutoff) if c.eof \
+                    else c.take_below(cutoff)
+                if p is not None:
+                    parts.append(p)
+            if parts:
+                return sort_frame(Frame.concat(parts))
+            for c in non_eof:
+                m = c.max_buffered_key()
+                eq = bool((m == cutoff).item()) \
+                    if isinstance(m, torch.Tensor) else m == cutoff
+                if eq:
+                    c.fill(len(c.buf) * 2)
 
 
 def reduce_reader(readers: List[Reader], schema, agg,

@@ -63,6 +63,46 @@ def test_merge_reader_multi_key_ties_across_batches():
         list(zip(k2.tolist(), s2.tolist())))
 
 
+def test_merge_reader_windows_stay_bounded():
+    # Regression: the window-growth loop used to grow the MINIMUM run
+    # on every window (the minimum is at its own cutoff by definition),
+    # buffering entire runs — a 10B-row external sort OOMed on it.
+    # With unique keys every window must stay near the chunk size.
+    runs = []
+    for r in range(4):
+        keys = torch.arange(r, 400_000, 4, dtype=torch.int64)
+        runs.append(frames_of([keys], batch=4096))
+    m = MergeReader(runs, chunk=8192)
+    total = 0
+    expected = None
+    while True:
+        f = m.read()
+        if f is None:
+            break
+        assert len(f) <= 8 * 8192, len(f)
+        col = f.columns[0]
+        if expected is not None:
+            assert int(col[0]) >= expected
+        expected = int(col[-1])
+        assert torch.equal(col, col.sort().values)
+        total += len(f)
+    assert total == 4 * 100_000
+
+
+def test_merge_reader_giant_duplicate_block():
+    # One run is a single giant equal-key block spanning many batches:
+    # the merge must grow ONLY that run (bounded by the block) and keep
+    # ordering with the other run's interleaved keys.
+    k1 = torch.full((50_000,), 7, dtype=torch.int64)
+    k2 = torch.tensor([1, 7, 7, 9, 12], dtype=torch.int64)
+    m = MergeReader([frames_of([k1], batch=1000),
+                     frames_of([k2], batch=2)], chunk=64)
+    out = read_all(m)
+    assert len(out) == 50_005
+    col = out.columns[0]
+    assert torch.equal(col, col.sort().values)
+
+
 def test_reduce_reader_combines_across_streams():
     schema = Schema([torch.int64, torch.int64], 1)
     a_k = torch.tensor([1, 1, 2, 3], dtype=torch.int64)
