@@ -525,6 +525,7 @@ class Frame:
 
 _HW_TOTAL = None
 _HW_SKIP = 0  # stride-cache countdown while safely below the mark
+_HW_FRAC = None  # the threshold the countdown was computed under
 
 
 def over_high_water() -> bool:
@@ -551,12 +552,14 @@ def over_high_water() -> bool:
             return False
     if _HW_TOTAL == 0:
         return False
-    if _HW_SKIP > 0:
-        _HW_SKIP -= 1
-        return False
     import os
     frac = float(os.environ.get("BIGSLICE_STORE_HIGH_WATER", "0.85"))
+    global _HW_FRAC
+    if _HW_SKIP > 0 and frac == _HW_FRAC:
+        _HW_SKIP -= 1
+        return False
     used = torch.cuda.memory_allocated()
     if used < 0.7 * frac * _HW_TOTAL:
         _HW_SKIP = 15
+        _HW_FRAC = frac
     return used > frac * _HW_TOTAL
