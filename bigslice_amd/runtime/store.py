@@ -20,6 +20,19 @@ import torch
 # readers must record_stream so the allocator defers block reuse
 _EVENT_MODE = os.environ.get("BIGSLICE_TASK_SYNC", "1") != "1"
 
+_CUDA_OK = None
+
+
+def _cuda_ok() -> bool:
+    """torch.cuda.is_available() is UNCACHED on ROCm (a full
+    hipGetDeviceCount per call, ~0.1 ms) and this path runs per frame
+    read; availability cannot change after process start."""
+    global _CUDA_OK
+    if _CUDA_OK is None:
+        _CUDA_OK = torch.cuda.is_available()
+    return _CUDA_OK
+
+
 from ..frame import Frame
 from ..sliceio import Reader, codec
 
@@ -188,8 +201,7 @@ class _FrameListReader(Reader):
             return None
         f = self.frames[self.i]
         self.i += 1
-        if f.device != "cpu" and _EVENT_MODE and \
-                torch.cuda.is_available():
+        if f.device != "cpu" and _EVENT_MODE and _cuda_ok():
             # cross-stream consumers: keep the caching allocator from
             # reusing these blocks until the reading stream passes
             # this point (the producer recorded its event; ordering is

@@ -142,7 +142,10 @@ class Spiller:
     def _to_host(self, frame: Frame):
         if frame.device == "cpu":
             return frame
-        pin = torch.cuda.is_available() if self._pin is None else self._pin
+        if self._pin is None:
+            from ..runtime.store import _cuda_ok
+            self._pin = _cuda_ok()
+        pin = self._pin
         if not pin:
             return frame.to("cpu")
         device = frame.device
