@@ -160,6 +160,30 @@ def test_typecheck_tool(tmp_path):
     assert p.stdout.count("\n") == 1  # only the bad call flagged
 
 
+def test_typecheck_tool_literal_types(tmp_path):
+    # annotated builder params vs literal call args (reference
+    # analysis/typecheck checks type compatibility, not just arity)
+    f = tmp_path / "types.py"
+    f.write_text(
+        "import bigslice_amd as bs\n"
+        "def build(nshard: int, path: str, scale: float):\n"
+        "    return None\n"
+        "fv = bs.func(build)\n"
+        "sess = bs.start()\n"
+        "sess.run(fv, 4, 'x', 0.5)\n"     # ok
+        "sess.run(fv, 4, 'x', 2)\n"       # ok: int satisfies float
+        "sess.run(fv, 'four', 'x', 1.0)\n"  # bad: str for int
+        "sess.run(fv, 4, 9, 1.0)\n"       # bad: int for str
+        "sess.run(fv, n, 'x', 1.0)\n")    # non-literal: skipped
+    p = subprocess.run(
+        [sys.executable, "-m", "bigslice_amd.tools.typecheck", str(f)],
+        capture_output=True, text=True)
+    assert p.returncode == 1
+    assert "argument 1 is str, builder annotates int" in p.stdout
+    assert "argument 2 is int, builder annotates str" in p.stdout
+    assert p.stdout.count("\n") == 2
+
+
 def test_kmeans_example_converges():
     sys.path.insert(0, "examples")
     import importlib
