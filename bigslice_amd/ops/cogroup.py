@@ -164,6 +164,13 @@ def _merge_sorted_unique(a, b):
         merged = torch.empty(n + m, dtype=a.dtype, device=a.device)
         merged[pos_a] = a
         merged[pos_b] = b
+    from .. import kernels
+    if (merged.is_cuda and merged.dtype == torch.int64
+            and kernels.have_extension() and merged.shape[0] > 0):
+        # one-pass dedup (K18) instead of compare + boolean index
+        # (whose nonzero sync + gather chain costs three passes)
+        uniq, _starts, cnt = kernels._C.runs_sorted(merged.contiguous())
+        return uniq[:int(cnt.item())]
     mask = torch.ones(merged.shape[0], dtype=torch.bool,
                       device=merged.device)
     mask[1:] = merged[1:] != merged[:-1]
